@@ -1,0 +1,38 @@
+"""beta9_amd — MI355X-native serverless GPU runtime.
+
+A from-scratch framework with the capabilities of beam-cloud/beta9 (Python
+SDK with @endpoint/@function/@task_queue/Pod/Sandbox, gRPC gateway,
+bin-packing scheduler, worker data plane) designed for AMD MI355X nodes:
+KFD/amdgpu device plane, xGMI-topology-aware multi-GPU placement, RCCL
+rendezvous provisioning, and a hand-written HIP cold-start path (HBM arena
+pre-alloc + pinned weight streaming + GPU snapshot/restore).
+"""
+
+from .sdk import (  # noqa: F401
+    Client,
+    ClientConfig,
+    Endpoint,
+    Function,
+    Image,
+    MapClient,
+    Pod,
+    PodInstance,
+    QueueClient,
+    QueueDepthAutoscaler,
+    Sandbox,
+    SandboxInstance,
+    SignalClient,
+    TaskQueue,
+    Volume,
+    asgi,
+    configure,
+    endpoint,
+    function,
+    get_config,
+    realtime,
+    schedule,
+    task_queue,
+)
+from .types import TaskPolicy  # noqa: F401
+
+__version__ = "0.1.0"

@@ -1,0 +1,155 @@
+"""RequestBuffer: per-stub request ring with container discovery, health
+probing, per-container concurrency tokens and HTTP dispatch.
+
+Reference behavior (reference: pkg/abstractions/endpoint/buffer.go —
+discoverContainers :451-556 via the container-address map + /health probe,
+requestTokens :558-636, handleHttpRequest :778, heartBeat :902, afterRequest
+keep-warm refresh :941-952).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import Any, Dict, Optional
+
+import httpx
+
+from ...common import keys
+from ...common.state import StateStore
+
+log = logging.getLogger(__name__)
+
+
+class _ContainerSlot:
+    def __init__(self, addr: str, tokens: int):
+        self.addr = addr
+        self.sema = asyncio.Semaphore(tokens)
+        self.healthy = False
+        self.last_probe = 0.0
+
+
+class RequestBuffer:
+    def __init__(
+        self,
+        store: StateStore,
+        container_repo,
+        stub_id: str,
+        concurrent_requests: int = 1,
+        keep_warm_seconds: float = 10.0,
+        request_timeout: float = 300.0,
+        on_inflight_change=None,
+    ):
+        self.store = store
+        self.containers = container_repo
+        self.stub_id = stub_id
+        self.tokens_per_container = max(1, concurrent_requests)
+        self.keep_warm_seconds = keep_warm_seconds
+        self.request_timeout = request_timeout
+        self.on_inflight_change = on_inflight_change
+        self.inflight = 0
+        self._slots: Dict[str, _ContainerSlot] = {}
+        self._client = httpx.AsyncClient(
+            timeout=httpx.Timeout(request_timeout, connect=5.0),
+            limits=httpx.Limits(max_connections=512, max_keepalive_connections=256),
+        )
+        self._probe_interval = 0.25
+
+    async def close(self) -> None:
+        await self._client.aclose()
+
+    async def _discover(self) -> Dict[str, _ContainerSlot]:
+        addrs = await self.containers.addresses_for_stub(self.stub_id)
+        for cid in list(self._slots.keys()):
+            if cid not in addrs:
+                del self._slots[cid]
+        now = time.monotonic()
+        for cid, addr in addrs.items():
+            slot = self._slots.get(cid)
+            if slot is None or slot.addr != addr:
+                slot = _ContainerSlot(addr, self.tokens_per_container)
+                self._slots[cid] = slot
+            if not slot.healthy and now - slot.last_probe > self._probe_interval:
+                slot.last_probe = now
+                slot.healthy = await self._probe(slot.addr)
+        return self._slots
+
+    async def _probe(self, addr: str) -> bool:
+        try:
+            r = await self._client.get(f"http://{addr}/health", timeout=2.0)
+            return r.status_code == 200
+        except Exception:
+            return False
+
+    async def _acquire_slot(self, deadline: float):
+        """Pick a healthy container with a free token; wait for one."""
+        while time.monotonic() < deadline:
+            slots = await self._discover()
+            best: Optional[tuple] = None
+            for cid, slot in slots.items():
+                if not slot.healthy:
+                    continue
+                free = slot.sema._value  # pick the most-free container
+                if not slot.sema.locked() and (best is None or free > best[2]):
+                    best = (cid, slot, free)
+            if best is not None:
+                cid, slot, _ = best
+                try:
+                    await asyncio.wait_for(slot.sema.acquire(), timeout=0.05)
+                    return cid, slot
+                except asyncio.TimeoutError:
+                    continue
+            await asyncio.sleep(0.005)
+        raise TimeoutError("no healthy container available")
+
+    async def forward(
+        self,
+        method: str,
+        path: str,
+        headers: Optional[Dict[str, str]] = None,
+        content: bytes = b"",
+        json_body: Any = None,
+        wait_timeout: float = 120.0,
+    ) -> httpx.Response:
+        self.inflight += 1
+        if self.on_inflight_change:
+            self.on_inflight_change(self.inflight)
+        try:
+            cid, slot = await self._acquire_slot(time.monotonic() + wait_timeout)
+            try:
+                url = f"http://{slot.addr}{path}"
+                kwargs: Dict[str, Any] = {"headers": headers or {}}
+                if json_body is not None:
+                    kwargs["json"] = json_body
+                elif content:
+                    kwargs["content"] = content
+                resp = await self._client.request(method, url, **kwargs)
+                await self._after_request(cid)
+                return resp
+            except (httpx.ConnectError, httpx.ReadError):
+                slot.healthy = False
+                raise
+            finally:
+                slot.sema.release()
+        finally:
+            self.inflight -= 1
+            if self.on_inflight_change:
+                self.on_inflight_change(self.inflight)
+
+    async def _after_request(self, cid: str) -> None:
+        if self.keep_warm_seconds > 0:
+            await self.store.set(
+                keys.container_keep_warm(self.stub_id, cid), 1, ttl=self.keep_warm_seconds
+            )
+
+    async def healthy_count(self) -> int:
+        return sum(1 for s in (await self._discover()).values() if s.healthy)
+
+    async def wait_healthy(self, n: int = 1, timeout: float = 120.0) -> bool:
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if await self.healthy_count() >= n:
+                return True
+            await asyncio.sleep(0.01)
+        return False

@@ -1,0 +1,159 @@
+"""Pod + Sandbox abstraction.
+
+Pods are arbitrary long-running containers with exposed ports; sandboxes add
+a process-manager surface (exec / run_code / fs ops) (reference:
+pkg/abstractions/pod/ — pod.proto 24 RPCs incl. the Sandbox file/process
+surface; worker-side goproc process manager, pkg/worker/sandbox.go:468-573).
+
+This build runs the process manager INSIDE the container as a first-party
+msgpack-gRPC server (beta9_amd/runner/sandbox.py — the goproc equivalent);
+the gateway proxies Sandbox* RPCs straight to the container address (single
+node, no per-node ContainerService hop needed).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import Any, Dict, Optional
+
+from ..common import keys
+from ..rpc.channel import RpcClient, not_found, rpc
+from ..types import ContainerRequest, StubConfig, new_id
+
+log = logging.getLogger(__name__)
+
+
+class PodService:
+    service_name = "PodService"
+
+    def __init__(self, gateway):
+        self.gw = gateway
+
+    async def _schedule_pod(self, auth: Dict, stub_id: str, wait_timeout: float = 60.0) -> Dict[str, Any]:
+        stub = self.gw.db.get_stub(stub_id)
+        if stub is None:
+            raise not_found(f"stub {stub_id}")
+        cfg: StubConfig = self.gw.db.get_stub_config(stub_id)
+        env = self.gw.runner_env(auth["workspace_id"], stub, cfg)
+        env.update(cfg.env)
+        if cfg.entry_point:
+            env["POD_ENTRYPOINT"] = " ".join(cfg.entry_point)
+        container_id = f"{stub['type'].split('/')[0]}-{stub_id[-8:]}-{new_id()}"
+        request = ContainerRequest(
+            container_id=container_id,
+            env=env,
+            cpu=cfg.runtime.cpu,
+            memory_mb=cfg.runtime.memory_mb,
+            gpu=cfg.runtime.gpu,
+            gpu_request=list(cfg.runtime.gpus),
+            gpu_count=cfg.runtime.gpu_count,
+            image_id=cfg.runtime.image_id,
+            stub_id=stub_id,
+            workspace_id=auth["workspace_id"],
+            stub_type=stub["type"],
+            pool_selector=cfg.pool,
+            mounts=list(cfg.volumes),
+            ports=list(cfg.ports),
+            timestamp=time.time(),
+        )
+        await self.gw.scheduler.run(request)
+        # wait for the container address (sandbox/pod containers serve a PM)
+        deadline = time.monotonic() + wait_timeout
+        addr = None
+        while time.monotonic() < deadline:
+            addr = await self.gw.container_repo.get_address(container_id)
+            if addr:
+                break
+            code = await self.gw.container_repo.get_exit_code(container_id)
+            if code is not None:
+                raise not_found(f"container exited early with code {code}")
+            await asyncio.sleep(0.01)
+        return {"container_id": container_id, "addr": addr or "", "ok": addr is not None}
+
+    @rpc()
+    async def CreatePod(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._schedule_pod(auth, req["stub_id"], float(req.get("timeout", 60.0)))
+
+    @rpc()
+    async def StopPod(self, req: Dict[str, Any], context, auth) -> bool:
+        await self.gw.scheduler.stop(req["container_id"])
+        return True
+
+    @rpc()
+    async def ListPods(self, req: Dict[str, Any], context, auth) -> list:
+        out = []
+        for cid, status in (await self.gw.container_repo.containers_for_stub(req["stub_id"])).items():
+            addr = await self.gw.container_repo.get_address(cid)
+            out.append({"container_id": cid, "status": status, "addr": addr})
+        return out
+
+    # -- sandbox process-manager proxy --------------------------------------
+    async def _pm_call(self, container_id: str, method: str, payload: Dict[str, Any]) -> Any:
+        addr = await self.gw.container_repo.get_address(container_id)
+        if not addr:
+            raise not_found(f"container {container_id} has no address")
+        client = RpcClient(addr)
+        try:
+            return await client.call("ProcessManager", method, payload, timeout=float(payload.get("timeout", 120.0)) + 10.0)
+        finally:
+            await client.close()
+
+    @rpc()
+    async def SandboxExec(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "Exec", req)
+
+    @rpc()
+    async def SandboxRunCode(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "RunCode", req)
+
+    @rpc()
+    async def SandboxStatus(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "Status", req)
+
+    @rpc()
+    async def SandboxListProcesses(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "ListProcesses", req)
+
+    @rpc()
+    async def SandboxKillProcess(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "KillProcess", req)
+
+    @rpc()
+    async def SandboxUploadFile(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "UploadFile", req)
+
+    @rpc()
+    async def SandboxDownloadFile(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "DownloadFile", req)
+
+    @rpc()
+    async def SandboxListFiles(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "ListFiles", req)
+
+    @rpc()
+    async def SandboxStatFile(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "StatFile", req)
+
+    @rpc()
+    async def SandboxDeleteFile(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "DeleteFile", req)
+
+    @rpc()
+    async def SandboxReplaceInFiles(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "ReplaceInFiles", req)
+
+    @rpc()
+    async def SandboxFindInFiles(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self._pm_call(req["container_id"], "FindInFiles", req)
+
+    @rpc()
+    async def SandboxExposePort(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Single-node: the sandbox PM binds the port directly; return the
+        address."""
+        addr = await self.gw.container_repo.get_address(req["container_id"])
+        if not addr:
+            raise not_found(f"container {req['container_id']}")
+        host = addr.split(":")[0]
+        return {"url": f"http://{host}:{req['port']}"}

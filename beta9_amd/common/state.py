@@ -1,0 +1,340 @@
+"""Embedded state store.
+
+The control-plane state plane: key/value with TTL, hashes, sorted sets
+(scheduler backlog), lists with blocking pop (per-worker container-request
+queues), distributed locks, and keyspace events. The reference keeps this
+state in Redis (reference: pkg/repository/*_redis.go, pkg/common/redis.go,
+keys.go, key_events.go); this framework embeds the store in the gateway
+process — workers reach it exclusively through the repository gRPC services,
+exactly like the reference's workers do (pkg/gateway/services/repository/) —
+so a single-node MI355X deployment needs no external Redis.
+
+All operations are async and run on the gateway event loop; consistency is by
+single-threaded execution (no cross-await mutation inside an op).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import fnmatch
+import time
+from dataclasses import dataclass, field
+from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
+
+from sortedcontainers import SortedList
+
+
+@dataclass
+class _Entry:
+    value: Any
+    expires_at: Optional[float] = None  # monotonic deadline
+
+    def expired(self, now: float) -> bool:
+        return self.expires_at is not None and now >= self.expires_at
+
+
+@dataclass
+class _ZSet:
+    # member -> score, plus score-ordered index
+    scores: Dict[str, float] = field(default_factory=dict)
+    index: SortedList = field(default_factory=lambda: SortedList())  # (score, member)
+
+
+class KeyEvent:
+    SET = "set"
+    DEL = "del"
+    EXPIRED = "expired"
+    LPUSH = "lpush"
+    HSET = "hset"
+
+
+class StateStore:
+    """In-process async KV store with the primitives the control plane needs."""
+
+    def __init__(self) -> None:
+        self._kv: Dict[str, _Entry] = {}
+        self._hashes: Dict[str, Dict[str, Any]] = {}
+        self._zsets: Dict[str, _ZSet] = {}
+        self._lists: Dict[str, List[Any]] = {}
+        self._list_cond: Dict[str, asyncio.Condition] = {}
+        self._subs: List[Tuple[str, asyncio.Queue]] = []
+        self._sweeper: Optional[asyncio.Task] = None
+
+    # -- lifecycle ---------------------------------------------------------
+    async def start(self) -> None:
+        if self._sweeper is None:
+            self._sweeper = asyncio.create_task(self._sweep_loop())
+
+    async def stop(self) -> None:
+        if self._sweeper is not None:
+            self._sweeper.cancel()
+            try:
+                await self._sweeper
+            except asyncio.CancelledError:
+                pass
+            self._sweeper = None
+
+    async def _sweep_loop(self) -> None:
+        while True:
+            await asyncio.sleep(0.05)
+            now = time.monotonic()
+            dead = [k for k, e in self._kv.items() if e.expired(now)]
+            for k in dead:
+                del self._kv[k]
+                self._publish(KeyEvent.EXPIRED, k)
+
+    # -- events ------------------------------------------------------------
+    def _publish(self, event: str, key: str) -> None:
+        for pattern, q in self._subs:
+            if fnmatch.fnmatchcase(key, pattern):
+                q.put_nowait((event, key))
+
+    async def subscribe(self, pattern: str) -> "Subscription":
+        q: asyncio.Queue = asyncio.Queue()
+        entry = (pattern, q)
+        self._subs.append(entry)
+        return Subscription(self, entry)
+
+    # -- kv ----------------------------------------------------------------
+    async def set(self, key: str, value: Any, ttl: Optional[float] = None) -> None:
+        exp = time.monotonic() + ttl if ttl else None
+        self._kv[key] = _Entry(value, exp)
+        self._publish(KeyEvent.SET, key)
+
+    async def set_nx(self, key: str, value: Any, ttl: Optional[float] = None) -> bool:
+        """Set if not exists (lock acquire). Returns True if set."""
+        e = self._kv.get(key)
+        if e is not None and not e.expired(time.monotonic()):
+            return False
+        await self.set(key, value, ttl)
+        return True
+
+    async def get(self, key: str) -> Any:
+        e = self._kv.get(key)
+        if e is None:
+            return None
+        if e.expired(time.monotonic()):
+            del self._kv[key]
+            self._publish(KeyEvent.EXPIRED, key)
+            return None
+        return e.value
+
+    async def delete(self, key: str) -> bool:
+        existed = False
+        if self._kv.pop(key, None) is not None:
+            existed = True
+        if self._hashes.pop(key, None) is not None:
+            existed = True
+        if self._zsets.pop(key, None) is not None:
+            existed = True
+        if self._lists.pop(key, None) is not None:
+            existed = True
+        if existed:
+            self._publish(KeyEvent.DEL, key)
+        return existed
+
+    async def exists(self, key: str) -> bool:
+        return (await self.get(key)) is not None or key in self._hashes
+
+    async def expire(self, key: str, ttl: float) -> bool:
+        e = self._kv.get(key)
+        if e is None or e.expired(time.monotonic()):
+            return False
+        e.expires_at = time.monotonic() + ttl
+        return True
+
+    async def ttl(self, key: str) -> Optional[float]:
+        e = self._kv.get(key)
+        if e is None or e.expires_at is None:
+            return None
+        return max(0.0, e.expires_at - time.monotonic())
+
+    async def keys(self, pattern: str) -> List[str]:
+        now = time.monotonic()
+        out = [
+            k
+            for k, e in self._kv.items()
+            if not e.expired(now) and fnmatch.fnmatchcase(k, pattern)
+        ]
+        out += [k for k in self._hashes if fnmatch.fnmatchcase(k, pattern) and k not in self._kv]
+        return out
+
+    async def incr(self, key: str, by: int = 1) -> int:
+        cur = await self.get(key) or 0
+        val = int(cur) + by
+        await self.set(key, val)
+        return val
+
+    # -- hashes --------------------------------------------------------------
+    async def hset(self, key: str, mapping: Dict[str, Any]) -> None:
+        self._hashes.setdefault(key, {}).update(mapping)
+        self._publish(KeyEvent.HSET, key)
+
+    async def hget(self, key: str, f: str) -> Any:
+        return self._hashes.get(key, {}).get(f)
+
+    async def hgetall(self, key: str) -> Dict[str, Any]:
+        return dict(self._hashes.get(key, {}))
+
+    async def hdel(self, key: str, *fields: str) -> None:
+        h = self._hashes.get(key)
+        if h:
+            for f in fields:
+                h.pop(f, None)
+            if not h:
+                del self._hashes[key]
+
+    async def hincr(self, key: str, f: str, by: int = 1) -> int:
+        h = self._hashes.setdefault(key, {})
+        h[f] = int(h.get(f, 0)) + by
+        return h[f]
+
+    # -- sorted sets (scheduler backlog) -------------------------------------
+    async def zadd(self, key: str, member: str, score: float) -> None:
+        z = self._zsets.setdefault(key, _ZSet())
+        old = z.scores.get(member)
+        if old is not None:
+            z.index.remove((old, member))
+        z.scores[member] = score
+        z.index.add((score, member))
+
+    async def zpop_min(self, key: str, n: int = 1) -> List[Tuple[str, float]]:
+        z = self._zsets.get(key)
+        if not z:
+            return []
+        out: List[Tuple[str, float]] = []
+        for _ in range(min(n, len(z.index))):
+            score, member = z.index.pop(0)
+            del z.scores[member]
+            out.append((member, score))
+        if not z.scores:
+            self._zsets.pop(key, None)
+        return out
+
+    async def zrem(self, key: str, member: str) -> bool:
+        z = self._zsets.get(key)
+        if not z or member not in z.scores:
+            return False
+        score = z.scores.pop(member)
+        z.index.remove((score, member))
+        if not z.scores:
+            self._zsets.pop(key, None)
+        return True
+
+    async def zcard(self, key: str) -> int:
+        z = self._zsets.get(key)
+        return len(z.scores) if z else 0
+
+    # -- lists (worker request queues) ----------------------------------------
+    def _cond(self, key: str) -> asyncio.Condition:
+        c = self._list_cond.get(key)
+        if c is None:
+            c = self._list_cond[key] = asyncio.Condition()
+        return c
+
+    async def rpush(self, key: str, *values: Any) -> int:
+        lst = self._lists.setdefault(key, [])
+        lst.extend(values)
+        self._publish(KeyEvent.LPUSH, key)
+        cond = self._cond(key)
+        async with cond:
+            cond.notify_all()
+        return len(lst)
+
+    async def lpop(self, key: str) -> Any:
+        lst = self._lists.get(key)
+        if not lst:
+            return None
+        v = lst.pop(0)
+        if not lst:
+            self._lists.pop(key, None)
+        return v
+
+    async def blpop(self, key: str, timeout: float) -> Any:
+        """Blocking left-pop with timeout; None on timeout."""
+        deadline = time.monotonic() + timeout
+        while True:
+            v = await self.lpop(key)
+            if v is not None:
+                return v
+            remaining = deadline - time.monotonic()
+            if remaining <= 0:
+                return None
+            cond = self._cond(key)
+            async with cond:
+                try:
+                    await asyncio.wait_for(cond.wait(), timeout=remaining)
+                except asyncio.TimeoutError:
+                    return None
+
+    async def llen(self, key: str) -> int:
+        return len(self._lists.get(key, []))
+
+    async def lrange(self, key: str, start: int, stop: int) -> List[Any]:
+        lst = self._lists.get(key, [])
+        if stop == -1:
+            return list(lst[start:])
+        return list(lst[start : stop + 1])
+
+
+class Subscription:
+    def __init__(self, store: StateStore, entry: Tuple[str, asyncio.Queue]):
+        self._store = store
+        self._entry = entry
+
+    async def __aenter__(self) -> "Subscription":
+        return self
+
+    async def __aexit__(self, *exc: Any) -> None:
+        self.close()
+
+    def close(self) -> None:
+        try:
+            self._store._subs.remove(self._entry)
+        except ValueError:
+            pass
+
+    async def get(self, timeout: Optional[float] = None) -> Optional[Tuple[str, str]]:
+        try:
+            if timeout is None:
+                return await self._entry[1].get()
+            return await asyncio.wait_for(self._entry[1].get(), timeout)
+        except asyncio.TimeoutError:
+            return None
+
+    async def events(self) -> AsyncIterator[Tuple[str, str]]:
+        while True:
+            yield await self._entry[1].get()
+
+
+class Lock:
+    """TTL lock over the store (reference: redislock usage in pkg/common)."""
+
+    def __init__(self, store: StateStore, key: str, ttl: float = 10.0):
+        self._store = store
+        self._key = "lock:" + key
+        self._ttl = ttl
+        self._held = False
+
+    async def acquire(self, timeout: float = 0.0) -> bool:
+        deadline = time.monotonic() + timeout
+        while True:
+            if await self._store.set_nx(self._key, 1, ttl=self._ttl):
+                self._held = True
+                return True
+            if time.monotonic() >= deadline:
+                return False
+            await asyncio.sleep(0.01)
+
+    async def release(self) -> None:
+        if self._held:
+            await self._store.delete(self._key)
+            self._held = False
+
+    async def __aenter__(self) -> "Lock":
+        if not await self.acquire(timeout=30.0):
+            raise TimeoutError(f"lock {self._key} not acquired")
+        return self
+
+    async def __aexit__(self, *exc: Any) -> None:
+        await self.release()

@@ -1,0 +1,234 @@
+"""Gateway: control-plane boot.
+
+Boots config -> embedded state store -> backend DB -> repos -> scheduler ->
+task dispatcher -> abstraction services -> gRPC server (+ HTTP API, http.py)
+(reference: pkg/gateway/gateway.go — NewGateway :105, registerServices :376,
+Start :618). Single process per deployment; workers connect over gRPC (or
+embed in-process for tests/bench).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+from typing import Dict, List, Optional
+
+from ..backend.db import BackendDB
+from ..backend.repos import ContainerRepo, WorkerRepo
+from ..common.config import AppConfig, load_config
+from ..common.state import StateStore
+from ..rpc.channel import RpcServer
+from ..scheduler.scheduler import Scheduler
+from ..task.dispatch import Dispatcher
+from ..types import StubConfig
+from .services.core import GatewayService, RunnerService
+from .services.repository import WorkerRepositoryService
+
+log = logging.getLogger(__name__)
+
+
+class Gateway:
+    def __init__(self, cfg: Optional[AppConfig] = None):
+        self.cfg = cfg or load_config()
+        self.store = StateStore()
+        self.db = BackendDB(self.cfg.database.path)
+        self.worker_repo = WorkerRepo(self.store, keepalive_ttl_s=self.cfg.worker.keepalive_ttl_s)
+        self.container_repo = ContainerRepo(self.store)
+        self.dispatcher = Dispatcher(
+            self.store,
+            self.db,
+            monitor_interval_s=self.cfg.task.monitor_interval_s,
+            heartbeat_ttl_s=self.cfg.task.heartbeat_ttl_s,
+        )
+        self.scheduler = Scheduler(self.cfg, self.store, self.worker_repo, self.container_repo)
+        self.rpc_server = RpcServer(
+            host=self.cfg.gateway.host, port=self.cfg.gateway.grpc_port, auth_fn=self._auth
+        )
+        self.grpc_port: int = 0
+        self.http_port: int = 0
+        self._http_server = None
+        self._services: List = []
+        self._instances = {}  # stub_id -> autoscaled instance (endpoint/taskqueue)
+        self._admin_token: str = ""
+        self._runner_tokens: Dict[str, str] = {}  # workspace_id -> token
+        self._started = False
+
+    # -- auth ----------------------------------------------------------------
+    async def _auth(self, token: str, full_method: str):
+        ctx = self.db.authorize_token(token)
+        if ctx is None:
+            return None
+        return {"workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"], "token": token}
+
+    def bootstrap_workspace(self, name: str = "default") -> Dict[str, str]:
+        """Create (or fetch) a workspace + token — first-boot convenience."""
+        ws = self.db.get_workspace_by_name(name)
+        if ws is None:
+            ws = self.db.create_workspace(name)
+        token = self.db.create_token(ws["id"])
+        return {"workspace_id": ws["id"], "workspace_name": name, "token": token}
+
+    def runner_token(self, workspace_id: str) -> str:
+        tok = self._runner_tokens.get(workspace_id)
+        if tok is None:
+            tok = self.db.create_token(workspace_id)
+            self._runner_tokens[workspace_id] = tok
+        return tok
+
+    def runner_env(self, workspace_id: str, stub: Dict, cfg: StubConfig) -> Dict[str, str]:
+        """Base env contract for in-container runners (reference env list:
+        SURVEY.md §A.10 / pkg/abstractions/endpoint/instance.go:51-151)."""
+        env = {
+            "B9_TOKEN": self.runner_token(workspace_id),
+            "B9_GATEWAY_HOST": self.cfg.gateway.host,
+            "B9_GATEWAY_PORT": str(self.grpc_port),
+            "STUB_ID": stub["id"],
+            "STUB_TYPE": stub["type"],
+            "WORKSPACE_ID": workspace_id,
+            "WORKERS": str(cfg.workers),
+            "KEEP_WARM_SECONDS": str(cfg.keep_warm_seconds),
+            "CONCURRENT_REQUESTS": str(cfg.concurrent_requests),
+            "CHECKPOINT_ENABLED": "1" if cfg.checkpoint_enabled else "0",
+            "CALLBACK_URL": cfg.callback_url,
+            "PYTHONUNBUFFERED": "1",
+        }
+        obj_id = (cfg.extra or {}).get("object_id")
+        if obj_id:
+            obj = self.db.get_object(obj_id)
+            if obj:
+                env["B9_WORKSPACE_DIR"] = self._extracted_dir(obj)
+        for name in cfg.secrets:
+            val = self.db.get_secret(workspace_id, name)
+            if val is not None:
+                env[name] = val
+        return env
+
+    def _extracted_dir(self, obj: Dict) -> str:
+        """Extract an uploaded workspace zip once, content-addressed."""
+        import zipfile
+
+        target = os.path.join(self.cfg.storage.root, "extracted", obj["hash"])
+        if not os.path.isdir(target):
+            os.makedirs(target + ".tmp", exist_ok=True)
+            try:
+                with zipfile.ZipFile(obj["path"]) as z:
+                    z.extractall(target + ".tmp")
+                os.rename(target + ".tmp", target)
+            except Exception:
+                log.exception("failed to extract object %s", obj["id"])
+                return ""
+        return target
+
+    # -- boot ------------------------------------------------------------------
+    async def start(self) -> None:
+        if self._started:
+            return
+        os.makedirs(self.cfg.storage.root, exist_ok=True)
+        await self.store.start()
+        await self.dispatcher.start()
+        await self.scheduler.start()
+        self._register_services()
+        self.grpc_port = await self.rpc_server.start()
+        await self._start_http()
+        self._started = True
+        log.info("gateway up: grpc=%d http=%d", self.grpc_port, self.http_port)
+
+    def _register_services(self) -> None:
+        from ..abstractions.endpoint import EndpointExecutor, EndpointService
+        from ..abstractions.function import EXECUTOR as FN_EXEC
+        from ..abstractions.function import FunctionExecutor, FunctionService
+        from ..abstractions.map_queue import MapService, SignalService, SimpleQueueService
+        from ..abstractions.secret import SecretService
+        from ..abstractions.taskqueue import TaskQueueExecutor, TaskQueueService
+        from ..abstractions.volume import VolumeService
+        from ..abstractions.output import OutputService
+        from ..abstractions.pod import PodService
+        from ..abstractions.image import ImageService
+
+        gw_svc = GatewayService(
+            self.db,
+            self.store,
+            self.dispatcher,
+            self.container_repo,
+            self.worker_repo,
+            self.scheduler,
+            objects_path=os.path.join(self.cfg.storage.root, "objects"),
+        )
+        runner_svc = RunnerService(self.dispatcher, self.store, self.db)
+        repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo)
+
+        fn_exec = FunctionExecutor(self.db, self.scheduler, self.store, self.runner_env)
+        self.dispatcher.register(FN_EXEC, fn_exec.execute)
+        fn_svc = FunctionService(self.dispatcher, self.db, self.store)
+
+        self.endpoint_exec = EndpointExecutor(self)
+        self.dispatcher.register("endpoint", self.endpoint_exec.execute)
+        ep_svc = EndpointService(self)
+
+        self.taskqueue_exec = TaskQueueExecutor(self)
+        self.dispatcher.register("taskqueue", self.taskqueue_exec.execute)
+        tq_svc = TaskQueueService(self)
+
+        self._services = [
+            gw_svc,
+            runner_svc,
+            repo_svc,
+            fn_svc,
+            ep_svc,
+            tq_svc,
+            SecretService(self.db),
+            MapService(self.store),
+            SimpleQueueService(self.store),
+            SignalService(self.store),
+            VolumeService(self.db, self.cfg.storage.root),
+            OutputService(self.db, self.cfg.storage.root),
+            PodService(self),
+            ImageService(self),
+        ]
+        for svc in self._services:
+            self.rpc_server.register_service(svc)
+
+    async def _start_http(self) -> None:
+        try:
+            from .http import build_http_app
+            import uvicorn
+
+            app = build_http_app(self)
+            config = uvicorn.Config(
+                app,
+                host=self.cfg.gateway.host,
+                port=self.cfg.gateway.http_port,
+                log_level="warning",
+                lifespan="off",
+            )
+            self._http_server = uvicorn.Server(config)
+            self._http_task = asyncio.create_task(self._http_server.serve())
+            # wait for bind
+            for _ in range(200):
+                if self._http_server.started:
+                    break
+                await asyncio.sleep(0.01)
+            servers = getattr(self._http_server, "servers", [])
+            if servers and servers[0].sockets:
+                self.http_port = servers[0].sockets[0].getsockname()[1]
+        except Exception:
+            log.exception("http server failed to start; gRPC-only mode")
+
+    async def shutdown(self) -> None:
+        if not self._started:
+            return
+        for inst in list(self._instances.values()):
+            await inst.shutdown()
+        await self.scheduler.shutdown()
+        await self.dispatcher.shutdown()
+        if self._http_server is not None:
+            self._http_server.should_exit = True
+            try:
+                await asyncio.wait_for(self._http_task, timeout=5.0)
+            except (asyncio.TimeoutError, asyncio.CancelledError):
+                pass
+        await self.rpc_server.stop()
+        await self.store.stop()
+        self.db.close()
+        self._started = False

@@ -1,0 +1,149 @@
+"""Gateway HTTP API: invoke routes + REST /api/v1.
+
+Reference surface (reference: pkg/gateway/gateway.go:227-291 echo groups,
+pkg/api/v1/*.go handlers, pkg/abstractions/endpoint/http.go:17-118 invoke
+routes): POST /endpoint/id/<stub> dispatches through the endpoint task path;
+/api/v1/{health,containers,tasks,workers,deployments,logs,spans,metrics}
+expose control-plane state. Auth: Bearer workspace token.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+from typing import Any, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse, Response
+
+log = logging.getLogger(__name__)
+
+
+def build_http_app(gateway) -> FastAPI:
+    app = FastAPI(title="beta9-amd gateway")
+
+    async def auth_ctx(request: Request) -> dict:
+        tok = request.headers.get("authorization", "")
+        if tok.startswith("Bearer "):
+            tok = tok[7:]
+        ctx = gateway.db.authorize_token(tok)
+        if ctx is None:
+            raise HTTPException(status_code=401, detail="invalid token")
+        return {"workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
+
+    @app.get("/api/v1/health")
+    async def health():
+        return {"status": "ok"}
+
+    # -- invoke routes ------------------------------------------------------
+    async def _invoke(stub_id: str, request: Request, auth: dict, path: str = "/"):
+        from ..abstractions.endpoint import get_instance
+
+        try:
+            instance = await get_instance(gateway, stub_id)
+        except KeyError:
+            raise HTTPException(status_code=404, detail=f"stub {stub_id} not found")
+        body = await request.body()
+        task = await gateway.dispatcher.send_and_execute(
+            "endpoint",
+            auth["workspace_id"],
+            stub_id,
+            {
+                "method": request.method,
+                "path": path,
+                "headers": {"content-type": request.headers.get("content-type", "application/json")},
+                "content": body,
+            },
+        )
+        done = await gateway.dispatcher.wait(task.task_id, timeout=instance.cfg.timeout + 60.0)
+        result = await gateway.dispatcher.get_result(task.task_id)
+        if result is None:
+            return JSONResponse({"task_id": task.task_id, "status": done.status.value}, status_code=500)
+        if isinstance(result, dict) and "status_code" in result:
+            return JSONResponse(
+                {"task_id": task.task_id, **(result.get("body") if isinstance(result.get("body"), dict) else {"body": result.get("body")})},
+                status_code=result["status_code"],
+            )
+        return JSONResponse({"task_id": task.task_id, "result": result})
+
+    @app.api_route("/endpoint/id/{stub_id}", methods=["GET", "POST"])
+    async def invoke_endpoint(stub_id: str, request: Request, auth: dict = Depends(auth_ctx)):
+        return await _invoke(stub_id, request, auth)
+
+    @app.api_route("/endpoint/id/{stub_id}/{path:path}", methods=["GET", "POST"])
+    async def invoke_endpoint_path(stub_id: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
+        return await _invoke(stub_id, request, auth, path="/" + path)
+
+    @app.api_route("/asgi/id/{stub_id}/{path:path}", methods=["GET", "POST"])
+    async def invoke_asgi(stub_id: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
+        return await _invoke(stub_id, request, auth, path="/" + path)
+
+    # -- deployment-name routing (reference: subdomain middleware A.6) -------
+    @app.api_route("/deployment/{name}/{path:path}", methods=["GET", "POST"])
+    async def invoke_deployment(name: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
+        dep = gateway.db.get_deployment(auth["workspace_id"], name)
+        if dep is None:
+            raise HTTPException(status_code=404, detail=f"deployment {name}")
+        return await _invoke(dep["stub_id"], request, auth, path="/" + path)
+
+    # -- REST API -----------------------------------------------------------
+    @app.get("/api/v1/containers")
+    async def list_containers(auth: dict = Depends(auth_ctx)):
+        out = []
+        for key in await gateway.store.keys("container:state:*"):
+            d = await gateway.store.get(key)
+            if d and d.get("workspace_id") == auth["workspace_id"]:
+                out.append(d)
+        return out
+
+    @app.post("/api/v1/containers/{container_id}/stop")
+    async def stop_container(container_id: str, auth: dict = Depends(auth_ctx)):
+        await gateway.scheduler.stop(container_id)
+        return {"ok": True}
+
+    @app.get("/api/v1/logs/{container_id}")
+    async def container_logs(container_id: str, tail: int = 200, auth: dict = Depends(auth_ctx)):
+        return await gateway.container_repo.get_logs(container_id, tail=tail)
+
+    @app.get("/api/v1/spans/{container_id}")
+    async def container_spans(container_id: str, auth: dict = Depends(auth_ctx)):
+        return await gateway.container_repo.get_spans(container_id)
+
+    @app.get("/api/v1/tasks")
+    async def list_tasks(stub_id: str = "", limit: int = 100, auth: dict = Depends(auth_ctx)):
+        return gateway.db.list_tasks(auth["workspace_id"], stub_id, limit)
+
+    @app.get("/api/v1/tasks/{task_id}")
+    async def get_task(task_id: str, auth: dict = Depends(auth_ctx)):
+        t = await gateway.dispatcher.get_task(task_id)
+        if t is None:
+            row = gateway.db.get_task(task_id)
+            if row is None:
+                raise HTTPException(status_code=404)
+            return row
+        return t.model_dump()
+
+    @app.get("/api/v1/workers")
+    async def list_workers(auth: dict = Depends(auth_ctx)):
+        return [w.model_dump() for w in await gateway.worker_repo.list()]
+
+    @app.get("/api/v1/deployments")
+    async def list_deployments(auth: dict = Depends(auth_ctx)):
+        return gateway.db.list_deployments(auth["workspace_id"])
+
+    @app.get("/api/v1/metrics")
+    async def metrics(auth: dict = Depends(auth_ctx)):
+        return {
+            "scheduler": gateway.scheduler.metrics.snapshot(),
+            "backlog": await gateway.scheduler.backlog_len(),
+        }
+
+    @app.get("/api/v1/outputs/{rel:path}")
+    async def get_output(rel: str, auth: dict = Depends(auth_ctx)):
+        path = os.path.realpath(os.path.join(gateway.cfg.storage.root, rel))
+        if not path.startswith(os.path.realpath(gateway.cfg.storage.root)) or not os.path.isfile(path):
+            raise HTTPException(status_code=404)
+        with open(path, "rb") as f:
+            return Response(content=f.read(), media_type="application/octet-stream")
+
+    return app

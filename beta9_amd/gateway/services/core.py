@@ -1,0 +1,255 @@
+"""GatewayService + RunnerService.
+
+GatewayService is the SDK/CLI-facing catch-all (reference:
+pkg/gateway/gateway.proto:12-444 — auth, object upload, stub get-or-create,
+deploy, container list/stop/logs, task list/status). RunnerService is the
+in-container loop the runners use (reference: the runner dials gateway gRPC
+with BETA9_TOKEN for StartTask/EndTask/args/results — SURVEY.md §A.10;
+containers are gateway clients too).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import os
+import time
+from typing import Any, AsyncIterator, Dict, Optional
+
+from ...backend.db import BackendDB
+from ...backend.repos import ContainerRepo, WorkerRepo
+from ...common import keys
+from ...common.state import StateStore
+from ...rpc.channel import RpcError, invalid, not_found, rpc
+from ...task.dispatch import Dispatcher
+from ...types import StubConfig, StubType, TaskStatus
+
+
+class GatewayService:
+    service_name = "GatewayService"
+
+    def __init__(
+        self,
+        db: BackendDB,
+        store: StateStore,
+        dispatcher: Dispatcher,
+        container_repo: ContainerRepo,
+        worker_repo: WorkerRepo,
+        scheduler,
+        objects_path: str,
+    ):
+        self.db = db
+        self.store = store
+        self.dispatcher = dispatcher
+        self.containers = container_repo
+        self.workers = worker_repo
+        self.scheduler = scheduler
+        self.objects_path = objects_path
+        os.makedirs(objects_path, exist_ok=True)
+
+    # -- auth ---------------------------------------------------------------
+    @rpc(open=True)
+    async def Authorize(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        token = (req or {}).get("token", "")
+        ctx = self.db.authorize_token(token)
+        if ctx is None:
+            return {"ok": False}
+        return {"ok": True, "workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
+
+    # -- objects (workspace code sync) ---------------------------------------
+    @rpc(kind="stream_unary")
+    async def PutObjectStream(self, request_iter, context, auth) -> Dict[str, Any]:
+        """Upload workspace code as chunked bytes; content-addressed dedupe
+        (reference: gateway.proto PutObjectStream + head object)."""
+        ws = auth["workspace_id"]
+        h = hashlib.sha256()
+        chunks = []
+        async for msg in request_iter:
+            data = msg.get("data", b"")
+            h.update(data)
+            chunks.append(data)
+        digest = h.hexdigest()
+        path = os.path.join(self.objects_path, ws, digest)
+        if not os.path.exists(path):
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            with open(path + ".tmp", "wb") as f:
+                for c in chunks:
+                    f.write(c)
+            os.rename(path + ".tmp", path)
+        size = sum(len(c) for c in chunks)
+        obj_id = self.db.create_object(ws, digest, size, path)
+        return {"object_id": obj_id, "hash": digest, "size": size}
+
+    @rpc()
+    async def HeadObject(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ws = auth["workspace_id"]
+        digest = req.get("hash", "")
+        path = os.path.join(self.objects_path, ws, digest)
+        if os.path.exists(path):
+            obj_id = self.db.create_object(ws, digest, os.path.getsize(path), path)
+            return {"exists": True, "object_id": obj_id}
+        return {"exists": False}
+
+    # -- stubs / deployments -------------------------------------------------
+    @rpc()
+    async def GetOrCreateStub(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ws = auth["workspace_id"]
+        name = req.get("name", "")
+        stub_type = req.get("stub_type", "")
+        try:
+            st = StubType(stub_type)
+        except ValueError:
+            raise invalid(f"unknown stub type {stub_type!r}")
+        cfg = StubConfig.model_validate(req.get("config", {}))
+        # normalize per type (reference: pkg/gateway/services/stub.go:32-75)
+        if st.kind in ("function", "schedule"):
+            cfg.keep_warm_seconds = 0
+        if st.is_serve:
+            cfg.autoscaler.min_containers = 0
+        app = self.db.get_or_create_app(ws, req.get("app_name", name or "default"))
+        row = self.db.get_or_create_stub(
+            ws, name, stub_type, cfg, app_id=app["id"], force_create=bool(req.get("force_create"))
+        )
+        return {"stub_id": row["id"], "app_id": app["id"]}
+
+    @rpc()
+    async def DeployStub(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ws = auth["workspace_id"]
+        stub_id = req["stub_id"]
+        name = req.get("name", "")
+        stub = self.db.get_stub(stub_id)
+        if stub is None:
+            raise not_found(f"stub {stub_id}")
+        dep = self.db.create_deployment(ws, name, stub_id, stub["type"], app_id=stub["app_id"])
+        await self.store.set(f"deploy:announce:{stub_id}", dep)
+        return {"deployment_id": dep["id"], "version": dep["version"], "invoke_url": f"/{stub['type'].split('/')[0]}/id/{stub_id}"}
+
+    @rpc()
+    async def GetStubConfig(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        cfg = self.db.get_stub_config(req["stub_id"])
+        if cfg is None:
+            raise not_found(f"stub {req['stub_id']}")
+        return cfg.model_dump()
+
+    @rpc()
+    async def ListDeployments(self, req: Dict[str, Any], context, auth) -> list:
+        return self.db.list_deployments(auth["workspace_id"])
+
+    @rpc()
+    async def StopDeployment(self, req: Dict[str, Any], context, auth) -> bool:
+        self.db.stop_deployment(req["deployment_id"])
+        return True
+
+    # -- containers ----------------------------------------------------------
+    @rpc()
+    async def ListContainers(self, req: Dict[str, Any], context, auth) -> list:
+        out = []
+        for key in await self.store.keys("container:state:*"):
+            d = await self.store.get(key)
+            if d and d.get("workspace_id") == auth["workspace_id"]:
+                out.append(d)
+        return out
+
+    @rpc()
+    async def StopContainer(self, req: Dict[str, Any], context, auth) -> bool:
+        await self.scheduler.stop(req["container_id"])
+        return True
+
+    @rpc()
+    async def GetContainerLogs(self, req: Dict[str, Any], context, auth) -> list:
+        return await self.containers.get_logs(req["container_id"], tail=int(req.get("tail", 200)))
+
+    @rpc()
+    async def GetContainerSpans(self, req: Dict[str, Any], context, auth) -> list:
+        return await self.containers.get_spans(req["container_id"])
+
+    # -- tasks ----------------------------------------------------------------
+    @rpc()
+    async def GetTask(self, req: Dict[str, Any], context, auth) -> Optional[Dict]:
+        t = await self.dispatcher.get_task(req["task_id"])
+        return t.model_dump() if t else None
+
+    @rpc()
+    async def ListTasks(self, req: Dict[str, Any], context, auth) -> list:
+        return self.db.list_tasks(auth["workspace_id"], req.get("stub_id", ""), int(req.get("limit", 100)))
+
+    @rpc()
+    async def StopTask(self, req: Dict[str, Any], context, auth) -> bool:
+        await self.dispatcher.cancel(req["task_id"])
+        return True
+
+    @rpc()
+    async def GetTaskResult(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        t = await self.dispatcher.get_task(req["task_id"])
+        if t is None:
+            raise not_found(f"task {req['task_id']}")
+        result = await self.dispatcher.get_result(req["task_id"])
+        return {"status": t.status.value, "result": result}
+
+    # -- workers (ops surface) -------------------------------------------------
+    @rpc()
+    async def ListWorkers(self, req: Dict[str, Any], context, auth) -> list:
+        return [w.model_dump() for w in await self.workers.list()]
+
+    @rpc()
+    async def CordonWorker(self, req: Dict[str, Any], context, auth) -> bool:
+        w = await self.workers.get(req["worker_id"])
+        if w is None:
+            return False
+        w.cordoned = bool(req.get("cordoned", True))
+        await self.workers.update(w)
+        return True
+
+
+class RunnerService:
+    """In-container runner loop: claim/heartbeat/args/result/end."""
+
+    service_name = "RunnerService"
+
+    def __init__(self, dispatcher: Dispatcher, store: StateStore, db: BackendDB):
+        self.dispatcher = dispatcher
+        self.store = store
+        self.db = db
+
+    @rpc()
+    async def StartTask(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ok = await self.dispatcher.claim(req["task_id"], req.get("container_id", ""))
+        return {"ok": ok}
+
+    @rpc()
+    async def GetTaskArgs(self, req: Dict[str, Any], context, auth) -> Any:
+        return await self.store.get(keys.task_args(req["task_id"]))
+
+    @rpc()
+    async def Heartbeat(self, req: Dict[str, Any], context, auth) -> bool:
+        await self.dispatcher.heartbeat(req["task_id"])
+        return True
+
+    @rpc()
+    async def IsCancelled(self, req: Dict[str, Any], context, auth) -> bool:
+        return await self.dispatcher.is_cancelled(req["task_id"])
+
+    @rpc()
+    async def EndTask(self, req: Dict[str, Any], context, auth) -> bool:
+        status = TaskStatus(req.get("status", TaskStatus.complete.value))
+        await self.dispatcher.complete(req["task_id"], req.get("result"), status=status)
+        return True
+
+    @rpc()
+    async def SetKeepWarm(self, req: Dict[str, Any], context, auth) -> bool:
+        """Refresh the keep-warm TTL key for a serving container."""
+        ttl = float(req.get("ttl", 10.0))
+        if ttl > 0:
+            await self.store.set(
+                keys.container_keep_warm(req["stub_id"], req["container_id"]), 1, ttl=ttl
+            )
+        return True
+
+    @rpc()
+    async def CheckpointReady(self, req: Dict[str, Any], context, auth) -> bool:
+        """Runner signals on_start finished and a hipsnap checkpoint may be
+        taken (reference: /criu/READY_FOR_CHECKPOINT sentinel —
+        sdk runner/common.py:618, pkg/worker/criu.go:2576)."""
+        await self.store.set(
+            f"checkpoint:ready:{req['stub_id']}:{req['container_id']}", time.time()
+        )
+        return True

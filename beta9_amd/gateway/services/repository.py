@@ -1,0 +1,105 @@
+"""WorkerRepositoryService — re-exports the embedded state repos to workers
+over gRPC so worker processes need no direct store access (reference:
+pkg/gateway/services/repository/worker_repo.proto:7-82, worker_repo.go:55-78).
+Method names mirror beta9_amd/worker/repo_client.py RemoteRepoClient.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+from ...backend.repos import ContainerRepo, WorkerRepo
+from ...rpc.channel import rpc
+from ...types import ContainerRequest, ContainerState, ContainerStatus, Worker
+
+
+class WorkerRepositoryService:
+    service_name = "WorkerRepositoryService"
+
+    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo):
+        self._w = worker_repo
+        self._c = container_repo
+
+    @rpc()
+    async def RegisterWorker(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._w.register(Worker.model_validate(req))
+        return True
+
+    @rpc()
+    async def RemoveWorker(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._w.remove(req["worker_id"])
+        return True
+
+    @rpc()
+    async def Keepalive(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._w.keepalive(req["worker_id"])
+        return True
+
+    @rpc()
+    async def GetWorker(self, req: Dict[str, Any], context, auth) -> Optional[Dict]:
+        w = await self._w.get(req["worker_id"])
+        return w.model_dump() if w else None
+
+    @rpc()
+    async def UpdateWorker(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._w.update(Worker.model_validate(req))
+        return True
+
+    @rpc()
+    async def GetNextContainerRequest(self, req: Dict[str, Any], context, auth) -> Optional[Dict]:
+        return await self._w.next_container_request(req["worker_id"], float(req.get("timeout", 1.0)))
+
+    @rpc()
+    async def AddContainerToWorker(self, req: Dict[str, Any], context, auth) -> bool:
+        return await self._w.ack_container(req["worker_id"], req["container_id"], req["token"])
+
+    @rpc()
+    async def ReleaseCapacity(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._w.release_capacity(
+            req["worker_id"], ContainerRequest.model_validate(req["request"])
+        )
+        return True
+
+    @rpc()
+    async def SetContainerState(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.set_state(ContainerState.model_validate(req))
+        return True
+
+    @rpc()
+    async def UpdateContainerStatus(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.update_status(req["container_id"], ContainerStatus(req["status"]))
+        return True
+
+    @rpc()
+    async def DeleteContainerState(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.delete_state(req["container_id"])
+        return True
+
+    @rpc()
+    async def SetContainerAddress(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.set_address(req["container_id"], req["stub_id"], req["addr"])
+        return True
+
+    @rpc()
+    async def RemoveContainerAddress(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.remove_address(req["container_id"], req["stub_id"])
+        return True
+
+    @rpc()
+    async def SetExitCode(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.set_exit_code(req["container_id"], int(req["code"]))
+        return True
+
+    @rpc()
+    async def StopRequested(self, req: Dict[str, Any], context, auth) -> bool:
+        return await self._c.stop_requested(req["container_id"])
+
+    @rpc()
+    async def AppendLog(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.append_log(req["container_id"], req["line"])
+        return True
+
+    @rpc()
+    async def PushSpan(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._c.push_span(req["container_id"], req["span"])
+        return True

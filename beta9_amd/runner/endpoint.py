@@ -1,0 +1,117 @@
+"""Endpoint runner: HTTP server wrapping the user handler.
+
+The in-container server the gateway's RequestBuffer dispatches to
+(reference: sdk/src/beta9/runner/endpoint.py:55-330 — gunicorn+uvicorn
+FastAPI app, /health, on_start before readiness, checkpoint READY signal).
+GPU containers run the MI355X boot path (HIP pre-init + HBM arena +
+optional hipsnap restore) BEFORE the handler import so weight streaming
+lands in reserved HBM.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+import sys
+import time
+from typing import Any
+
+from .common import (
+    GPUBoot,
+    RunnerClient,
+    RunnerEnv,
+    call_handler,
+    load_callable,
+    run_on_start,
+    setup_environment,
+)
+
+log = logging.getLogger("b9.runner.endpoint")
+
+
+def build_app(env: RunnerEnv, gpu: GPUBoot):
+    from fastapi import FastAPI, Request
+    from fastapi.responses import JSONResponse
+
+    app = FastAPI()
+    state: dict = {"ready": False, "on_start_value": None, "handler": None, "client": None}
+
+    @app.get("/health")
+    async def health():
+        if not state["ready"]:
+            return JSONResponse({"status": "starting"}, status_code=503)
+        return {"status": "ok", "gpu_phases": gpu.phases, "restored": gpu.restored}
+
+    @app.post("/{path:path}")
+    async def invoke(request: Request, path: str = ""):
+        try:
+            body = await request.json() if (await request.body()) else {}
+        except Exception:
+            body = {}
+        try:
+            result = await asyncio.wait_for(
+                call_handler(state["handler"], body), timeout=env.timeout
+            )
+            return JSONResponse({"result": result})
+        except asyncio.TimeoutError:
+            return JSONResponse({"error": "timeout"}, status_code=504)
+        except Exception as e:
+            log.exception("handler error")
+            return JSONResponse({"error": str(e)}, status_code=500)
+
+    @app.get("/{path:path}")
+    async def invoke_get(request: Request, path: str = ""):
+        try:
+            result = await asyncio.wait_for(
+                call_handler(state["handler"], dict(request.query_params)),
+                timeout=env.timeout,
+            )
+            return JSONResponse({"result": result})
+        except Exception as e:
+            return JSONResponse({"error": str(e)}, status_code=500)
+
+    async def startup() -> None:
+        client = RunnerClient(env)
+        state["client"] = client
+        state["on_start_value"] = await run_on_start(env)
+        state["handler"] = load_callable(env.handler)
+        state["ready"] = True
+        if env.checkpoint_enabled and not gpu.restored:
+            # on_start done -> signal checkpointability (reference:
+            # /criu/READY_FOR_CHECKPOINT, runner/common.py:618)
+            try:
+                await client.checkpoint_ready()
+            except Exception:
+                log.warning("checkpoint-ready signal failed", exc_info=True)
+
+    app.add_event_handler("startup", startup)
+    return app
+
+
+def main() -> int:
+    logging.basicConfig(level=logging.INFO)
+    env = RunnerEnv.from_env()
+    setup_environment(env)
+    t0 = time.perf_counter()
+    gpu = GPUBoot(env)
+    gpu.boot()
+    app = build_app(env, gpu)
+    import uvicorn
+
+    # Single proc, event-loop concurrency; `workers` >1 would fork and
+    # re-init HIP per proc — on MI355X one proc per GPU is the model, and
+    # in-process async concurrency covers CPU handlers.
+    uvicorn.run(
+        app,
+        host="0.0.0.0",
+        port=env.bind_port,
+        log_level="warning",
+        lifespan="on",
+        access_log=False,
+    )
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

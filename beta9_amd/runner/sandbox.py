@@ -1,0 +1,288 @@
+"""Sandbox process manager — the in-container exec/fs server.
+
+First-party replacement for the reference's goproc binary (reference:
+beam-cloud/goproc, proxied via pkg/worker/sandbox.go:468-573 and
+container_server.go:671-1403): a msgpack-gRPC ProcessManager serving
+exec/run_code/process mgmt/file ops on the container's BIND_PORT. The
+gateway's PodService proxies Sandbox* RPCs here directly.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import base64
+import fnmatch
+import logging
+import os
+import re
+import signal
+import sys
+import time
+from typing import Any, Dict, List, Optional
+
+from ..rpc.channel import RpcServer, invalid, not_found, rpc
+from .common import GPUBoot, RunnerEnv, setup_environment
+
+log = logging.getLogger("b9.runner.sandbox")
+
+
+class _Process:
+    def __init__(self, pid: int, proc: asyncio.subprocess.Process, cmd: str):
+        self.pid = pid
+        self.proc = proc
+        self.cmd = cmd
+        self.stdout = b""
+        self.stderr = b""
+        self.exit_code: Optional[int] = None
+        self.started_at = time.time()
+
+
+class ProcessManager:
+    service_name = "ProcessManager"
+
+    def __init__(self, root: str = "/"):
+        self.root = root
+        self.procs: Dict[int, _Process] = {}
+
+    def _path(self, p: str) -> str:
+        if not os.path.isabs(p):
+            p = os.path.join(os.getcwd(), p)
+        return p
+
+    # -- exec ----------------------------------------------------------------
+    async def _spawn(self, argv: List[str], cwd: str = "", env: Optional[Dict] = None) -> _Process:
+        full_env = dict(os.environ)
+        if env:
+            full_env.update({str(k): str(v) for k, v in env.items()})
+        proc = await asyncio.create_subprocess_exec(
+            *argv,
+            cwd=cwd or None,
+            env=full_env,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            start_new_session=True,
+        )
+        entry = _Process(proc.pid, proc, " ".join(argv))
+        self.procs[proc.pid] = entry
+
+        async def collect() -> None:
+            out, err = await proc.communicate()
+            entry.stdout, entry.stderr = out, err
+            entry.exit_code = proc.returncode
+
+        asyncio.create_task(collect())
+        return entry
+
+    @rpc()
+    async def Exec(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        cmd = req.get("command")
+        if isinstance(cmd, str):
+            argv = ["bash", "-c", cmd]
+        elif isinstance(cmd, list):
+            argv = cmd
+        else:
+            raise invalid("command must be str or list")
+        entry = await self._spawn(argv, cwd=req.get("cwd", ""), env=req.get("env"))
+        if not req.get("background"):
+            timeout = float(req.get("timeout", 120.0))
+            try:
+                await asyncio.wait_for(entry.proc.wait(), timeout=timeout)
+                # give the collector a tick to store output
+                for _ in range(100):
+                    if entry.exit_code is not None:
+                        break
+                    await asyncio.sleep(0.005)
+            except asyncio.TimeoutError:
+                return {"pid": entry.pid, "running": True, "timeout": True}
+        return {
+            "pid": entry.pid,
+            "running": entry.exit_code is None,
+            "exit_code": entry.exit_code,
+            "stdout": entry.stdout.decode(errors="replace"),
+            "stderr": entry.stderr.decode(errors="replace"),
+        }
+
+    @rpc()
+    async def RunCode(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._run_code(req)
+
+    async def _run_code(self, req: Dict[str, Any]) -> Dict[str, Any]:
+        entry = await self._spawn([sys.executable, "-c", req.get("code", "")], cwd=req.get("cwd", ""), env=req.get("env"))
+        try:
+            await asyncio.wait_for(entry.proc.wait(), timeout=float(req.get("timeout", 120.0)))
+            for _ in range(100):
+                if entry.exit_code is not None:
+                    break
+                await asyncio.sleep(0.005)
+        except asyncio.TimeoutError:
+            return {"pid": entry.pid, "running": True, "timeout": True}
+        return {
+            "pid": entry.pid,
+            "running": False,
+            "exit_code": entry.exit_code,
+            "stdout": entry.stdout.decode(errors="replace"),
+            "stderr": entry.stderr.decode(errors="replace"),
+        }
+
+    @rpc()
+    async def Status(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        pid = req.get("pid")
+        if pid is not None:
+            entry = self.procs.get(int(pid))
+            if entry is None:
+                raise not_found(f"pid {pid}")
+            return {
+                "pid": entry.pid,
+                "running": entry.exit_code is None,
+                "exit_code": entry.exit_code,
+                "stdout": entry.stdout.decode(errors="replace"),
+                "stderr": entry.stderr.decode(errors="replace"),
+            }
+        return {"ok": True, "pid": os.getpid(), "uptime": time.time()}
+
+    @rpc()
+    async def ListProcesses(self, req: Dict[str, Any], context, auth) -> List[Dict]:
+        return [
+            {"pid": p.pid, "cmd": p.cmd, "running": p.exit_code is None, "exit_code": p.exit_code}
+            for p in self.procs.values()
+        ]
+
+    @rpc()
+    async def KillProcess(self, req: Dict[str, Any], context, auth) -> bool:
+        entry = self.procs.get(int(req["pid"]))
+        if entry is None or entry.exit_code is not None:
+            return False
+        try:
+            os.killpg(entry.pid, int(req.get("signal", signal.SIGKILL)))
+        except ProcessLookupError:
+            pass
+        return True
+
+    # -- files ----------------------------------------------------------------
+    @rpc()
+    async def UploadFile(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        path = self._path(req["path"])
+        os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+        data = req.get("data", b"")
+        if isinstance(data, str):
+            data = base64.b64decode(data)
+        mode = "ab" if req.get("append") else "wb"
+        with open(path, mode) as f:
+            f.write(data)
+        return {"path": path, "size": len(data)}
+
+    @rpc()
+    async def DownloadFile(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        path = self._path(req["path"])
+        if not os.path.isfile(path):
+            raise not_found(path)
+        offset = int(req.get("offset", 0))
+        length = int(req.get("length", 0)) or None
+        with open(path, "rb") as f:
+            f.seek(offset)
+            data = f.read(length) if length else f.read()
+        return {"data": data, "size": len(data)}
+
+    @rpc()
+    async def ListFiles(self, req: Dict[str, Any], context, auth) -> List[Dict]:
+        path = self._path(req.get("path", "."))
+        if not os.path.isdir(path):
+            raise not_found(path)
+        out = []
+        for name in sorted(os.listdir(path)):
+            fp = os.path.join(path, name)
+            try:
+                st = os.stat(fp)
+                out.append(
+                    {
+                        "name": name,
+                        "is_dir": os.path.isdir(fp),
+                        "size": st.st_size,
+                        "mtime": st.st_mtime,
+                    }
+                )
+            except OSError:
+                pass
+        return out
+
+    @rpc()
+    async def StatFile(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        path = self._path(req["path"])
+        if not os.path.exists(path):
+            return {"exists": False}
+        st = os.stat(path)
+        return {"exists": True, "is_dir": os.path.isdir(path), "size": st.st_size, "mtime": st.st_mtime}
+
+    @rpc()
+    async def DeleteFile(self, req: Dict[str, Any], context, auth) -> bool:
+        path = self._path(req["path"])
+        if os.path.isdir(path):
+            import shutil
+
+            shutil.rmtree(path, ignore_errors=True)
+        elif os.path.exists(path):
+            os.unlink(path)
+        return True
+
+    @rpc()
+    async def ReplaceInFiles(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        root = self._path(req["path"])
+        old, new = req["old"], req["new"]
+        pattern = req.get("pattern", "*")
+        changed = 0
+        for dirpath, _dirs, files in os.walk(root):
+            for fname in files:
+                if not fnmatch.fnmatch(fname, pattern):
+                    continue
+                fp = os.path.join(dirpath, fname)
+                try:
+                    with open(fp, "r", errors="ignore") as f:
+                        content = f.read()
+                except OSError:
+                    continue
+                if old in content:
+                    with open(fp, "w") as f:
+                        f.write(content.replace(old, new))
+                    changed += 1
+        return {"files_changed": changed}
+
+    @rpc()
+    async def FindInFiles(self, req: Dict[str, Any], context, auth) -> List[Dict]:
+        root = self._path(req["path"])
+        regex = re.compile(req["pattern"])
+        out: List[Dict] = []
+        for dirpath, _dirs, files in os.walk(root):
+            for fname in files:
+                fp = os.path.join(dirpath, fname)
+                try:
+                    with open(fp, "r", errors="ignore") as f:
+                        for i, line in enumerate(f, 1):
+                            if regex.search(line):
+                                out.append({"path": fp, "line": i, "text": line.rstrip()})
+                                if len(out) >= int(req.get("limit", 1000)):
+                                    return out
+                except OSError:
+                    continue
+        return out
+
+
+async def main() -> int:
+    env = RunnerEnv.from_env()
+    setup_environment(env)
+    gpu = GPUBoot(env)
+    gpu.boot()
+    pm = ProcessManager()
+    server = RpcServer(host="0.0.0.0", port=env.bind_port)
+    server.register_service(pm)
+    await server.start()
+    log.info("sandbox process manager up on :%d", env.bind_port)
+    await asyncio.Event().wait()  # run until killed
+    return 0
+
+
+if __name__ == "__main__":
+    logging.basicConfig(level=logging.INFO)
+    try:
+        sys.exit(asyncio.run(main()))
+    except KeyboardInterrupt:
+        sys.exit(0)

@@ -1,0 +1,763 @@
+"""Python SDK: decorators and clients.
+
+The user surface of the framework (reference: sdk/src/beta9/ —
+@endpoint/@function/@task_queue decorators, Pod, Sandbox, Image, Volume,
+abstractions/base/runner.py:569-811 prepare_runtime, mixins.py:42 deploy).
+Sync user code drives the async RPC layer through a background event-loop
+thread.
+
+Usage:
+    from beta9_amd import function, endpoint, task_queue, Image, Sandbox
+
+    @function(cpu=1000, memory=512)
+    def hello(x):
+        return x * 2
+
+    hello.remote(21)  # -> 42
+"""
+
+from __future__ import annotations
+
+import asyncio
+import atexit
+import functools
+import hashlib
+import io
+import os
+import sys
+import threading
+import time
+import zipfile
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, Iterable, List, Optional
+
+from .rpc.channel import RpcClient
+from .types import (
+    AutoscalerConfig,
+    AutoscalerType,
+    Mount,
+    MountType,
+    RuntimeConfig,
+    StubConfig,
+    TaskPolicy,
+)
+
+# ---------------------------------------------------------------------------
+# client plumbing
+# ---------------------------------------------------------------------------
+
+
+class _LoopThread:
+    _instance: Optional["_LoopThread"] = None
+
+    def __init__(self) -> None:
+        self.loop = asyncio.new_event_loop()
+        self.thread = threading.Thread(target=self._run, daemon=True, name="b9-sdk-loop")
+        self.thread.start()
+
+    def _run(self) -> None:
+        asyncio.set_event_loop(self.loop)
+        self.loop.run_forever()
+
+    @classmethod
+    def get(cls) -> "_LoopThread":
+        if cls._instance is None or not cls._instance.thread.is_alive():
+            cls._instance = cls()
+        return cls._instance
+
+    def run(self, coro):
+        return asyncio.run_coroutine_threadsafe(coro, self.loop).result()
+
+
+@dataclass
+class ClientConfig:
+    gateway_addr: str = ""
+    token: str = ""
+
+    @classmethod
+    def from_env(cls) -> "ClientConfig":
+        return cls(
+            gateway_addr=os.environ.get("B9_GATEWAY", "127.0.0.1:1993"),
+            token=os.environ.get("B9_TOKEN", ""),
+        )
+
+
+_default_config: Optional[ClientConfig] = None
+
+
+def configure(gateway_addr: str, token: str) -> None:
+    """Point the SDK at a gateway (tests/benches call this; otherwise env)."""
+    global _default_config
+    _default_config = ClientConfig(gateway_addr=gateway_addr, token=token)
+
+
+def get_config() -> ClientConfig:
+    return _default_config or ClientConfig.from_env()
+
+
+class Client:
+    """Sync gateway client."""
+
+    def __init__(self, config: Optional[ClientConfig] = None):
+        self.config = config or get_config()
+        self._rpc: Optional[RpcClient] = None
+        self._lt = _LoopThread.get()
+
+    def _client(self) -> RpcClient:
+        if self._rpc is None:
+            self._rpc = RpcClient(self.config.gateway_addr, token=self.config.token)
+        return self._rpc
+
+    def call(self, service: str, method: str, payload: Any = None, timeout: Optional[float] = None) -> Any:
+        return self._lt.run(self._client().call(service, method, payload, timeout=timeout))
+
+    def stream(self, service: str, method: str, payload: Any = None) -> Iterable[Any]:
+        q: "asyncio.Queue" = None  # type: ignore
+        done = object()
+        import queue as _queue
+
+        out: "_queue.Queue" = _queue.Queue()
+
+        async def pump():
+            try:
+                async for item in self._client().stream(service, method, payload):
+                    out.put(item)
+            except Exception as e:
+                out.put(e)
+            finally:
+                out.put(done)
+
+        fut = asyncio.run_coroutine_threadsafe(pump(), self._lt.loop)
+        while True:
+            item = out.get()
+            if item is done:
+                break
+            if isinstance(item, Exception):
+                raise item
+            yield item
+        fut.result()
+
+    def client_stream(self, service: str, method: str, payloads: Iterable[Any], timeout: Optional[float] = None) -> Any:
+        async def gen():
+            for p in payloads:
+                yield p
+
+        return self._lt.run(self._client().client_stream(service, method, gen(), timeout=timeout))
+
+    def close(self) -> None:
+        if self._rpc is not None:
+            self._lt.run(self._rpc.close())
+            self._rpc = None
+
+
+# ---------------------------------------------------------------------------
+# Image
+# ---------------------------------------------------------------------------
+
+
+@dataclass
+class Image:
+    """Build recipe for a container image (reference: sdk Image surface,
+    SURVEY.md §A.7)."""
+
+    python_version: str = ""
+    python_packages: List[str] = field(default_factory=list)
+    commands: List[str] = field(default_factory=list)
+    base_image: str = ""
+    env_vars: List[str] = field(default_factory=list)
+
+    def add_python_packages(self, pkgs: List[str]) -> "Image":
+        self.python_packages.extend(pkgs)
+        return self
+
+    def add_commands(self, cmds: List[str]) -> "Image":
+        self.commands.extend(cmds)
+        return self
+
+    def with_envs(self, envs: List[str]) -> "Image":
+        self.env_vars.extend(envs)
+        return self
+
+    def from_registry(self, ref: str) -> "Image":
+        self.base_image = ref
+        return self
+
+    @property
+    def is_empty(self) -> bool:
+        return not (self.python_packages or self.commands or self.base_image or self.env_vars)
+
+    def build_payload(self) -> Dict[str, Any]:
+        return {
+            "python_version": self.python_version,
+            "python_packages": self.python_packages,
+            "commands": self.commands,
+            "base_image": self.base_image,
+            "env_vars": self.env_vars,
+        }
+
+    def build(self, client: Optional[Client] = None, verbose: bool = False) -> str:
+        client = client or Client()
+        v = client.call("ImageService", "VerifyImageBuild", self.build_payload())
+        if v["exists"]:
+            return v["image_id"]
+        image_id = ""
+        for msg in client.stream("ImageService", "BuildImage", self.build_payload()):
+            if verbose and msg.get("type") == "log":
+                print(msg["line"])
+            if msg.get("type") == "done":
+                image_id = msg["image_id"]
+            if msg.get("type") == "error":
+                raise RuntimeError(f"image build failed: {msg.get('line')}")
+        if not image_id:
+            raise RuntimeError("image build did not complete")
+        return image_id
+
+
+# ---------------------------------------------------------------------------
+# Volumes / outputs / autoscalers
+# ---------------------------------------------------------------------------
+
+
+@dataclass
+class Volume:
+    name: str
+    mount_path: str
+
+    def mount(self) -> Mount:
+        return Mount(local_path=self.name, mount_path=self.mount_path, mount_type=MountType.volume)
+
+
+@dataclass
+class QueueDepthAutoscaler:
+    max_containers: int = 1
+    tasks_per_container: int = 1
+    min_containers: int = 0
+
+    def to_config(self) -> AutoscalerConfig:
+        return AutoscalerConfig(
+            type=AutoscalerType.queue_depth,
+            max_containers=self.max_containers,
+            tasks_per_container=self.tasks_per_container,
+            min_containers=self.min_containers,
+        )
+
+
+# ---------------------------------------------------------------------------
+# runtime prep shared by all decorators
+# ---------------------------------------------------------------------------
+
+
+def _handler_spec(fn: Callable) -> str:
+    module = fn.__module__
+    if module == "__main__":
+        main_file = getattr(sys.modules.get("__main__"), "__file__", "")
+        module = os.path.splitext(os.path.basename(main_file))[0] if main_file else "__main__"
+    return f"{module}:{fn.__qualname__}"
+
+
+def _zip_workspace(root: str) -> bytes:
+    buf = io.BytesIO()
+    skip_dirs = {".git", "__pycache__", ".pytest_cache", "node_modules", ".venv"}
+    with zipfile.ZipFile(buf, "w", zipfile.ZIP_DEFLATED) as z:
+        for dirpath, dirnames, filenames in os.walk(root):
+            dirnames[:] = [d for d in dirnames if d not in skip_dirs]
+            for fname in filenames:
+                if fname.endswith((".pyc", ".so")):
+                    continue
+                fp = os.path.join(dirpath, fname)
+                if os.path.getsize(fp) > 64 * (1 << 20):
+                    continue
+                z.write(fp, os.path.relpath(fp, root))
+    return buf.getvalue()
+
+
+class RunnerAbstraction:
+    """Shared deploy/runtime-prep machinery (reference:
+    abstractions/base/runner.py prepare_runtime + mixins.py deploy)."""
+
+    stub_kind = "function"
+
+    def __init__(
+        self,
+        func: Optional[Callable] = None,
+        cpu: int = 1000,
+        memory: int = 1024,
+        gpu: str = "",
+        gpu_count: int = 0,
+        image: Optional[Image] = None,
+        timeout: float = 3600.0,
+        retries: int = 3,
+        workers: int = 1,
+        concurrent_requests: int = 1,
+        keep_warm_seconds: float = 10.0,
+        max_pending_tasks: int = 100,
+        on_start: Optional[Callable] = None,
+        volumes: Optional[List[Volume]] = None,
+        secrets: Optional[List[str]] = None,
+        env: Optional[Dict[str, str]] = None,
+        autoscaler: Optional[QueueDepthAutoscaler] = None,
+        checkpoint_enabled: bool = False,
+        pool: str = "",
+        name: str = "",
+        sync_workspace: bool = False,
+        client: Optional[Client] = None,
+    ):
+        self.func = func
+        self.cpu = cpu
+        self.memory = memory
+        self.gpu = gpu
+        self.gpu_count = gpu_count if gpu_count else (1 if gpu else 0)
+        self.image = image or Image()
+        self.timeout = timeout
+        self.retries = retries
+        self.workers = workers
+        self.concurrent_requests = concurrent_requests
+        self.keep_warm_seconds = keep_warm_seconds
+        self.max_pending_tasks = max_pending_tasks
+        self.on_start = on_start
+        self.volumes = volumes or []
+        self.secrets = secrets or []
+        self.env = env or {}
+        self.autoscaler = autoscaler
+        self.checkpoint_enabled = checkpoint_enabled
+        self.pool = pool
+        self.name = name
+        self.sync_workspace = sync_workspace
+        self._client = client
+        self._stub_id: Optional[str] = None
+        if func is not None:
+            functools.update_wrapper(self, func)
+
+    @property
+    def client(self) -> Client:
+        if self._client is None:
+            self._client = Client()
+        return self._client
+
+    def __call__(self, *args, **kwargs):
+        if self.func is None and len(args) == 1 and callable(args[0]) and not kwargs:
+            # decorator used with arguments: @function(cpu=..)
+            self.func = args[0]
+            functools.update_wrapper(self, self.func)
+            return self
+        return self.func(*args, **kwargs)
+
+    def local(self, *args, **kwargs):
+        return self.func(*args, **kwargs)
+
+    def _stub_config(self) -> StubConfig:
+        cfg = StubConfig(
+            runtime=RuntimeConfig(
+                cpu=self.cpu,
+                memory_mb=self.memory,
+                gpu=self.gpu,
+                gpus=[self.gpu] if self.gpu else [],
+                gpu_count=self.gpu_count,
+                image_id="",
+            ),
+            handler=_handler_spec(self.func) if self.func else "",
+            on_start=_handler_spec(self.on_start) if self.on_start else "",
+            keep_warm_seconds=self.keep_warm_seconds,
+            max_pending_tasks=self.max_pending_tasks,
+            task_policy=TaskPolicy(max_retries=self.retries, timeout=self.timeout),
+            workers=self.workers,
+            concurrent_requests=self.concurrent_requests,
+            volumes=[v.mount() for v in self.volumes],
+            secrets=list(self.secrets),
+            env=dict(self.env),
+            autoscaler=self.autoscaler.to_config() if self.autoscaler else AutoscalerConfig(),
+            checkpoint_enabled=self.checkpoint_enabled,
+            pool=self.pool,
+            timeout=self.timeout,
+        )
+        return cfg
+
+    def prepare_runtime(self, stub_type: Optional[str] = None, force: bool = False) -> str:
+        if self._stub_id is not None and not force:
+            return self._stub_id
+        cfg = self._stub_config()
+        if not self.image.is_empty:
+            cfg.runtime.image_id = self.image.build(self.client)
+        if self.sync_workspace:
+            data = _zip_workspace(os.getcwd())
+            r = self.client.client_stream(
+                "GatewayService",
+                "PutObjectStream",
+                [{"data": data[i : i + (1 << 20)]} for i in range(0, len(data), 1 << 20)] or [{"data": b""}],
+            )
+            cfg.extra["object_id"] = r["object_id"]
+        name = self.name or (self.func.__name__ if self.func else "anon")
+        r = self.client.call(
+            "GatewayService",
+            "GetOrCreateStub",
+            {
+                "name": name,
+                "stub_type": stub_type or self.stub_kind,
+                "config": cfg.model_dump(),
+                "app_name": name,
+            },
+        )
+        self._stub_id = r["stub_id"]
+        return self._stub_id
+
+    def deploy(self, name: str = "") -> Dict[str, Any]:
+        stub_id = self.prepare_runtime(stub_type=f"{self.stub_kind}/deployment", force=True)
+        return self.client.call(
+            "GatewayService",
+            "DeployStub",
+            {"stub_id": stub_id, "name": name or self.name or self.func.__name__},
+        )
+
+
+# ---------------------------------------------------------------------------
+# @function
+# ---------------------------------------------------------------------------
+
+
+class Function(RunnerAbstraction):
+    stub_kind = "function"
+
+    def remote(self, *args, **kwargs) -> Any:
+        stub_id = self.prepare_runtime()
+        result = None
+        status = "ERROR"
+        for msg in self.client.stream(
+            "FunctionService",
+            "FunctionInvoke",
+            {"stub_id": stub_id, "args": {"args": list(args), "kwargs": kwargs}},
+        ):
+            if msg.get("type") == "result":
+                status = msg.get("status")
+                result = msg.get("result")
+        if status != "COMPLETE":
+            raise RuntimeError(f"function failed: status={status} result={result!r}")
+        return result
+
+    def map(self, inputs: Iterable[Any]) -> List[Any]:
+        """Parallel fan-out over inputs (reference: function .map())."""
+        stub_id = self.prepare_runtime()
+        lt = _LoopThread.get()
+
+        async def one(x):
+            rpc = RpcClient(self.client.config.gateway_addr, token=self.client.config.token)
+            try:
+                result, status = None, "ERROR"
+                async for msg in rpc.stream(
+                    "FunctionService",
+                    "FunctionInvoke",
+                    {"stub_id": stub_id, "args": {"args": [x], "kwargs": {}}},
+                ):
+                    if msg.get("type") == "result":
+                        status, result = msg.get("status"), msg.get("result")
+                if status != "COMPLETE":
+                    raise RuntimeError(f"map item failed: {result!r}")
+                return result
+            finally:
+                await rpc.close()
+
+        async def run_all(xs):
+            return await asyncio.gather(*(one(x) for x in xs))
+
+        return lt.run(run_all(list(inputs)))
+
+
+def function(*dargs, **dkwargs):
+    if dargs and callable(dargs[0]) and not dkwargs:
+        return Function(dargs[0])
+    return lambda fn: Function(fn, **dkwargs)
+
+
+def schedule(*dargs, **dkwargs):
+    interval = dkwargs.pop("interval_s", 60.0)
+
+    def deco(fn):
+        f = Function(fn, **dkwargs)
+        f.stub_kind = "schedule"
+        f.interval_s = interval
+        return f
+
+    if dargs and callable(dargs[0]):
+        return deco(dargs[0])
+    return deco
+
+
+# ---------------------------------------------------------------------------
+# @endpoint / @asgi
+# ---------------------------------------------------------------------------
+
+
+class Endpoint(RunnerAbstraction):
+    stub_kind = "endpoint"
+
+    def remote(self, timeout: float = 300.0, **kwargs) -> Any:
+        stub_id = self.prepare_runtime()
+        r = self.client.call(
+            "EndpointService",
+            "EndpointInvoke",
+            {"stub_id": stub_id, "args": kwargs, "timeout": timeout},
+            timeout=timeout + 30.0,
+        )
+        if r.get("status") != "COMPLETE":
+            raise RuntimeError(f"endpoint invoke failed: {r!r}")
+        result = r.get("result")
+        if isinstance(result, dict) and "body" in result:
+            body = result["body"]
+            return body.get("result", body) if isinstance(body, dict) else body
+        return result
+
+    def serve(self, timeout: float = 600.0) -> str:
+        stub_id = self.prepare_runtime(stub_type=f"{self.stub_kind}/serve", force=True)
+        self.client.call("EndpointService", "StartEndpointServe", {"stub_id": stub_id, "timeout": timeout})
+        return stub_id
+
+
+def endpoint(*dargs, **dkwargs):
+    if dargs and callable(dargs[0]) and not dkwargs:
+        return Endpoint(dargs[0])
+    return lambda fn: Endpoint(fn, **dkwargs)
+
+
+def asgi(*dargs, **dkwargs):
+    def deco(fn):
+        e = Endpoint(fn, **dkwargs)
+        e.stub_kind = "asgi"
+        return e
+
+    if dargs and callable(dargs[0]):
+        return deco(dargs[0])
+    return deco
+
+
+def realtime(*dargs, **dkwargs):
+    # realtime = asgi with websocket dispatch (reference groups them)
+    return asgi(*dargs, **dkwargs)
+
+
+# ---------------------------------------------------------------------------
+# @task_queue
+# ---------------------------------------------------------------------------
+
+
+class TaskQueue(RunnerAbstraction):
+    stub_kind = "taskqueue"
+
+    def put(self, *args, **kwargs) -> str:
+        stub_id = self.prepare_runtime()
+        r = self.client.call(
+            "TaskQueueService",
+            "TaskQueuePut",
+            {"stub_id": stub_id, "args": {"args": list(args), "kwargs": kwargs}},
+        )
+        if not r.get("ok"):
+            raise RuntimeError(f"put failed: {r.get('error')}")
+        return r["task_id"]
+
+    def get_task(self, task_id: str) -> Dict[str, Any]:
+        return self.client.call("GatewayService", "GetTaskResult", {"task_id": task_id})
+
+    def wait(self, task_id: str, timeout: float = 300.0, poll: float = 0.05) -> Any:
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            r = self.get_task(task_id)
+            if r["status"] in ("COMPLETE", "ERROR", "CANCELLED", "TIMEOUT"):
+                if r["status"] != "COMPLETE":
+                    raise RuntimeError(f"task {task_id} {r['status']}: {r.get('result')!r}")
+                return r.get("result")
+            time.sleep(poll)
+        raise TimeoutError(task_id)
+
+
+def task_queue(*dargs, **dkwargs):
+    if dargs and callable(dargs[0]) and not dkwargs:
+        return TaskQueue(dargs[0])
+    return lambda fn: TaskQueue(fn, **dkwargs)
+
+
+# ---------------------------------------------------------------------------
+# Pod / Sandbox
+# ---------------------------------------------------------------------------
+
+
+class Pod(RunnerAbstraction):
+    stub_kind = "pod"
+
+    def __init__(self, entrypoint: Optional[List[str]] = None, ports: Optional[List[int]] = None, **kwargs):
+        super().__init__(func=None, **kwargs)
+        self.entrypoint = entrypoint or []
+        self.ports = ports or []
+        self.name = kwargs.get("name") or "pod"
+
+    def _stub_config(self) -> StubConfig:
+        cfg = super()._stub_config()
+        cfg.entry_point = list(self.entrypoint)
+        cfg.ports = list(self.ports)
+        if self.keep_warm_seconds == 10.0:
+            cfg.keep_warm_seconds = -1  # pods default to run-until-stopped
+        return cfg
+
+    def create(self, timeout: float = 60.0) -> "PodInstance":
+        stub_id = self.prepare_runtime(stub_type="pod/run", force=True)
+        r = self.client.call("PodService", "CreatePod", {"stub_id": stub_id, "timeout": timeout}, timeout=timeout + 10.0)
+        if not r.get("ok"):
+            raise RuntimeError("pod did not become ready")
+        return PodInstance(self.client, r["container_id"], r["addr"])
+
+
+@dataclass
+class PodInstance:
+    client: Client
+    container_id: str
+    addr: str
+
+    def terminate(self) -> None:
+        self.client.call("PodService", "StopPod", {"container_id": self.container_id})
+
+
+class Sandbox(Pod):
+    stub_kind = "sandbox"
+
+    def __init__(self, **kwargs):
+        kwargs.setdefault("name", "sandbox")
+        super().__init__(**kwargs)
+        self.entrypoint = []  # sandbox always runs the PM
+
+    def create(self, timeout: float = 60.0) -> "SandboxInstance":
+        stub_id = self.prepare_runtime(stub_type="sandbox", force=True)
+        r = self.client.call("PodService", "CreatePod", {"stub_id": stub_id, "timeout": timeout}, timeout=timeout + 10.0)
+        if not r.get("ok"):
+            raise RuntimeError("sandbox did not become ready")
+        return SandboxInstance(self.client, r["container_id"], r["addr"])
+
+
+class SandboxInstance:
+    def __init__(self, client: Client, container_id: str, addr: str):
+        self.client = client
+        self.container_id = container_id
+        self.addr = addr
+        self.fs = SandboxFS(self)
+        self.process = SandboxProcessAPI(self)
+
+    def _call(self, method: str, payload: Dict[str, Any]) -> Any:
+        payload = {"container_id": self.container_id, **payload}
+        return self.client.call("PodService", method, payload, timeout=float(payload.get("timeout", 120.0)) + 15.0)
+
+    def exec(self, command, timeout: float = 120.0, background: bool = False, cwd: str = "", env: Optional[Dict] = None) -> Dict:
+        return self._call("SandboxExec", {"command": command, "timeout": timeout, "background": background, "cwd": cwd, "env": env})
+
+    def run_code(self, code: str, timeout: float = 120.0) -> Dict:
+        return self._call("SandboxRunCode", {"code": code, "timeout": timeout})
+
+    def status(self) -> Dict:
+        return self._call("SandboxStatus", {})
+
+    def expose_port(self, port: int) -> str:
+        return self._call("SandboxExposePort", {"port": port})["url"]
+
+    def terminate(self) -> None:
+        self.client.call("PodService", "StopPod", {"container_id": self.container_id})
+
+
+class SandboxFS:
+    def __init__(self, sb: SandboxInstance):
+        self._sb = sb
+
+    def upload(self, path: str, data: bytes) -> Dict:
+        return self._sb._call("SandboxUploadFile", {"path": path, "data": data})
+
+    def download(self, path: str) -> bytes:
+        return self._sb._call("SandboxDownloadFile", {"path": path})["data"]
+
+    def ls(self, path: str = ".") -> List[Dict]:
+        return self._sb._call("SandboxListFiles", {"path": path})
+
+    def stat(self, path: str) -> Dict:
+        return self._sb._call("SandboxStatFile", {"path": path})
+
+    def rm(self, path: str) -> None:
+        self._sb._call("SandboxDeleteFile", {"path": path})
+
+    def replace_in_files(self, path: str, old: str, new: str, pattern: str = "*") -> Dict:
+        return self._sb._call("SandboxReplaceInFiles", {"path": path, "old": old, "new": new, "pattern": pattern})
+
+    def find_in_files(self, path: str, pattern: str) -> List[Dict]:
+        return self._sb._call("SandboxFindInFiles", {"path": path, "pattern": pattern})
+
+
+class SandboxProcessAPI:
+    def __init__(self, sb: SandboxInstance):
+        self._sb = sb
+
+    def list(self) -> List[Dict]:
+        return self._sb._call("SandboxListProcesses", {})
+
+    def kill(self, pid: int, sig: int = 9) -> bool:
+        return self._sb._call("SandboxKillProcess", {"pid": pid, "signal": sig})
+
+    def status(self, pid: int) -> Dict:
+        return self._sb._call("SandboxStatus", {"pid": pid})
+
+
+# ---------------------------------------------------------------------------
+# Map / Queue / Signal client objects
+# ---------------------------------------------------------------------------
+
+
+class MapClient:
+    def __init__(self, name: str, client: Optional[Client] = None):
+        self.name = name
+        self.client = client or Client()
+
+    def set(self, key: str, value: Any) -> None:
+        self.client.call("MapService", "MapSet", {"name": self.name, "key": key, "value": value})
+
+    def get(self, key: str) -> Any:
+        return self.client.call("MapService", "MapGet", {"name": self.name, "key": key})
+
+    def delete(self, key: str) -> None:
+        self.client.call("MapService", "MapDelete", {"name": self.name, "key": key})
+
+    def keys(self) -> List[str]:
+        return self.client.call("MapService", "MapKeys", {"name": self.name})
+
+    def __len__(self) -> int:
+        return self.client.call("MapService", "MapCount", {"name": self.name})
+
+    __setitem__ = set
+    __getitem__ = get
+
+
+class QueueClient:
+    def __init__(self, name: str, client: Optional[Client] = None):
+        self.name = name
+        self.client = client or Client()
+
+    def put(self, value: Any) -> None:
+        self.client.call("SimpleQueueService", "SimpleQueuePut", {"name": self.name, "value": value})
+
+    def pop(self, timeout: float = 0.0) -> Any:
+        r = self.client.call(
+            "SimpleQueueService", "SimpleQueuePop", {"name": self.name, "timeout": timeout},
+            timeout=timeout + 10.0,
+        )
+        return r["value"] if r["ok"] else None
+
+    def empty(self) -> bool:
+        return self.client.call("SimpleQueueService", "SimpleQueueEmpty", {"name": self.name})
+
+    def __len__(self) -> int:
+        return self.client.call("SimpleQueueService", "SimpleQueueSize", {"name": self.name})
+
+
+class SignalClient:
+    def __init__(self, name: str, client: Optional[Client] = None):
+        self.name = name
+        self.client = client or Client()
+
+    def set(self, ttl: float = 0.0) -> None:
+        self.client.call("SignalService", "SignalSet", {"name": self.name, "ttl": ttl})
+
+    def clear(self) -> None:
+        self.client.call("SignalService", "SignalClear", {"name": self.name})
+
+    def is_set(self) -> bool:
+        return self.client.call("SignalService", "SignalMonitor", {"name": self.name})

@@ -1,0 +1,209 @@
+"""Container lifecycle: the cold-start hot path.
+
+RunContainer mirrors the reference's fully phase-instrumented path
+(reference: pkg/worker/lifecycle.go:581 RunContainer, :1532 spawn, :1975
+runContainer): port reservation -> spec from request -> GPU assignment +
+device/env injection (+ RCCL rendezvous for gpu_count>1) -> checkpoint
+attach -> runtime exec -> address publication -> exit normalization. Every
+phase emits a lifecycle span (common/spans.py) — these are the numbers
+bench.py reports as the cold-start decomposition.
+
+MI355X specifics vs the reference: GPU assignment is xGMI-topology-aware
+(gpu.py), the runner pre-initializes HIP + pre-allocates its HBM arena before
+handler import (runner/common.py + ops hipsnap), and scale-to-zero restore
+streams an HBM snapshot through pinned staging instead of CRIU
+(worker/snapshot.py).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+import socket
+import sys
+import time
+from typing import Dict, Optional
+
+from ..common.spans import SpanRecorder
+from ..types import ContainerRequest, ContainerState, ContainerStatus, StubType
+from .gpu import AllocationError, GPUManager
+from .runtime.base import ContainerSpec, Runtime
+
+log = logging.getLogger(__name__)
+
+# Stub kinds whose containers serve HTTP and need an address published.
+_HTTP_KINDS = {"endpoint", "asgi", "pod", "sandbox"}
+
+_RUNNER_MODULES = {
+    "endpoint": "beta9_amd.runner.endpoint",
+    "asgi": "beta9_amd.runner.endpoint",
+    "function": "beta9_amd.runner.function",
+    "schedule": "beta9_amd.runner.function",
+    "taskqueue": "beta9_amd.runner.taskqueue",
+    "pod": "beta9_amd.runner.pod",
+    "sandbox": "beta9_amd.runner.sandbox",
+    "container": "beta9_amd.runner.function",
+    "shell": "beta9_amd.runner.pod",
+}
+
+
+def free_port() -> int:
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+class ContainerInstance:
+    def __init__(self, request: ContainerRequest):
+        self.request = request
+        self.port: int = 0
+        self.exit_code: Optional[int] = None
+        self.started_at = 0.0
+        self.done = asyncio.Event()
+
+
+class Lifecycle:
+    def __init__(
+        self,
+        worker_id: str,
+        repo,  # DirectRepoClient | RemoteRepoClient
+        runtime: Runtime,
+        gpu_manager: Optional[GPUManager],
+        gateway_host: str,
+        gateway_grpc_port: int,
+        worker_host: str = "127.0.0.1",
+    ):
+        self.worker_id = worker_id
+        self.repo = repo
+        self.runtime = runtime
+        self.gpu = gpu_manager
+        self.gateway_host = gateway_host
+        self.gateway_grpc_port = gateway_grpc_port
+        self.worker_host = worker_host
+        self.containers: Dict[str, ContainerInstance] = {}
+
+    async def run_container(self, request: ContainerRequest) -> int:
+        cid = request.container_id
+        inst = ContainerInstance(request)
+        self.containers[cid] = inst
+
+        def sink(span: Dict) -> None:
+            asyncio.ensure_future(self.repo.push_span(cid, span))
+
+        rec = SpanRecorder(cid, sink=sink)
+        start_total = time.time()
+        gpu_assignment = None
+        try:
+            with rec.span("worker.spec_from_request"):
+                spec, needs_addr = self._spec_from_request(request, inst)
+
+            if request.requires_gpu():
+                with rec.span("worker.gpu_assignment"):
+                    if self.gpu is None or self.gpu.free_count < request.gpu_count:
+                        raise AllocationError("no GPUs available on worker")
+                    gpu_assignment = self.gpu.assign(cid, max(1, request.gpu_count))
+                    spec.env.update(gpu_assignment.env)
+                    spec.devices.extend(gpu_assignment.devices)
+                    rdzv_env = self.gpu.provision_rendezvous(gpu_assignment)
+                    spec.env.update(rdzv_env)
+                    if request.rendezvous is not None:
+                        spec.env.update(request.rendezvous.env)
+
+            if request.checkpoint is not None and request.checkpoint_enabled:
+                # Attach the hipsnap checkpoint location; the runner streams
+                # HBM state back through pinned staging during boot.
+                spec.env["B9_CHECKPOINT_KEY"] = request.checkpoint.remote_key
+                spec.env["B9_CHECKPOINT_ID"] = request.checkpoint.checkpoint_id
+
+            with rec.span("worker.runtime_start"):
+                on_log = self._make_log_cb(cid)
+                handle = await self.runtime.run(spec, on_log=on_log)
+                inst.started_at = time.time()
+
+            await self.repo.update_container_status(cid, ContainerStatus.running.value)
+            if needs_addr:
+                with rec.span("worker.publish_address"):
+                    addr = f"{self.worker_host}:{inst.port}"
+                    await self.repo.set_container_address(cid, request.stub_id, addr)
+
+            rec.record("worker.cold_start", start_total, time.time())
+
+            stop_watch = asyncio.create_task(self._watch_stop(cid))
+            try:
+                exit_code = await self.runtime.wait(cid)
+            finally:
+                stop_watch.cancel()
+            inst.exit_code = exit_code
+            return exit_code
+        except AllocationError as e:
+            log.warning("gpu allocation failed for %s: %s", cid, e)
+            inst.exit_code = 125
+            return 125
+        except Exception:
+            log.exception("container %s failed", cid)
+            inst.exit_code = 1
+            return 1
+        finally:
+            with rec.span("worker.cleanup"):
+                if gpu_assignment is not None and self.gpu is not None:
+                    self.gpu.unassign(cid)
+                await self.runtime.delete(cid)
+                if inst.port:
+                    await self.repo.remove_container_address(cid, request.stub_id)
+                await self.repo.set_exit_code(cid, inst.exit_code if inst.exit_code is not None else 1)
+                await self.repo.delete_container_state(cid)
+            inst.done.set()
+            self.containers.pop(cid, None)
+
+    def _spec_from_request(self, request: ContainerRequest, inst: ContainerInstance):
+        kind = StubType(request.stub_type).kind if request.stub_type else "function"
+        needs_addr = kind in _HTTP_KINDS
+        env = dict(request.env)
+        env.setdefault("CONTAINER_ID", request.container_id)
+        env.setdefault("STUB_ID", request.stub_id)
+        env.setdefault("STUB_TYPE", request.stub_type)
+        env.setdefault("WORKSPACE_ID", request.workspace_id)
+        env.setdefault("B9_GATEWAY_HOST", self.gateway_host)
+        env.setdefault("B9_GATEWAY_PORT", str(self.gateway_grpc_port))
+        env.setdefault("CONTAINER_HOSTNAME", request.hostname or request.container_id)
+        if request.task_id:
+            env.setdefault("TASK_ID", request.task_id)
+        if needs_addr:
+            inst.port = free_port()
+            env["BIND_PORT"] = str(inst.port)
+        entry = list(request.entry_point)
+        if not entry:
+            module = _RUNNER_MODULES.get(kind, "beta9_amd.runner.function")
+            entry = [sys.executable, "-u", "-m", module]
+        spec = ContainerSpec(
+            container_id=request.container_id,
+            entry_point=entry,
+            env=env,
+            cpu_millicores=request.cpu,
+            memory_mb=request.memory_mb,
+            hostname=request.hostname or request.container_id,
+            block_network=request.block_network,
+        )
+        return spec, needs_addr
+
+    def _make_log_cb(self, cid: str):
+        def cb(stream: str, line: str) -> None:
+            asyncio.ensure_future(self.repo.append_log(cid, line))
+
+        return cb
+
+    async def _watch_stop(self, cid: str) -> None:
+        """Poll for stop requests (reference: worker
+        processStopContainerEvents via keyspace events)."""
+        try:
+            while True:
+                await asyncio.sleep(0.2)
+                if await self.repo.stop_requested(cid):
+                    log.info("stop requested for %s", cid)
+                    await self.runtime.kill(cid, 15)
+                    await asyncio.sleep(5.0)
+                    await self.runtime.kill(cid, 9)
+                    return
+        except asyncio.CancelledError:
+            pass

@@ -1,0 +1,165 @@
+"""Worker: the node data plane.
+
+Mirrors the reference's worker main loop (reference: pkg/worker/worker.go —
+NewWorker :351, Run :663: stream GetNextContainerRequest, ack with delivery
+token :749-777, semaphore-bounded concurrent starts :777-816, capacity
+keepalive, idle shutdown :1006). One worker process (or embedded instance)
+per node; on an MI355X node it owns all 8 GPUs through the GPUManager.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+from typing import Optional
+
+from ..types import ContainerRequest, Worker as WorkerState, WorkerStatus
+from .gpu import GPUManager
+from .lifecycle import Lifecycle
+from .runtime.base import Runtime
+from .runtime.proc import ProcRuntime
+
+log = logging.getLogger(__name__)
+
+
+class Worker:
+    def __init__(
+        self,
+        worker_id: str,
+        repo,  # DirectRepoClient | RemoteRepoClient
+        cpu_limit: int = 64000,
+        memory_limit_mb: int = 131072,
+        gpu_type: str = "",
+        gpu_count: int = 0,
+        pool_name: str = "default",
+        runtime: Optional[Runtime] = None,
+        gpu_manager: Optional[GPUManager] = None,
+        gateway_host: str = "127.0.0.1",
+        gateway_grpc_port: int = 0,
+        container_start_limit: int = 32,
+        keepalive_interval_s: float = 2.0,
+        idle_shutdown_s: float = 0.0,
+        machine_id: str = "",
+    ):
+        self.id = worker_id
+        self.repo = repo
+        self.pool_name = pool_name
+        self.runtime = runtime or ProcRuntime()
+        if gpu_manager is None and gpu_type:
+            gpu_manager = GPUManager()
+        self.gpu_manager = gpu_manager
+        actual_gpus = gpu_manager.total if gpu_manager else 0
+        if gpu_count and gpu_manager:
+            actual_gpus = min(gpu_count, gpu_manager.total)
+        self.state = WorkerState(
+            id=worker_id,
+            status=WorkerStatus.available,
+            total_cpu=cpu_limit,
+            total_memory_mb=memory_limit_mb,
+            total_gpu_count=actual_gpus,
+            free_cpu=cpu_limit,
+            free_memory_mb=memory_limit_mb,
+            free_gpu_count=actual_gpus,
+            gpu=gpu_type or (gpu_manager.gpu_type if gpu_manager else ""),
+            pool_name=pool_name,
+            machine_id=machine_id,
+        )
+        self.lifecycle = Lifecycle(
+            worker_id=worker_id,
+            repo=repo,
+            runtime=self.runtime,
+            gpu_manager=gpu_manager,
+            gateway_host=gateway_host,
+            gateway_grpc_port=gateway_grpc_port,
+        )
+        self._sema = asyncio.Semaphore(container_start_limit)
+        self._keepalive_interval_s = keepalive_interval_s
+        self._idle_shutdown_s = idle_shutdown_s
+        self._tasks: set = set()
+        self._stopping = asyncio.Event()
+        self._run_task: Optional[asyncio.Task] = None
+        self._idle_since = None
+
+    async def start(self) -> None:
+        await self.repo.register_worker(self.state)
+        self._run_task = asyncio.create_task(self._run())
+        ka = asyncio.create_task(self._keepalive_loop())
+        self._tasks.add(ka)
+
+    async def _run(self) -> None:
+        import time as _time
+
+        self._idle_since = _time.monotonic()
+        while not self._stopping.is_set():
+            try:
+                d = await self.repo.next_container_request(self.id, timeout=1.0)
+            except asyncio.CancelledError:
+                raise
+            except Exception:
+                if self._stopping.is_set():
+                    return
+                log.exception("request stream error; retrying")
+                await asyncio.sleep(0.5)
+                continue
+            if d is None:
+                if (
+                    self._idle_shutdown_s > 0
+                    and not self.lifecycle.containers
+                    and _time.monotonic() - self._idle_since > self._idle_shutdown_s
+                ):
+                    log.info("worker %s idle shutdown", self.id)
+                    await self.shutdown()
+                    return
+                continue
+            request = ContainerRequest.model_validate(d)
+            ok = await self.repo.ack_container(self.id, request.container_id, request.delivery_token)
+            if not ok:
+                log.warning("stale delivery token for %s; skipping", request.container_id)
+                continue
+            self._idle_since = _time.monotonic()
+            t = asyncio.create_task(self._handle(request))
+            self._tasks.add(t)
+            t.add_done_callback(self._tasks.discard)
+
+    async def _handle(self, request: ContainerRequest) -> None:
+        async with self._sema:
+            try:
+                await self.lifecycle.run_container(request)
+            finally:
+                await self.repo.release_capacity(self.id, request)
+                import time as _time
+
+                self._idle_since = _time.monotonic()
+
+    async def _keepalive_loop(self) -> None:
+        try:
+            while not self._stopping.is_set():
+                await self.repo.keepalive(self.id)
+                await asyncio.sleep(self._keepalive_interval_s)
+        except asyncio.CancelledError:
+            pass
+        except Exception:
+            log.exception("keepalive loop died")
+
+    async def drain(self, timeout: float = 30.0) -> None:
+        """Wait for running containers to finish (graceful stop)."""
+        deadline = asyncio.get_event_loop().time() + timeout
+        while self.lifecycle.containers and asyncio.get_event_loop().time() < deadline:
+            await asyncio.sleep(0.1)
+
+    async def shutdown(self) -> None:
+        self._stopping.set()
+        for cid in list(self.lifecycle.containers):
+            await self.runtime.kill(cid, 15)
+        await self.drain(timeout=10.0)
+        for cid in list(self.lifecycle.containers):
+            await self.runtime.kill(cid, 9)
+        for t in list(self._tasks):
+            t.cancel()
+        if self._run_task is not None and self._run_task is not asyncio.current_task():
+            self._run_task.cancel()
+        try:
+            await self.repo.remove_worker(self.id)
+        except Exception:
+            pass

@@ -1,0 +1,46 @@
+"""Fixture user handlers for e2e tests (reference analog:
+sdk/tests/fixtures/customer_apps)."""
+
+import os
+import time
+
+
+def hello(x=1):
+    return {"hello": x * 2}
+
+
+def add(a, b):
+    return a + b
+
+
+def envdump():
+    return {
+        "container_id": os.environ.get("CONTAINER_ID", ""),
+        "stub_type": os.environ.get("STUB_TYPE", ""),
+        "secret": os.environ.get("TEST_SECRET", ""),
+    }
+
+
+def slow(seconds=1.0):
+    time.sleep(float(seconds))
+    return "done"
+
+
+def boom():
+    raise ValueError("expected failure")
+
+
+def echo(**kwargs):
+    return kwargs
+
+
+_started = {"on_start_ran": False}
+
+
+def on_start():
+    _started["on_start_ran"] = True
+    return "warm"
+
+
+def check_started(**kwargs):
+    return _started["on_start_ran"]
