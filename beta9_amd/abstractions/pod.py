@@ -59,7 +59,8 @@ class PodService:
             timestamp=time.time(),
         )
         await self.gw.scheduler.run(request)
-        # wait for the container address (sandbox/pod containers serve a PM)
+        # wait for the container address, then probe the PM port until it
+        # accepts (reference: waitForProcessManager pkg/worker/sandbox.go:468)
         deadline = time.monotonic() + wait_timeout
         addr = None
         while time.monotonic() < deadline:
@@ -70,7 +71,22 @@ class PodService:
             if code is not None:
                 raise not_found(f"container exited early with code {code}")
             await asyncio.sleep(0.01)
-        return {"container_id": container_id, "addr": addr or "", "ok": addr is not None}
+        ready = False
+        if addr:
+            host, _, port = addr.partition(":")
+            while time.monotonic() < deadline:
+                try:
+                    _r, w = await asyncio.wait_for(
+                        asyncio.open_connection(host, int(port)), timeout=1.0
+                    )
+                    w.close()
+                    ready = True
+                    break
+                except OSError:
+                    await asyncio.sleep(0.02)
+                except asyncio.TimeoutError:
+                    pass
+        return {"container_id": container_id, "addr": addr or "", "ok": ready}
 
     @rpc()
     async def CreatePod(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:

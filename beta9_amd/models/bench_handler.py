@@ -1,0 +1,72 @@
+"""Bench endpoint handler: the GPU bf16 serving workload dispatched through
+the full platform path (gateway -> scheduler -> worker -> container -> this
+module). Importable by the endpoint runner as
+`beta9_amd.models.bench_handler:infer` with `:load` as on_start.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Any, Dict, Optional
+
+_model = None
+_load_info: Dict[str, Any] = {}
+
+
+def load() -> Dict[str, Any]:
+    """on_start: build the model (random-init weights into the HBM arena).
+    The time spent here IS the cold-start weight phase the platform
+    overlaps/snapshots."""
+    global _model, _load_info
+    if _model is not None:
+        return _load_info
+    t0 = time.perf_counter()
+    import torch
+
+    hidden = int(os.environ.get("B9_BENCH_HIDDEN", "4096"))
+    inter = int(os.environ.get("B9_BENCH_INTER", "14336"))
+    layers = int(os.environ.get("B9_BENCH_LAYERS", "8"))
+    if torch.cuda.is_available():
+        from .mlp import SyntheticServingModel
+
+        _model = SyntheticServingModel(hidden=hidden, inter=inter, layers=layers, device="cuda:0")
+        native = True
+        param_bytes = _model.param_bytes
+    else:
+        # CPU dev-path: tiny torch model so the full dispatch loop still runs
+        _model = _CPUModel(hidden=min(hidden, 256), layers=2)
+        native = False
+        param_bytes = _model.param_bytes
+    _load_info = {
+        "load_ms": (time.perf_counter() - t0) * 1000,
+        "native": native,
+        "param_bytes": param_bytes,
+    }
+    return _load_info
+
+
+class _CPUModel:
+    def __init__(self, hidden: int = 256, layers: int = 2):
+        import torch
+
+        self.ws = [torch.randn(hidden, hidden) * 0.02 for _ in range(layers)]
+        self.param_bytes = sum(w.numel() * 4 for w in self.ws)
+        self.hidden = hidden
+
+    def infer(self, batch_tokens: int = 16) -> Dict:
+        import torch
+
+        x = torch.randn(min(batch_tokens, 32), self.hidden)
+        t0 = time.perf_counter()
+        for w in self.ws:
+            x = torch.relu(x @ w)
+        return {"ms": (time.perf_counter() - t0) * 1000, "tflops": 0.0, "checksum": float(x.abs().mean())}
+
+
+def infer(tokens: int = 256, **kwargs) -> Dict[str, Any]:
+    if _model is None:
+        load()
+    r = _model.infer(batch_tokens=int(tokens))
+    r["load"] = _load_info
+    return r

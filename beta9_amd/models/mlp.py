@@ -1,0 +1,111 @@
+"""Synthetic bf16 serving model for GPU endpoint benches.
+
+A llama-style transformer MLP stack (rmsnorm -> gate/up projections ->
+silu*mul -> down projection, residual) with random-init weights, running
+entirely on the hand-written gfx950 kernels (beta9_amd/ops) with weights
+resident in the hipsnap HBM arena — this is the "single-GPU bf16 handler"
+of BASELINE.json config 2, and the payload whose cold-start weight
+streaming the framework optimizes.
+
+There is no network in the target environment, so weights are random-init
+by design (synthetic handlers / random-init weights).
+"""
+
+from __future__ import annotations
+
+import math
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from ..ops import gemm as ops
+
+
+class MLPBlock:
+    def __init__(self, hidden: int, inter: int, device: str, arena=None, prefix: str = ""):
+        self.hidden = hidden
+        self.inter = inter
+
+        def make(name: str, shape):
+            n = math.prod(shape)
+            if arena is not None and arena.native:
+                t = arena.tensor(f"{prefix}{name}", shape, dtype="bfloat16")
+            else:
+                t = torch.empty(shape, dtype=torch.bfloat16, device=device)
+            with torch.no_grad():
+                t.normal_(0, 0.02)
+            return t
+
+        # weights in torch Linear layout [out, in] = the GEMM's Bt operand
+        self.w_gate = make("w_gate", (inter, hidden))
+        self.w_up = make("w_up", (inter, hidden))
+        self.w_down = make("w_down", (hidden, inter))
+        self.norm_w = make("norm_w", (hidden,))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        h = ops.rmsnorm(x, self.norm_w)
+        gate = ops.gemm_nt(h, self.w_gate)
+        up = ops.gemm_nt(h, self.w_up)
+        act = ops.silu_mul(gate, up)
+        return x + ops.gemm_nt(act, self.w_down)
+
+
+class SyntheticServingModel:
+    """N MLP blocks; forward = one serving "inference" of the bench handler."""
+
+    def __init__(
+        self,
+        hidden: int = 4096,
+        inter: int = 14336,
+        layers: int = 4,
+        device: str = "cuda:0",
+        use_arena: bool = True,
+    ):
+        self.hidden = hidden
+        self.inter = inter
+        self.device = device
+        self.arena = None
+        if use_arena and torch.cuda.is_available():
+            try:
+                from ..ops import hipsnap
+
+                nbytes = layers * (2 * inter * hidden + hidden * inter + hidden) * 2
+                self.arena = hipsnap.Arena(
+                    device=int(device.split(":")[-1]) if ":" in device else 0,
+                    size_bytes=int(nbytes * 1.25) + (64 << 20),
+                )
+                if not self.arena.native:
+                    self.arena = None
+            except Exception:
+                self.arena = None
+        self.blocks = [
+            MLPBlock(hidden, inter, device, arena=self.arena, prefix=f"l{i}.")
+            for i in range(layers)
+        ]
+
+    @property
+    def param_bytes(self) -> int:
+        return sum(
+            b.w_gate.numel() + b.w_up.numel() + b.w_down.numel() + b.norm_w.numel()
+            for b in self.blocks
+        ) * 2
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        for b in self.blocks:
+            x = b.forward(x)
+        return x
+
+    def infer(self, batch_tokens: int = 128) -> Dict:
+        x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
+        t0 = time.perf_counter()
+        y = self.forward(x)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        flops = 2 * batch_tokens * 3 * self.hidden * self.inter * len(self.blocks)
+        return {
+            "tokens": batch_tokens,
+            "ms": dt * 1000,
+            "tflops": flops / dt / 1e12,
+            "checksum": float(y.float().abs().mean()),
+        }

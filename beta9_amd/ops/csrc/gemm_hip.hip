@@ -1,0 +1,315 @@
+// MFMA bf16 GEMM + fused elementwise ops for gfx950 (MI355X, CDNA4).
+//
+// Hand-written for the CDNA4 execution model (64-wide waves, MFMA matrix
+// cores, LDS staging via global_load_lds): C[M,N] = A[M,K] @ B^T where B is
+// stored transposed [N,K] (torch Linear weight layout), bf16 in / fp32
+// accumulate / bf16 out.
+//
+// Structure (cdna_hip_programming.md §5 ladder, step-3 class):
+//   - 128x128 output tile per 256-thread block (4 waves as 2x2, each wave a
+//     64x64 subtile = 4x4 fragments of v_mfma_f32_16x16x32_bf16)
+//   - BK=64 K-step, double-buffered LDS (A and B images both [128][64] bf16)
+//   - global -> LDS via __builtin_amdgcn_global_load_lds width 16 (lane-
+//     linear LDS image, per-lane global addresses)
+//   - XCD-aware bijective blockIdx swizzle (8 XCDs each with private 4 MiB
+//     L2: adjacent tiles land on one XCD for L2 reuse when HBM-bound)
+//
+// No library GEMM under this op: the fused serving path
+// (gemm -> silu*mul -> gemm) runs entirely on these kernels.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                    \
+                               hipGetErrorString(_e));                         \
+    }                                                                          \
+  } while (0)
+
+typedef __hip_bfloat16 bf16_t;
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef short bf16x8 __attribute__((ext_vector_type(8)));  // 8 bf16 in 4 VGPRs
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int WAVES = 4;  // 2x2, each wave 64x64
+constexpr int THREADS = WAVES * 64;
+
+// ---------------------------------------------------------------------------
+// GEMM kernel: A [M,K] row-major, Bt [N,K] row-major, C [M,N] row-major bf16.
+// M % 128 == 0, N % 128 == 0, K % 64 == 0 (host wrapper pads/asserts).
+// ---------------------------------------------------------------------------
+
+template <bool XCD_SWIZZLE>
+__global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ Bt,
+    bf16_t* __restrict__ C, int M, int N, int K) {
+  // One __shared__ object only (a second one forces vmcnt(0) before every
+  // ds_read of a glds pipeline — §5 trap 4a).
+  __shared__ bf16_t lds[2 * 2 * BM * BK];  // [buf][A|B][128][64]
+
+  int nwg_n = N / BN;
+  int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    // bijective XCD remap (guide: "XCD swizzle must be bijective")
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = (wg / nwg_n) * BM;
+  int tile_n = (wg % nwg_n) * BN;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 1, wc = wave & 1;  // wave's 64x64 subtile
+
+  // ---- glds staging ------------------------------------------------------
+  // Each image is 16 KiB = 16 pieces of 1 KiB (one wave-instruction each).
+  // Piece p covers rows p*8 .. p*8+7 (128 B per row); lane l handles
+  // row p*8 + l/8, 16-byte chunk l%8.
+  int ld_row = lane >> 3;          // 0..7 within piece
+  int ld_col = (lane & 7) * 8;     // bf16 column of 16B chunk
+  auto stage = [&](int buf, int kt) {
+    bf16_t* Ab = lds + buf * 2 * BM * BK;
+    bf16_t* Bb = Ab + BM * BK;
+    const bf16_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
+    const bf16_t* b_src = Bt + (int64_t)(tile_n)*K + kt * BK;
+#pragma unroll
+    for (int call = 0; call < 4; call++) {
+      int p = wave + WAVES * call;  // piece 0..15
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(a_src +
+                                                             (int64_t)row * K +
+                                                             ld_col),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(b_src +
+                                                             (int64_t)row * K +
+                                                             ld_col),
+          (__attribute__((address_space(3))) uint32_t*)(Bb + p * 512),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+
+  int ntiles = K / BK;
+  stage(0, 0);
+
+  // fragment read offsets: lane l -> row (l&15), k (l>>4)*8
+  int fr_row = lane & 15;
+  int fr_k = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < ntiles; kt++) {
+    int buf = kt & 1;
+    bf16_t* Ab = lds + buf * 2 * BM * BK;
+    bf16_t* Bb = Ab + BM * BK;
+    asm volatile("s_waitcnt vmcnt(0)");
+    __syncthreads();
+    if (kt + 1 < ntiles) stage(buf ^ 1, kt + 1);
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ks++) {  // two k=32 steps per BK=64 tile
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        int row = wr * 64 + i * 16 + fr_row;
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + ks * 32 + fr_k);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        int col = wc * 64 + j * 16 + fr_row;
+        b_frag[j] = *(const bf16x8*)(Bb + col * BK + ks * 32 + fr_k);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r ------------
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int col = tile_n + wc * 64 + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 64 + i * 16 + c_row0 + r;
+        C[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused SiLU-mul: out = silu(gate) * up, bf16, vectorized 8-wide.
+// The MLP seam (gate/up projections -> activation) fused into one
+// bandwidth-bound pass instead of three.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void silu_mul_kernel(
+    const bf16_t* __restrict__ gate, const bf16_t* __restrict__ up,
+    bf16_t* __restrict__ out, int64_t n) {
+  int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (; idx + 8 <= n; idx += stride) {
+    bf16x8 g = *(const bf16x8*)(gate + idx);
+    bf16x8 u = *(const bf16x8*)(up + idx);
+    bf16x8 o;
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      float gv = (float)((const bf16_t*)&g)[i];
+      float uv = (float)((const bf16_t*)&u)[i];
+      float s = gv / (1.0f + __expf(-gv));
+      ((bf16_t*)&o)[i] = (bf16_t)(s * uv);
+    }
+    *(bf16x8*)(out + idx) = o;
+  }
+  // tail (n not multiple of 8)
+  if (blockIdx.x == 0 && threadIdx.x < 8) {
+    int64_t tail_start = (n / 8) * 8;
+    int64_t t = tail_start + threadIdx.x;
+    if (t < n) {
+      float gv = (float)gate[t];
+      float uv = (float)up[t];
+      out[t] = (bf16_t)(gv / (1.0f + __expf(-gv)) * uv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm (bf16 x, bf16 weight, fp32 accum): one row per workgroup.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rmsnorm_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ out, int rows, int dim, float eps) {
+  int row = blockIdx.x;
+  if (row >= rows) return;
+  const bf16_t* xr = x + (int64_t)row * dim;
+  bf16_t* orow = out + (int64_t)row * dim;
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i + 8 <= dim; i += blockDim.x * 8) {
+    bf16x8 v = *(const bf16x8*)(xr + i);
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      float f = (float)((const bf16_t*)&v)[k];
+      ss += f * f;
+    }
+  }
+  // wave + block reduce
+  __shared__ float red[4];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off, 64);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = red[0] + red[1] + red[2] + red[3];
+    red[0] = rsqrtf(total / dim + eps);
+  }
+  __syncthreads();
+  float scale = red[0];
+  for (int i = threadIdx.x * 8; i + 8 <= dim; i += blockDim.x * 8) {
+    bf16x8 v = *(const bf16x8*)(xr + i);
+    bf16x8 wv = *(const bf16x8*)(w + i);
+    bf16x8 o;
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      float f = (float)((const bf16_t*)&v)[k];
+      float wf = (float)((const bf16_t*)&wv)[k];
+      ((bf16_t*)&o)[k] = (bf16_t)(f * scale * wf);
+    }
+    *(bf16x8*)(orow + i) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static void check_bf16_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.dtype() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 2 && t.is_contiguous(), name,
+              " must be 2D contiguous");
+}
+
+torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
+                           bool xcd_swizzle) {
+  check_bf16_2d(A, "A");
+  check_bf16_2d(Bt, "Bt");
+  int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
+              "shapes must be multiples of 128/128/64 (use gemm() wrapper "
+              "for padding)");
+  auto C = torch::empty({M, N}, A.options());
+  dim3 grid((M / BM) * (N / BN));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (xcd_swizzle) {
+    hipLaunchKernelGGL(gemm_bf16_nt_kernel<true>, grid, dim3(THREADS), 0,
+                       stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                       (int)M, (int)N, (int)K);
+  } else {
+    hipLaunchKernelGGL(gemm_bf16_nt_kernel<false>, grid, dim3(THREADS), 0,
+                       stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                       (int)M, (int)N, (int)K);
+  }
+  return C;
+}
+
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
+  TORCH_CHECK(gate.is_cuda() && gate.dtype() == torch::kBFloat16);
+  TORCH_CHECK(gate.sizes() == up.sizes());
+  TORCH_CHECK(gate.is_contiguous() && up.is_contiguous());
+  auto out = torch::empty_like(gate);
+  int64_t n = gate.numel();
+  int64_t blocks = std::min<int64_t>((n / 8 + 255) / 256, 4096);
+  if (blocks < 1) blocks = 1;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(silu_mul_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16_t*)gate.data_ptr(),
+                     (const bf16_t*)up.data_ptr(), (bf16_t*)out.data_ptr(), n);
+  return out;
+}
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
+  int rows = x.size(0), dim = x.size(1);
+  TORCH_CHECK(dim % 8 == 0, "dim must be multiple of 8");
+  TORCH_CHECK(w.numel() == dim);
+  auto out = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(rmsnorm_kernel, dim3(rows), dim3(256), 0, stream,
+                     (const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
+                     (bf16_t*)out.data_ptr(), rows, dim, (float)eps);
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "hand-written gfx950 MFMA kernels: bf16 GEMM, silu_mul, rmsnorm";
+  m.def("gemm_bf16_nt", &gemm_bf16_nt, py::arg("A"), py::arg("Bt"),
+        py::arg("xcd_swizzle") = true);
+  m.def("silu_mul", &silu_mul);
+  m.def("rmsnorm", &rmsnorm, py::arg("x"), py::arg("w"),
+        py::arg("eps") = 1e-6);
+}

@@ -8,7 +8,8 @@ optional hipsnap restore) BEFORE the handler import so weight streaming
 lands in reserved HBM.
 """
 
-from __future__ import annotations
+# NOTE: no `from __future__ import annotations` here — FastAPI resolves the
+# route handlers' `Request` annotations at def time from build_app's locals.
 
 import asyncio
 import logging
@@ -31,11 +32,33 @@ log = logging.getLogger("b9.runner.endpoint")
 
 
 def build_app(env: RunnerEnv, gpu: GPUBoot):
+    from contextlib import asynccontextmanager
+
     from fastapi import FastAPI, Request
     from fastapi.responses import JSONResponse
 
-    app = FastAPI()
     state: dict = {"ready": False, "on_start_value": None, "handler": None, "client": None}
+
+    async def startup() -> None:
+        client = RunnerClient(env)
+        state["client"] = client
+        state["on_start_value"] = await run_on_start(env)
+        state["handler"] = load_callable(env.handler)
+        state["ready"] = True
+        if env.checkpoint_enabled and not gpu.restored:
+            # on_start done -> signal checkpointability (reference:
+            # /criu/READY_FOR_CHECKPOINT, runner/common.py:618)
+            try:
+                await client.checkpoint_ready()
+            except Exception:
+                log.warning("checkpoint-ready signal failed", exc_info=True)
+
+    @asynccontextmanager
+    async def lifespan(app):
+        await startup()
+        yield
+
+    app = FastAPI(lifespan=lifespan)
 
     @app.get("/health")
     async def health():
@@ -71,21 +94,6 @@ def build_app(env: RunnerEnv, gpu: GPUBoot):
         except Exception as e:
             return JSONResponse({"error": str(e)}, status_code=500)
 
-    async def startup() -> None:
-        client = RunnerClient(env)
-        state["client"] = client
-        state["on_start_value"] = await run_on_start(env)
-        state["handler"] = load_callable(env.handler)
-        state["ready"] = True
-        if env.checkpoint_enabled and not gpu.restored:
-            # on_start done -> signal checkpointability (reference:
-            # /criu/READY_FOR_CHECKPOINT, runner/common.py:618)
-            try:
-                await client.checkpoint_ready()
-            except Exception:
-                log.warning("checkpoint-ready signal failed", exc_info=True)
-
-    app.add_event_handler("startup", startup)
     return app
 
 
