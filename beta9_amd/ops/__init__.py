@@ -1,0 +1,1 @@
+import torch  # load torch libs before the extensions

@@ -61,9 +61,14 @@ class Arena {
   int64_t base_ptr() const { return reinterpret_cast<int64_t>(base_); }
   int64_t size() const { return size_; }
 
-  // H2D: host buffer -> arena[offset], pinned double-buffered.
-  void upload(int64_t offset, const std::string& data) {
-    upload_raw(offset, data.data(), (int64_t)data.size());
+  // H2D: host buffer -> arena[offset], pinned double-buffered. Accepts any
+  // buffer-protocol object without copying; GIL released for the transfer.
+  void upload(int64_t offset, py::buffer data) {
+    py::buffer_info info = data.request();
+    const char* src = (const char*)info.ptr;
+    int64_t nbytes = (int64_t)(info.size * info.itemsize);
+    py::gil_scoped_release release;
+    upload_raw(offset, src, nbytes);
   }
 
   void upload_raw(int64_t offset, const char* src, int64_t nbytes) {
@@ -121,11 +126,21 @@ class Arena {
     return nbytes;
   }
 
-  // D2H into a bytes object.
+  // D2H into a bytes object (allocated once; GIL released for the copy).
   py::bytes download(int64_t offset, int64_t nbytes) {
     check_range(offset, nbytes);
     HIP_CHECK(hipSetDevice(device_));
-    std::string out((size_t)nbytes, '\0');
+    PyObject* obj = PyBytes_FromStringAndSize(nullptr, (Py_ssize_t)nbytes);
+    if (!obj) throw std::bad_alloc();
+    char* out = PyBytes_AS_STRING(obj);
+    {
+      py::gil_scoped_release release;
+      download_raw(offset, out, nbytes);
+    }
+    return py::reinterpret_steal<py::bytes>(obj);
+  }
+
+  void download_raw(int64_t offset, char* out, int64_t nbytes) {
     int64_t done = 0;
     int buf = 0;
     int64_t pending_off[2] = {-1, -1};
@@ -151,7 +166,6 @@ class Arena {
       }
     }
     d2h_bytes_ += nbytes;
-    return py::bytes(out);
   }
 
   // Stream an arena region to a file (snapshot path).
@@ -238,10 +252,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("base_ptr", &Arena::base_ptr)
       .def("size", &Arena::size)
       .def("upload", &Arena::upload)
-      .def("upload_file", &Arena::upload_file)
+      .def("upload_file", &Arena::upload_file,
+           py::call_guard<py::gil_scoped_release>())
       .def("download", &Arena::download)
       .def("download_file", &Arena::download_file, py::arg("offset"),
-           py::arg("path"), py::arg("nbytes"), py::arg("append") = false)
+           py::arg("path"), py::arg("nbytes"), py::arg("append") = false,
+           py::call_guard<py::gil_scoped_release>())
       .def("tensor_view", &Arena::tensor_view)
       .def("stats", &Arena::stats);
 }

@@ -1,0 +1,83 @@
+"""Prewarmed runner shell.
+
+The MI355X cold-start answer to the reference's preallocated network-slot
+pool (reference: pkg/worker/network.go:542-624 maintainNetworkSlotPool —
+cold start skips netns/veth creation): on a dedicated GPU node the expensive
+part of a cold start is not the namespace, it is the Python+torch import and
+the HIP context creation. The worker keeps N of these shells per GPU —
+spawned with the GPU pinned (HIP_VISIBLE_DEVICES fixed at spawn), torch
+imported, HIP context initialized — parked on a unix socket. Starting a
+container then costs only: send the spec, import the handler, stream
+weights.
+
+Protocol: one JSON line {env: {...}, module: "beta9_amd.runner.endpoint"}
+over the socket; the shell replies "ok\n", merges env, and execs the runner
+module in-process (it IS the container from then on).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import runpy
+import socket
+import sys
+
+
+def main() -> int:
+    sock_path = os.environ["B9_PREWARM_SOCK"]
+    # ---- warm the expensive imports -------------------------------------
+    import importlib
+
+    import torch  # noqa: F401  (the big one)
+
+    if os.environ.get("WORKER_GPU_DEVICES", "") != "":
+        try:
+            torch.cuda.init()
+            torch.zeros(1, device="cuda:0")  # force full HIP context + pools
+        except Exception:
+            pass
+    for mod in ("fastapi", "uvicorn", "grpc", "msgpack", "httpx"):
+        try:
+            importlib.import_module(mod)
+        except ImportError:
+            pass
+    import beta9_amd.runner.common  # noqa: F401
+    import beta9_amd.runner.endpoint  # noqa: F401
+    import beta9_amd.runner.function  # noqa: F401
+    import beta9_amd.runner.sandbox  # noqa: F401
+    import beta9_amd.runner.taskqueue  # noqa: F401
+
+    # ---- park on the socket ----------------------------------------------
+    try:
+        os.unlink(sock_path)
+    except FileNotFoundError:
+        pass
+    srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    srv.bind(sock_path)
+    srv.listen(1)
+    print(f"[prewarm] ready on {sock_path}", flush=True)
+    conn, _ = srv.accept()
+    data = b""
+    while not data.endswith(b"\n"):
+        chunk = conn.recv(65536)
+        if not chunk:
+            return 0
+        data += chunk
+    spec = json.loads(data.decode())
+    os.environ.update({str(k): str(v) for k, v in spec.get("env", {}).items()})
+    conn.sendall(b"ok\n")
+    conn.close()
+    srv.close()
+    try:
+        os.unlink(sock_path)
+    except FileNotFoundError:
+        pass
+    module = spec["module"]
+    sys.argv = [module]
+    runpy.run_module(module, run_name="__main__")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

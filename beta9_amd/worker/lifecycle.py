@@ -73,6 +73,7 @@ class Lifecycle:
         gateway_host: str,
         gateway_grpc_port: int,
         worker_host: str = "127.0.0.1",
+        prewarm=None,  # PrewarmPool
     ):
         self.worker_id = worker_id
         self.repo = repo
@@ -81,6 +82,7 @@ class Lifecycle:
         self.gateway_host = gateway_host
         self.gateway_grpc_port = gateway_grpc_port
         self.worker_host = worker_host
+        self.prewarm = prewarm
         self.containers: Dict[str, ContainerInstance] = {}
 
     async def run_container(self, request: ContainerRequest) -> int:
@@ -118,7 +120,27 @@ class Lifecycle:
 
             with rec.span("worker.runtime_start"):
                 on_log = self._make_log_cb(cid)
-                handle = await self.runtime.run(spec, on_log=on_log)
+                handle = None
+                # Prewarmed shell path: torch+HIP context already up on the
+                # assigned GPU; the shell becomes the container.
+                if self.prewarm is not None and hasattr(self.runtime, "adopt"):
+                    from .prewarm import runner_module_from_entry
+
+                    module = runner_module_from_entry(spec.entry_point)
+                    ordinals = gpu_assignment.ordinals if gpu_assignment else []
+                    if module is not None and len(ordinals) <= 1:
+                        shell = await self.prewarm.acquire(ordinals)
+                        if shell is not None:
+                            if await shell.activate(spec.env, module):
+                                handle = await self.runtime.adopt(spec, shell.proc, on_log)
+                                rec.record("worker.prewarm_hit", time.time(), time.time())
+                            else:
+                                try:
+                                    shell.proc.kill()
+                                except ProcessLookupError:
+                                    pass
+                if handle is None:
+                    handle = await self.runtime.run(spec, on_log=on_log)
                 inst.started_at = time.time()
 
             await self.repo.update_container_status(cid, ContainerStatus.running.value)

@@ -1,0 +1,158 @@
+"""Worker-side prewarm pool: keeps runner shells (torch+HIP pre-initialized)
+per GPU ordinal and hands them to the lifecycle on container start.
+
+See beta9_amd/runner/prewarm.py for the shell side. Pool shape: one shell
+per GPU ordinal (the GPU is pinned at spawn — HIP visibility cannot be
+changed after context creation) plus a small CPU pool. A shell consumed by a
+container start is replenished in the background.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import os
+import sys
+import tempfile
+import time
+from typing import Dict, List, Optional, Tuple
+
+log = logging.getLogger(__name__)
+
+_RUNNER_MODULES = {
+    "beta9_amd.runner.endpoint",
+    "beta9_amd.runner.function",
+    "beta9_amd.runner.taskqueue",
+    "beta9_amd.runner.sandbox",
+    "beta9_amd.runner.pod",
+}
+
+
+class PrewarmedShell:
+    def __init__(self, proc: asyncio.subprocess.Process, sock_path: str, key: Tuple[int, ...]):
+        self.proc = proc
+        self.sock_path = sock_path
+        self.key = key
+        self.ready = False
+
+    async def wait_ready(self, timeout: float = 120.0) -> bool:
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if self.proc.returncode is not None:
+                return False
+            if os.path.exists(self.sock_path):
+                # socket exists once the shell is parked
+                self.ready = True
+                return True
+            await asyncio.sleep(0.02)
+        return False
+
+    async def activate(self, env: Dict[str, str], module: str) -> bool:
+        """Send the spec; on success the shell becomes the container."""
+        try:
+            reader, writer = await asyncio.open_unix_connection(self.sock_path)
+            writer.write((json.dumps({"env": env, "module": module}) + "\n").encode())
+            await writer.drain()
+            resp = await asyncio.wait_for(reader.readline(), timeout=10.0)
+            writer.close()
+            return resp.strip() == b"ok"
+        except (OSError, asyncio.TimeoutError):
+            return False
+
+
+class PrewarmPool:
+    def __init__(
+        self,
+        gpu_ordinals: List[int],
+        cpu_shells: int = 2,
+        base_env: Optional[Dict[str, str]] = None,
+        visible_env_names: Optional[List[str]] = None,
+    ):
+        self.gpu_ordinals = gpu_ordinals
+        self.cpu_shells = cpu_shells
+        self.base_env = base_env or {}
+        self.visible_env_names = visible_env_names or [
+            "HIP_VISIBLE_DEVICES",
+            "ROCR_VISIBLE_DEVICES",
+        ]
+        self._shells: Dict[Tuple[int, ...], List[PrewarmedShell]] = {}
+        self._dir = tempfile.mkdtemp(prefix="b9-prewarm-")
+        self._replenish_lock = asyncio.Lock()
+        self._n = 0
+        self._stopping = False
+
+    async def start(self) -> None:
+        tasks = [self._spawn((o,)) for o in self.gpu_ordinals]
+        tasks += [self._spawn(()) for _ in range(self.cpu_shells)]
+        await asyncio.gather(*tasks)
+
+    async def _spawn(self, key: Tuple[int, ...]) -> None:
+        if self._stopping:
+            return
+        self._n += 1
+        sock_path = os.path.join(self._dir, f"shell-{self._n}.sock")
+        env = dict(os.environ)
+        env.update(self.base_env)
+        env["B9_PREWARM_SOCK"] = sock_path
+        if key:
+            vis = ",".join(str(o) for o in key)
+            for name in self.visible_env_names:
+                env[name] = vis
+            env["WORKER_GPU_DEVICES"] = vis
+        else:
+            env.pop("WORKER_GPU_DEVICES", None)
+        proc = await asyncio.create_subprocess_exec(
+            sys.executable,
+            "-u",
+            "-m",
+            "beta9_amd.runner.prewarm",
+            env=env,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            start_new_session=True,
+        )
+        shell = PrewarmedShell(proc, sock_path, key)
+        self._shells.setdefault(key, []).append(shell)
+
+    async def acquire(self, ordinals: List[int]) -> Optional[PrewarmedShell]:
+        """Take a ready shell matching the GPU set (or CPU for [])."""
+        key = tuple(ordinals)
+        shells = self._shells.get(key, [])
+        for shell in list(shells):
+            if shell.proc.returncode is not None:
+                shells.remove(shell)
+                continue
+            if not shell.ready and not os.path.exists(shell.sock_path):
+                continue
+            shells.remove(shell)
+            asyncio.create_task(self._spawn(key))  # replenish in background
+            return shell
+        return None
+
+    async def shutdown(self) -> None:
+        self._stopping = True
+        for shells in self._shells.values():
+            for shell in shells:
+                if shell.proc.returncode is None:
+                    try:
+                        shell.proc.terminate()
+                    except ProcessLookupError:
+                        pass
+        for shells in self._shells.values():
+            for shell in shells:
+                try:
+                    await asyncio.wait_for(shell.proc.wait(), timeout=3.0)
+                except asyncio.TimeoutError:
+                    try:
+                        shell.proc.kill()
+                    except ProcessLookupError:
+                        pass
+
+
+def runner_module_from_entry(entry_point: List[str]) -> Optional[str]:
+    """Extract the runner module from a standard entrypoint, or None if the
+    entrypoint is custom (no prewarm for those)."""
+    if len(entry_point) >= 3 and entry_point[-2] == "-m" and entry_point[-1] in _RUNNER_MODULES:
+        return entry_point[-1]
+    return None

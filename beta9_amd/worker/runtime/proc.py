@@ -104,6 +104,32 @@ class ProcRuntime(Runtime):
             asyncio.create_task(self._drain(proc.stderr))
         return ContainerHandle(container_id=spec.container_id, pid=proc.pid)
 
+    async def adopt(
+        self,
+        spec: ContainerSpec,
+        proc: asyncio.subprocess.Process,
+        on_log: Optional[LogCallback] = None,
+    ) -> ContainerHandle:
+        """Adopt an already-running process (a prewarmed runner shell) as a
+        container: move it into the resource cgroup and take over its
+        lifecycle."""
+        cg = self._setup_cgroup(spec)
+        entry = _Proc(proc, cg)
+        self._containers[spec.container_id] = entry
+        if cg:
+            try:
+                with open(os.path.join(cg, "cgroup.procs"), "w") as f:
+                    f.write(str(proc.pid))
+            except OSError:
+                pass
+        if on_log is not None and proc.stdout is not None:
+            asyncio.create_task(self._pump(proc.stdout, "stdout", on_log))
+            asyncio.create_task(self._pump(proc.stderr, "stderr", on_log))
+        elif proc.stdout is not None:
+            asyncio.create_task(self._drain(proc.stdout))
+            asyncio.create_task(self._drain(proc.stderr))
+        return ContainerHandle(container_id=spec.container_id, pid=proc.pid)
+
     @staticmethod
     async def _pump(stream: asyncio.StreamReader, name: str, cb: LogCallback) -> None:
         while True:

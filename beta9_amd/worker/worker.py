@@ -41,6 +41,8 @@ class Worker:
         keepalive_interval_s: float = 2.0,
         idle_shutdown_s: float = 0.0,
         machine_id: str = "",
+        prewarm: bool = True,
+        prewarm_cpu_shells: int = 2,
     ):
         self.id = worker_id
         self.repo = repo
@@ -65,6 +67,14 @@ class Worker:
             pool_name=pool_name,
             machine_id=machine_id,
         )
+        self.prewarm_pool = None
+        if prewarm:
+            from .prewarm import PrewarmPool
+
+            ordinals = [d.ordinal for d in gpu_manager.devices] if gpu_manager else []
+            self.prewarm_pool = PrewarmPool(
+                gpu_ordinals=ordinals, cpu_shells=prewarm_cpu_shells
+            )
         self.lifecycle = Lifecycle(
             worker_id=worker_id,
             repo=repo,
@@ -72,6 +82,7 @@ class Worker:
             gpu_manager=gpu_manager,
             gateway_host=gateway_host,
             gateway_grpc_port=gateway_grpc_port,
+            prewarm=self.prewarm_pool,
         )
         self._sema = asyncio.Semaphore(container_start_limit)
         self._keepalive_interval_s = keepalive_interval_s
@@ -83,9 +94,27 @@ class Worker:
 
     async def start(self) -> None:
         await self.repo.register_worker(self.state)
+        if self.prewarm_pool is not None:
+            await self.prewarm_pool.start()
         self._run_task = asyncio.create_task(self._run())
         ka = asyncio.create_task(self._keepalive_loop())
         self._tasks.add(ka)
+
+    async def wait_prewarmed(self, timeout: float = 120.0) -> bool:
+        """Block until every prewarm shell is parked (bench setup)."""
+        if self.prewarm_pool is None:
+            return True
+        import time as _time
+
+        deadline = _time.monotonic() + timeout
+        while _time.monotonic() < deadline:
+            shells = [s for lst in self.prewarm_pool._shells.values() for s in lst]
+            if shells and all(
+                s.proc.returncode is None and os.path.exists(s.sock_path) for s in shells
+            ):
+                return True
+            await asyncio.sleep(0.05)
+        return False
 
     async def _run(self) -> None:
         import time as _time
@@ -150,6 +179,8 @@ class Worker:
 
     async def shutdown(self) -> None:
         self._stopping.set()
+        if self.prewarm_pool is not None:
+            await self.prewarm_pool.shutdown()
         for cid in list(self.lifecycle.containers):
             await self.runtime.kill(cid, 15)
         await self.drain(timeout=10.0)
