@@ -90,6 +90,7 @@ class Gateway:
             "KEEP_WARM_SECONDS": str(cfg.keep_warm_seconds),
             "CONCURRENT_REQUESTS": str(cfg.concurrent_requests),
             "CHECKPOINT_ENABLED": "1" if cfg.checkpoint_enabled else "0",
+            "B9_CHECKPOINT_DIR": self.cfg.checkpoint.store_path,
             "CALLBACK_URL": cfg.callback_url,
             "PYTHONUNBUFFERED": "1",
         }
@@ -155,7 +156,17 @@ class Gateway:
             self.scheduler,
             objects_path=os.path.join(self.cfg.storage.root, "objects"),
         )
-        runner_svc = RunnerService(self.dispatcher, self.store, self.db)
+        os.makedirs(self.cfg.checkpoint.store_path, exist_ok=True)
+        runner_svc = RunnerService(
+            self.dispatcher, self.store, self.db,
+            checkpoint_dir=self.cfg.checkpoint.store_path if self.cfg.checkpoint.enabled else "",
+        )
+        # scheduler attaches the latest available checkpoint to
+        # checkpoint-enabled requests (reference: prepareWorkerRequest +
+        # scheduler/checkpoint.go attachLatestCheckpoint)
+        self.scheduler.checkpoint_lookup = (
+            lambda ws, stub_id: self.db.latest_checkpoint(ws, stub_id)
+        )
         repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo)
 
         fn_exec = FunctionExecutor(self.db, self.scheduler, self.store, self.runner_env)

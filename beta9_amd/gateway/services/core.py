@@ -205,10 +205,11 @@ class RunnerService:
 
     service_name = "RunnerService"
 
-    def __init__(self, dispatcher: Dispatcher, store: StateStore, db: BackendDB):
+    def __init__(self, dispatcher: Dispatcher, store: StateStore, db: BackendDB, checkpoint_dir: str = ""):
         self.dispatcher = dispatcher
         self.store = store
         self.db = db
+        self.checkpoint_dir = checkpoint_dir
 
     @rpc()
     async def StartTask(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
@@ -245,11 +246,47 @@ class RunnerService:
         return True
 
     @rpc()
-    async def CheckpointReady(self, req: Dict[str, Any], context, auth) -> bool:
-        """Runner signals on_start finished and a hipsnap checkpoint may be
-        taken (reference: /criu/READY_FOR_CHECKPOINT sentinel —
-        sdk runner/common.py:618, pkg/worker/criu.go:2576)."""
+    async def CheckpointReady(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Runner signals on_start finished; if the stub has no checkpoint
+        yet, THIS container is granted the (single-flight, stub-scoped)
+        hipsnap snapshot (reference: READY sentinel + stub-level Redis lock —
+        runner/common.py:618, pkg/worker/criu.go:475-510)."""
+        stub_id = req["stub_id"]
+        ws = auth["workspace_id"]
         await self.store.set(
-            f"checkpoint:ready:{req['stub_id']}:{req['container_id']}", time.time()
+            f"checkpoint:ready:{stub_id}:{req['container_id']}", time.time()
         )
+        if not self.checkpoint_dir:
+            return {"take": False}
+        if self.db.latest_checkpoint(ws, stub_id) is not None:
+            return {"take": False}
+        lock_key = f"checkpoint:taking:{stub_id}"
+        if not await self.store.set_nx(lock_key, req["container_id"], ttl=600.0):
+            return {"take": False}
+        key = f"{ws}-{stub_id}"
+        return {"take": True, "key": key, "dir": self.checkpoint_dir}
+
+    @rpc()
+    async def CommitCheckpoint(self, req: Dict[str, Any], context, auth) -> bool:
+        """Record a completed hipsnap snapshot (scheduler attaches it to
+        future requests for scale-to-zero resume)."""
+        stub_id = req["stub_id"]
+        ws = auth["workspace_id"]
+        stub = self.db.get_stub(stub_id)
+        self.db.upsert_checkpoint(
+            {
+                "id": req["key"],
+                "workspace_id": ws,
+                "stub_id": stub_id,
+                "source_container_id": req.get("container_id", ""),
+                "status": "available",
+                "remote_key": req["key"],
+                "stub_type": stub["type"] if stub else "",
+                "accelerator": req.get("accelerator", "MI355X"),
+                "gpu_count": int(req.get("gpu_count", 1)),
+                "size_bytes": int(req.get("size_bytes", 0)),
+                "created_at": time.time(),
+            }
+        )
+        await self.store.delete(f"checkpoint:taking:{stub_id}")
         return True

@@ -28,11 +28,24 @@ def load() -> Dict[str, Any]:
     inter = int(os.environ.get("B9_BENCH_INTER", "14336"))
     layers = int(os.environ.get("B9_BENCH_LAYERS", "8"))
     if torch.cuda.is_available():
+        from ..runner.common import get_gpu_boot
         from .mlp import SyntheticServingModel
 
-        _model = SyntheticServingModel(hidden=hidden, inter=inter, layers=layers, device="cuda:0")
+        gb = get_gpu_boot()
+        restored = False
+        if gb is not None and gb.restored and gb.arena is not None:
+            # scale-to-zero resume: weights already streamed into HBM from
+            # the hipsnap snapshot during GPU boot — just remap views.
+            _model = SyntheticServingModel.from_arena(gb.arena, hidden, inter, layers)
+            restored = True
+        else:
+            _model = SyntheticServingModel(
+                hidden=hidden, inter=inter, layers=layers, device="cuda:0",
+                arena=(gb.arena if gb is not None else None),
+            )
         native = True
         param_bytes = _model.param_bytes
+        _load_info["restored"] = restored
     else:
         # CPU dev-path: tiny torch model so the full dispatch loop still runs
         _model = _CPUModel(hidden=min(hidden, 256), layers=2)

@@ -61,12 +61,18 @@ class SyntheticServingModel:
         layers: int = 4,
         device: str = "cuda:0",
         use_arena: bool = True,
+        arena=None,
     ):
         self.hidden = hidden
         self.inter = inter
         self.device = device
-        self.arena = None
-        if use_arena and torch.cuda.is_available():
+        need = layers * (2 * inter * hidden + hidden * inter + hidden) * 2
+        if arena is not None and getattr(arena, "native", False):
+            free = arena.size_bytes - arena._offset
+            self.arena = arena if free >= int(need * 1.1) else None
+        else:
+            self.arena = None
+        if self.arena is None and use_arena and torch.cuda.is_available():
             try:
                 from ..ops import hipsnap
 
@@ -83,6 +89,32 @@ class SyntheticServingModel:
             MLPBlock(hidden, inter, device, arena=self.arena, prefix=f"l{i}.")
             for i in range(layers)
         ]
+
+    @classmethod
+    def from_arena(cls, arena, hidden: int, inter: int, layers: int, device: str = "cuda:0"):
+        """Rebuild the model as views over a RESTORED hipsnap arena — the
+        scale-to-zero resume path: no weight init, no weight load; the HBM
+        snapshot already streamed in during GPU boot."""
+        import torch
+
+        self = cls.__new__(cls)
+        self.hidden, self.inter, self.device, self.arena = hidden, inter, device, arena
+        regions = {n: (o, s) for (n, o, s) in arena.regions()}
+
+        def view(name, shape):
+            off, size = regions[name]
+            return arena._impl.tensor_view(off, size).view(torch.bfloat16).reshape(shape)
+
+        self.blocks = []
+        for i in range(layers):
+            b = MLPBlock.__new__(MLPBlock)
+            b.hidden, b.inter = hidden, inter
+            b.w_gate = view(f"l{i}.w_gate", (inter, hidden))
+            b.w_up = view(f"l{i}.w_up", (inter, hidden))
+            b.w_down = view(f"l{i}.w_down", (hidden, inter))
+            b.norm_w = view(f"l{i}.norm_w", (hidden,))
+            self.blocks.append(b)
+        return self
 
     @property
     def param_bytes(self) -> int:

@@ -143,6 +143,18 @@ def snapshot_path(store_dir: str, key: str) -> str:
     return os.path.join(store_dir, f"{key}.hipsnap")
 
 
+def peek_meta(path: str) -> Optional[Dict]:
+    """Read just the snapshot header (region list) without the payload."""
+    try:
+        with open(path, "rb") as f:
+            hlen = int.from_bytes(f.read(8), "little")
+            if hlen <= 0 or hlen > (1 << 24):
+                return None
+            return json.loads(f.read(hlen).decode())
+    except (OSError, ValueError):
+        return None
+
+
 def snapshot(arena: Arena, store_dir: str, key: str) -> Dict:
     """Dump all registered arena regions to a snapshot file (D2H through
     pinned staging)."""
@@ -162,10 +174,16 @@ def snapshot(arena: Arena, store_dir: str, key: str) -> Dict:
         header = json.dumps(meta).encode()
         f.write(len(header).to_bytes(8, "little"))
         f.write(header)
+    if arena._impl is not None:
+        # native path: HBM -> pinned ring -> file, appended per region
         for name, off, nbytes in arena.regions():
-            data = arena.download(off, nbytes)
-            f.write(data)
+            arena._impl.download_file(off, path + ".tmp", nbytes, True)
             total += nbytes
+    else:
+        with open(path + ".tmp", "ab") as f:
+            for name, off, nbytes in arena.regions():
+                f.write(arena.download(off, nbytes))
+                total += nbytes
     os.rename(path + ".tmp", path)
     dt = time.perf_counter() - t0
     return {"path": path, "bytes": total, "seconds": dt, "gbps": total / max(dt, 1e-9) / 1e9}
@@ -181,13 +199,21 @@ def restore_snapshot(arena: Arena, key_or_path: str, store_dir: str = "") -> boo
     with open(path, "rb") as f:
         hlen = int.from_bytes(f.read(8), "little")
         meta = json.loads(f.read(hlen).decode())
-        arena._regions = []
-        arena._offset = 0
-        for region in meta["regions"]:
-            off = arena.alloc(region["name"], region["nbytes"])
-            assert off == region["offset"], "arena layout mismatch on restore"
-            data = f.read(region["nbytes"])
-            arena.upload(off, data)
+        payload_start = 8 + hlen
+    arena._regions = []
+    arena._offset = 0
+    file_off = payload_start
+    for region in meta["regions"]:
+        off = arena.alloc(region["name"], region["nbytes"])
+        assert off == region["offset"], "arena layout mismatch on restore"
+        if arena._impl is not None:
+            # file -> pinned ring -> HBM (the cold-start restore hot path)
+            arena._impl.upload_file(off, path, file_off, region["nbytes"])
+        else:
+            with open(path, "rb") as f:
+                f.seek(file_off)
+                arena.upload(off, f.read(region["nbytes"]))
+        file_off += region["nbytes"]
     return True
 
 

@@ -99,6 +99,15 @@ def setup_environment(env: RunnerEnv) -> None:
         sys.path.insert(0, os.getcwd())
 
 
+_gpu_boot: Optional["GPUBoot"] = None
+
+
+def get_gpu_boot() -> Optional["GPUBoot"]:
+    """The container's GPU boot context (arena + restore state); handlers use
+    this to rebuild models from a restored HBM snapshot."""
+    return _gpu_boot
+
+
 class GPUBoot:
     """HIP pre-init + HBM arena + snapshot restore for GPU containers.
 
@@ -115,6 +124,8 @@ class GPUBoot:
         self.phases: Dict[str, float] = {}
 
     def boot(self) -> None:
+        global _gpu_boot
+        _gpu_boot = self
         if not self.env.gpu_devices:
             return
         t0 = time.perf_counter()
@@ -129,11 +140,20 @@ class GPUBoot:
             t1 = time.perf_counter()
             from ..ops import hipsnap
 
-            self.arena = hipsnap.Arena(device=0)
+            ckpt_dir = os.environ.get("B9_CHECKPOINT_DIR", "")
+            size = 0
+            snap_path = ""
+            if self.env.checkpoint_key and ckpt_dir:
+                snap_path = hipsnap.snapshot_path(ckpt_dir, self.env.checkpoint_key)
+                meta = hipsnap.peek_meta(snap_path)
+                if meta is not None:
+                    total = sum(r["nbytes"] for r in meta["regions"])
+                    size = int(total * 1.25) + (64 << 20)
+            self.arena = hipsnap.Arena(device=0, size_bytes=size)
             self.phases["hbm_prealloc_ms"] = (time.perf_counter() - t1) * 1000
-            if self.env.checkpoint_key:
+            if snap_path:
                 t2 = time.perf_counter()
-                self.restored = hipsnap.restore_snapshot(self.arena, self.env.checkpoint_key)
+                self.restored = hipsnap.restore_snapshot(self.arena, snap_path)
                 self.phases["hbm_restore_ms"] = (time.perf_counter() - t2) * 1000
         except Exception:
             log.exception("GPU boot failed; continuing CPU-only")
@@ -230,12 +250,50 @@ class RunnerClient:
             },
         )
 
-    async def checkpoint_ready(self) -> None:
-        await self.rpc.call(
+    async def checkpoint_ready(self) -> Dict[str, Any]:
+        """Signal on_start done; gateway may grant this container the
+        checkpoint (single-flight per stub)."""
+        return await self.rpc.call(
             "RunnerService",
             "CheckpointReady",
             {"stub_id": self.env.stub_id, "container_id": self.env.container_id},
         )
+
+    async def commit_checkpoint(self, key: str, size_bytes: int, gpu_count: int = 1) -> None:
+        await self.rpc.call(
+            "RunnerService",
+            "CommitCheckpoint",
+            {
+                "stub_id": self.env.stub_id,
+                "container_id": self.env.container_id,
+                "key": key,
+                "size_bytes": size_bytes,
+                "gpu_count": gpu_count,
+            },
+        )
+
+    async def take_checkpoint_if_granted(self, gpu_boot: "GPUBoot") -> bool:
+        """The full runner-side checkpoint flow: ask, snapshot the arena,
+        commit. Returns True if a checkpoint was written."""
+        try:
+            grant = await self.checkpoint_ready()
+        except Exception:
+            log.warning("checkpoint-ready signal failed", exc_info=True)
+            return False
+        if not isinstance(grant, dict) or not grant.get("take"):
+            return False
+        arena = gpu_boot.arena if gpu_boot else None
+        if arena is None or not getattr(arena, "native", False):
+            return False
+        from ..ops import hipsnap
+
+        info = hipsnap.snapshot(arena, grant["dir"], grant["key"])
+        await self.commit_checkpoint(grant["key"], info["bytes"])
+        log.info(
+            "checkpoint %s written: %.1f GB at %.1f GB/s",
+            grant["key"], info["bytes"] / 1e9, info["gbps"],
+        )
+        return True
 
     def start_heartbeat(self, task_id: str, interval: float = 5.0) -> None:
         async def loop() -> None:

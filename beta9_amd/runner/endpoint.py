@@ -46,12 +46,12 @@ def build_app(env: RunnerEnv, gpu: GPUBoot):
         state["handler"] = load_callable(env.handler)
         state["ready"] = True
         if env.checkpoint_enabled and not gpu.restored:
-            # on_start done -> signal checkpointability (reference:
-            # /criu/READY_FOR_CHECKPOINT, runner/common.py:618)
+            # on_start done -> take the stub checkpoint if granted
+            # (reference: READY sentinel -> worker startAutoCheckpoint)
             try:
-                await client.checkpoint_ready()
+                await client.take_checkpoint_if_granted(gpu)
             except Exception:
-                log.warning("checkpoint-ready signal failed", exc_info=True)
+                log.warning("checkpoint flow failed", exc_info=True)
 
     @asynccontextmanager
     async def lifespan(app):

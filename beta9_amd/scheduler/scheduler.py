@@ -79,6 +79,8 @@ class Scheduler:
         self.containers = container_repo
         self.pools: Dict[str, PoolController] = {}
         self.metrics = SchedulerMetrics()
+        # set by the gateway: (workspace_id, stub_id) -> checkpoint row dict
+        self.checkpoint_lookup = None
         self._loop_task: Optional[asyncio.Task] = None
         self._wake = asyncio.Event()
 
@@ -89,6 +91,21 @@ class Scheduler:
     async def run(self, request: ContainerRequest) -> None:
         """Accept a container request: normalize, record state, enqueue."""
         self._normalize_gpu(request)
+        if request.checkpoint_enabled and self.checkpoint_lookup is not None:
+            cp = self.checkpoint_lookup(request.workspace_id, request.stub_id)
+            if cp is not None and (not cp.get("accelerator") or not request.gpu or cp["accelerator"] == request.gpu):
+                from ..types import CheckpointState, CheckpointStatus
+
+                request.checkpoint = CheckpointState(
+                    checkpoint_id=cp["id"],
+                    stub_id=cp["stub_id"],
+                    workspace_id=cp["workspace_id"],
+                    status=CheckpointStatus.available,
+                    remote_key=cp["remote_key"],
+                    accelerator=cp.get("accelerator", ""),
+                    gpu_count=cp.get("gpu_count", 0),
+                    size_bytes=cp.get("size_bytes", 0),
+                )
         st = ContainerState(
             container_id=request.container_id,
             stub_id=request.stub_id,
