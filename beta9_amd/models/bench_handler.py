@@ -45,16 +45,20 @@ def load() -> Dict[str, Any]:
             )
         native = True
         param_bytes = _model.param_bytes
-        _load_info["restored"] = restored
     else:
         # CPU dev-path: tiny torch model so the full dispatch loop still runs
         _model = _CPUModel(hidden=min(hidden, 256), layers=2)
         native = False
         param_bytes = _model.param_bytes
+    from ..runner.common import get_gpu_boot as _ggb
+
+    gb = _ggb()
     _load_info = {
         "load_ms": (time.perf_counter() - t0) * 1000,
         "native": native,
         "param_bytes": param_bytes,
+        "restored": bool(gb is not None and gb.restored),
+        "gpu_phases": dict(gb.phases) if gb is not None else {},
     }
     return _load_info
 
