@@ -27,6 +27,31 @@ def load() -> Dict[str, Any]:
     hidden = int(os.environ.get("B9_BENCH_HIDDEN", "4096"))
     inter = int(os.environ.get("B9_BENCH_INTER", "14336"))
     layers = int(os.environ.get("B9_BENCH_LAYERS", "8"))
+    world = int(os.environ.get("WORLD_SIZE", "1") or 1)
+    if world > 1:
+        # tensor-parallel over the container's RCCL rendezvous (mgpu runner)
+        from ..runner.common import get_gpu_boot
+        from ..runner.mgpu import get_rendezvous
+        from .mlp_tp import TPServingModel
+
+        rdzv = get_rendezvous()
+        assert rdzv is not None, "WORLD_SIZE>1 but no rendezvous initialized"
+        gb = get_gpu_boot()
+        device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        _model = TPServingModel(
+            hidden=hidden, inter=inter, layers=layers,
+            rank=rdzv.rank, world=rdzv.world, device=device,
+            arena=(gb.arena if gb is not None else None),
+        )
+        _load_info = {
+            "load_ms": (time.perf_counter() - t0) * 1000,
+            "native": torch.cuda.is_available(),
+            "param_bytes": _model.param_bytes,
+            "restored": False,
+            "world": world,
+            "backend": rdzv.backend,
+        }
+        return _load_info
     if torch.cuda.is_available():
         from ..runner.common import get_gpu_boot
         from .mlp import SyntheticServingModel

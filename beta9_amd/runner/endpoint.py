@@ -31,13 +31,24 @@ from .common import (
 log = logging.getLogger("b9.runner.endpoint")
 
 
-def build_app(env: RunnerEnv, gpu: GPUBoot):
+def build_app(env: RunnerEnv, gpu: GPUBoot, rdzv=None):
     from contextlib import asynccontextmanager
 
     from fastapi import FastAPI, Request
     from fastapi.responses import JSONResponse
 
-    state: dict = {"ready": False, "on_start_value": None, "handler": None, "client": None}
+    state: dict = {"ready": False, "on_start_value": None, "handler": None, "client": None, "dispatcher": None}
+    if rdzv is not None:
+        from .mgpu import Rank0Dispatcher
+
+        state["dispatcher"] = Rank0Dispatcher(rdzv)
+
+    async def run_handler(args):
+        # multi-GPU: broadcast args so every rank executes the same handler
+        # call (RCCL collectives inside the handler line up across ranks)
+        if state["dispatcher"] is not None:
+            return await state["dispatcher"].dispatch(state["handler"], args)
+        return await call_handler(state["handler"], args)
 
     async def startup() -> None:
         client = RunnerClient(env)
@@ -73,9 +84,7 @@ def build_app(env: RunnerEnv, gpu: GPUBoot):
         except Exception:
             body = {}
         try:
-            result = await asyncio.wait_for(
-                call_handler(state["handler"], body), timeout=env.timeout
-            )
+            result = await asyncio.wait_for(run_handler(body), timeout=env.timeout)
             return JSONResponse({"result": result})
         except asyncio.TimeoutError:
             return JSONResponse({"error": "timeout"}, status_code=504)
@@ -87,8 +96,7 @@ def build_app(env: RunnerEnv, gpu: GPUBoot):
     async def invoke_get(request: Request, path: str = ""):
         try:
             result = await asyncio.wait_for(
-                call_handler(state["handler"], dict(request.query_params)),
-                timeout=env.timeout,
+                run_handler(dict(request.query_params)), timeout=env.timeout
             )
             return JSONResponse({"result": result})
         except Exception as e:
@@ -99,6 +107,13 @@ def build_app(env: RunnerEnv, gpu: GPUBoot):
 
 def main() -> int:
     logging.basicConfig(level=logging.INFO)
+    world = int(os.environ.get("WORLD_SIZE", "1") or 1)
+    if world > 1:
+        # gpu_count>1 container: RCCL-over-xGMI rendezvous, one process per
+        # GPU (mgpu.py) — must run BEFORE any HIP context is created here.
+        from .mgpu import run_multi_gpu
+
+        return run_multi_gpu(world)
     env = RunnerEnv.from_env()
     setup_environment(env)
     t0 = time.perf_counter()
