@@ -1,0 +1,306 @@
+"""b9 CLI — operator and developer surface.
+
+(reference: sdk/src/beta9/cli/main.py click multi-group: config/container/
+deploy/deployment/run/secret/serve/task/token/volume/worker/pool). Run as
+`python -m beta9_amd.cli` or the `b9` console script.
+"""
+
+from __future__ import annotations
+
+import importlib
+import json
+import os
+import sys
+from typing import List, Optional
+
+import typer
+import yaml
+
+app = typer.Typer(help="beta9_amd — MI355X-native serverless GPU runtime")
+config_app = typer.Typer(help="client contexts")
+task_app = typer.Typer(help="tasks")
+container_app = typer.Typer(help="containers")
+deployment_app = typer.Typer(help="deployments")
+secret_app = typer.Typer(help="secrets")
+volume_app = typer.Typer(help="volumes")
+worker_app = typer.Typer(help="workers")
+app.add_typer(config_app, name="config")
+app.add_typer(task_app, name="task")
+app.add_typer(container_app, name="container")
+app.add_typer(deployment_app, name="deployment")
+app.add_typer(secret_app, name="secret")
+app.add_typer(volume_app, name="volume")
+app.add_typer(worker_app, name="worker")
+
+CONFIG_PATH = os.path.expanduser("~/.b9/config.yaml")
+
+
+def _load_context() -> dict:
+    if os.path.exists(CONFIG_PATH):
+        with open(CONFIG_PATH) as f:
+            return yaml.safe_load(f) or {}
+    return {}
+
+
+def _client():
+    import beta9_amd
+
+    ctx = _load_context()
+    if ctx.get("gateway_addr"):
+        beta9_amd.configure(ctx["gateway_addr"], ctx.get("token", ""))
+    return beta9_amd.Client()
+
+
+def _load_handler(spec: str):
+    """Load module.py:fn or module:fn from cwd."""
+    mod_name, _, fn_name = spec.partition(":")
+    if mod_name.endswith(".py"):
+        mod_name = mod_name[:-3].replace("/", ".")
+    sys.path.insert(0, os.getcwd())
+    module = importlib.import_module(mod_name)
+    obj = module
+    for part in fn_name.split("."):
+        obj = getattr(obj, part)
+    return obj
+
+
+# -- servers ---------------------------------------------------------------
+
+
+@app.command()
+def gateway(
+    config: Optional[str] = typer.Option(None, help="config yaml path"),
+    bootstrap: bool = typer.Option(True, help="create default workspace+token on first boot"),
+):
+    """Run the gateway (control plane) on this node."""
+    import asyncio
+
+    from ..common.config import load_config
+    from ..gateway.gateway import Gateway
+    from ..scheduler.pool import LocalProcessPoolController
+
+    async def amain():
+        cfg = load_config(path=config)
+        gw = Gateway(cfg)
+        await gw.start()
+        system = gw.bootstrap_workspace("system")
+        for pool_cfg in cfg.scheduler.pools:
+            gw.scheduler.register_pool(
+                LocalProcessPoolController(
+                    cfg,
+                    pool_cfg,
+                    gateway_addr=f"{cfg.gateway.host}:{gw.grpc_port}",
+                    worker_token=system["token"],
+                )
+            )
+        if bootstrap:
+            creds = gw.bootstrap_workspace("default")
+            typer.echo(f"workspace: {creds['workspace_id']}")
+            typer.echo(f"token:     {creds['token']}")
+            typer.echo(f"grpc:      {cfg.gateway.host}:{gw.grpc_port}")
+            typer.echo(f"http:      {cfg.gateway.host}:{gw.http_port}")
+            os.makedirs(os.path.dirname(CONFIG_PATH), exist_ok=True)
+            with open(CONFIG_PATH, "w") as f:
+                yaml.dump(
+                    {"gateway_addr": f"{cfg.gateway.host}:{gw.grpc_port}", "token": creds["token"]},
+                    f,
+                )
+        stop = asyncio.Event()
+        import signal as _signal
+
+        loop = asyncio.get_event_loop()
+        for s in (_signal.SIGINT, _signal.SIGTERM):
+            loop.add_signal_handler(s, stop.set)
+        await stop.wait()
+        await gw.shutdown()
+
+    asyncio.run(amain())
+
+
+@worker_app.command("start")
+def worker_start():
+    """Run a worker against GATEWAY_ADDR (env contract)."""
+    import asyncio
+
+    from ..worker.__main__ import amain
+
+    asyncio.run(amain())
+
+
+# -- config ----------------------------------------------------------------
+
+
+@config_app.command("set")
+def config_set(gateway_addr: str, token: str):
+    os.makedirs(os.path.dirname(CONFIG_PATH), exist_ok=True)
+    with open(CONFIG_PATH, "w") as f:
+        yaml.dump({"gateway_addr": gateway_addr, "token": token}, f)
+    typer.echo(f"context saved to {CONFIG_PATH}")
+
+
+@config_app.command("show")
+def config_show():
+    typer.echo(json.dumps(_load_context(), indent=2))
+
+
+# -- dev loop ---------------------------------------------------------------
+
+
+@app.command()
+def run(handler: str, args_json: str = typer.Option("{}", "--args")):
+    """Invoke a @function once: b9 run app.py:hello --args '{"x": 2}'"""
+    fn = _load_handler(handler)
+    kwargs = json.loads(args_json)
+    if hasattr(fn, "remote"):
+        result = fn.remote(**kwargs)
+    else:
+        result = fn(**kwargs)
+    typer.echo(json.dumps(result, default=str))
+
+
+@app.command()
+def deploy(handler: str, name: str = typer.Option("", help="deployment name")):
+    """Deploy a decorated handler: b9 deploy app.py:api --name my-api"""
+    fn = _load_handler(handler)
+    if not hasattr(fn, "deploy"):
+        typer.echo("handler is not a deployable abstraction", err=True)
+        raise typer.Exit(1)
+    r = fn.deploy(name or getattr(fn, "name", "") or "app")
+    typer.echo(json.dumps(r, indent=2))
+
+
+@app.command()
+def serve(handler: str, timeout: float = typer.Option(600.0)):
+    """Serve mode (hot dev loop): keeps one container up while active."""
+    fn = _load_handler(handler)
+    if not hasattr(fn, "serve"):
+        typer.echo("handler does not support serve", err=True)
+        raise typer.Exit(1)
+    stub_id = fn.serve(timeout=timeout)
+    typer.echo(f"serving stub {stub_id} (lock ttl {timeout}s)")
+
+
+# -- tasks -------------------------------------------------------------------
+
+
+@task_app.command("list")
+def task_list(stub_id: str = typer.Option("", "--stub"), limit: int = 50):
+    typer.echo(json.dumps(_client().call("GatewayService", "ListTasks", {"stub_id": stub_id, "limit": limit}), indent=2, default=str))
+
+
+@task_app.command("get")
+def task_get(task_id: str):
+    typer.echo(json.dumps(_client().call("GatewayService", "GetTaskResult", {"task_id": task_id}), indent=2, default=str))
+
+
+@task_app.command("stop")
+def task_stop(task_id: str):
+    _client().call("GatewayService", "StopTask", {"task_id": task_id})
+    typer.echo("stopped")
+
+
+# -- containers ---------------------------------------------------------------
+
+
+@container_app.command("list")
+def container_list():
+    typer.echo(json.dumps(_client().call("GatewayService", "ListContainers", {}), indent=2, default=str))
+
+
+@container_app.command("stop")
+def container_stop(container_id: str):
+    _client().call("GatewayService", "StopContainer", {"container_id": container_id})
+    typer.echo("stop requested")
+
+
+@container_app.command("logs")
+def container_logs(container_id: str, tail: int = 200):
+    for line in _client().call("GatewayService", "GetContainerLogs", {"container_id": container_id, "tail": tail}):
+        typer.echo(line)
+
+
+@container_app.command("spans")
+def container_spans(container_id: str):
+    spans = _client().call("GatewayService", "GetContainerSpans", {"container_id": container_id})
+    for s in spans:
+        typer.echo(f"{s['name']:40s} {s['duration_ms']:10.2f} ms")
+
+
+# -- deployments ---------------------------------------------------------------
+
+
+@deployment_app.command("list")
+def deployment_list():
+    typer.echo(json.dumps(_client().call("GatewayService", "ListDeployments", {}), indent=2, default=str))
+
+
+@deployment_app.command("stop")
+def deployment_stop(deployment_id: str):
+    _client().call("GatewayService", "StopDeployment", {"deployment_id": deployment_id})
+    typer.echo("stopped")
+
+
+# -- secrets --------------------------------------------------------------------
+
+
+@secret_app.command("create")
+def secret_create(name: str, value: str):
+    _client().call("SecretService", "SecretCreate", {"name": name, "value": value})
+    typer.echo("created")
+
+
+@secret_app.command("list")
+def secret_list():
+    typer.echo(json.dumps(_client().call("SecretService", "SecretList", {}), indent=2))
+
+
+@secret_app.command("delete")
+def secret_delete(name: str):
+    _client().call("SecretService", "SecretDelete", {"name": name})
+    typer.echo("deleted")
+
+
+# -- volumes ----------------------------------------------------------------------
+
+
+@volume_app.command("ls")
+def volume_ls(volume_name: str, path: str = ""):
+    typer.echo(json.dumps(_client().call("VolumeService", "ListPath", {"volume_name": volume_name, "path": path}), indent=2))
+
+
+@volume_app.command("cp")
+def volume_cp(volume_name: str, src: str, dst: str):
+    """Upload a local file into a volume."""
+    with open(src, "rb") as f:
+        data = f.read()
+    chunks = [{"volume_name": volume_name, "path": dst, "data": data[i : i + (1 << 20)]} for i in range(0, max(len(data), 1), 1 << 20)]
+    r = _client().client_stream("VolumeService", "UploadFile", chunks)
+    typer.echo(json.dumps(r))
+
+
+@volume_app.command("cat")
+def volume_cat(volume_name: str, path: str):
+    for msg in _client().stream("VolumeService", "DownloadFile", {"volume_name": volume_name, "path": path}):
+        sys.stdout.buffer.write(msg["data"])
+
+
+# -- workers / ops -----------------------------------------------------------------
+
+
+@worker_app.command("list")
+def worker_list():
+    typer.echo(json.dumps(_client().call("GatewayService", "ListWorkers", {}), indent=2, default=str))
+
+
+@worker_app.command("cordon")
+def worker_cordon(worker_id: str, uncordon: bool = typer.Option(False)):
+    _client().call("GatewayService", "CordonWorker", {"worker_id": worker_id, "cordoned": not uncordon})
+    typer.echo("ok")
+
+
+def main():
+    app()
+
+
+if __name__ == "__main__":
+    main()

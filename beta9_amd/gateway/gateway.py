@@ -131,6 +131,15 @@ class Gateway:
         await self.scheduler.start()
         self._register_services()
         self.grpc_port = await self.rpc_server.start()
+        if self.cfg.cache.enabled:
+            from ..cache.transport import RawTransportServer
+
+            self.cache_transport = RawTransportServer(
+                self.content_store,
+                host=self.cfg.gateway.host,
+                inflight_bytes_cap=self.cfg.cache.inflight_bytes_cap,
+            )
+            self.cache_service.transport_port = await self.cache_transport.start()
         await self._start_http()
         self._started = True
         log.info("gateway up: grpc=%d http=%d", self.grpc_port, self.http_port)
@@ -181,10 +190,23 @@ class Gateway:
         self.dispatcher.register("taskqueue", self.taskqueue_exec.execute)
         tq_svc = TaskQueueService(self)
 
+        from ..cache.service import CacheCoordinator, CacheService
+        from ..cache.store import ContentStore
+
+        self.content_store = ContentStore(
+            self.cfg.cache.root,
+            page_size=self.cfg.cache.page_size,
+            quota_bytes=self.cfg.cache.disk_quota_bytes,
+            eviction_watermark=self.cfg.cache.eviction_watermark,
+        )
+        self.cache_coordinator = CacheCoordinator(self.store)
+        self.cache_service = CacheService(self.content_store, self.cache_coordinator)
+
         self._services = [
             gw_svc,
             runner_svc,
             repo_svc,
+            self.cache_service,
             fn_svc,
             ep_svc,
             tq_svc,
@@ -239,6 +261,8 @@ class Gateway:
                 await asyncio.wait_for(self._http_task, timeout=5.0)
             except (asyncio.TimeoutError, asyncio.CancelledError):
                 pass
+        if getattr(self, "cache_transport", None) is not None:
+            await self.cache_transport.stop()
         await self.rpc_server.stop()
         await self.store.stop()
         self.db.close()
