@@ -10,6 +10,7 @@ keep-warm refresh :941-952).
 from __future__ import annotations
 
 import asyncio
+import sys
 import logging
 import time
 from typing import Any, Dict, Optional
@@ -28,9 +29,15 @@ class _ContainerSlot:
         self.sema = asyncio.Semaphore(tokens)
         self.healthy = False
         self.last_probe = 0.0
+        self.rpc: "RpcClient" = None  # grpc protocol connection
 
 
 class RequestBuffer:
+    """protocol: "grpc" = the fast internal dispatch to our endpoint runner
+    (msgpack-gRPC Invoke/Health — ~10x cheaper per request than the pure-
+    python HTTP stack in this image); "http" = reverse proxy to user ASGI
+    apps (reference behavior: buffer.go:778 handleHttpRequest)."""
+
     def __init__(
         self,
         store: StateStore,
@@ -40,6 +47,7 @@ class RequestBuffer:
         keep_warm_seconds: float = 10.0,
         request_timeout: float = 300.0,
         on_inflight_change=None,
+        protocol: str = "grpc",
     ):
         self.store = store
         self.containers = container_repo
@@ -48,6 +56,7 @@ class RequestBuffer:
         self.keep_warm_seconds = keep_warm_seconds
         self.request_timeout = request_timeout
         self.on_inflight_change = on_inflight_change
+        self.protocol = protocol
         self.inflight = 0
         self._slots: Dict[str, _ContainerSlot] = {}
         self._client = httpx.AsyncClient(
@@ -58,12 +67,17 @@ class RequestBuffer:
 
     async def close(self) -> None:
         await self._client.aclose()
+        for slot in self._slots.values():
+            if slot.rpc is not None:
+                await slot.rpc.close()
 
     async def _discover(self) -> Dict[str, _ContainerSlot]:
         addrs = await self.containers.addresses_for_stub(self.stub_id)
         for cid in list(self._slots.keys()):
             if cid not in addrs:
-                del self._slots[cid]
+                old = self._slots.pop(cid)
+                if old.rpc is not None:
+                    asyncio.ensure_future(old.rpc.close())
         now = time.monotonic()
         for cid, addr in addrs.items():
             slot = self._slots.get(cid)
@@ -72,12 +86,22 @@ class RequestBuffer:
                 self._slots[cid] = slot
             if not slot.healthy and now - slot.last_probe > self._probe_interval:
                 slot.last_probe = now
-                slot.healthy = await self._probe(slot.addr)
+                slot.healthy = await self._probe(slot)
         return self._slots
 
-    async def _probe(self, addr: str) -> bool:
+    async def _probe(self, slot: "_ContainerSlot") -> bool:
+        if self.protocol == "grpc":
+            try:
+                if slot.rpc is None:
+                    from ...rpc.channel import RpcClient
+
+                    slot.rpc = RpcClient(slot.addr)
+                r = await slot.rpc.call("Runner", "Health", {}, timeout=2.0)
+                return bool(r.get("ok"))
+            except Exception:
+                return False
         try:
-            r = await self._client.get(f"http://{addr}/health", timeout=2.0)
+            r = await self._client.get(f"http://{slot.addr}/health", timeout=2.0)
             return r.status_code == 200
         except Exception:
             return False
@@ -103,21 +127,37 @@ class RequestBuffer:
             await asyncio.sleep(0.005)
         raise TimeoutError("no healthy container available")
 
-    async def forward(
+    async def dispatch(
         self,
-        method: str,
-        path: str,
+        method: str = "POST",
+        path: str = "/",
         headers: Optional[Dict[str, str]] = None,
         content: bytes = b"",
         json_body: Any = None,
         wait_timeout: float = 120.0,
-    ) -> httpx.Response:
+    ) -> Dict[str, Any]:
+        """Dispatch one request; returns {"status_code", "body"}."""
         self.inflight += 1
         if self.on_inflight_change:
             self.on_inflight_change(self.inflight)
         try:
             cid, slot = await self._acquire_slot(time.monotonic() + wait_timeout)
             try:
+                if self.protocol == "grpc":
+                    args = json_body
+                    if args is None and content:
+                        import json as _json
+
+                        try:
+                            args = _json.loads(content)
+                        except ValueError:
+                            args = {}
+                    r = await slot.rpc.call(
+                        "Runner", "Invoke", {"args": args, "path": path},
+                        timeout=self.request_timeout,
+                    )
+                    await self._after_request(cid)
+                    return r  # {"status_code", "body"}
                 url = f"http://{slot.addr}{path}"
                 kwargs: Dict[str, Any] = {"headers": headers or {}}
                 if json_body is not None:
@@ -126,9 +166,24 @@ class RequestBuffer:
                     kwargs["content"] = content
                 resp = await self._client.request(method, url, **kwargs)
                 await self._after_request(cid)
-                return resp
+                try:
+                    body = resp.json()
+                except ValueError:
+                    body = resp.content
+                return {"status_code": resp.status_code, "body": body}
             except (httpx.ConnectError, httpx.ReadError):
                 slot.healthy = False
+                raise
+            except Exception:
+                if self.protocol == "grpc":
+                    import grpc as _grpc
+
+                    exc = sys.exc_info()[1]
+                    if isinstance(exc, _grpc.aio.AioRpcError) and exc.code() in (
+                        _grpc.StatusCode.UNAVAILABLE,
+                        _grpc.StatusCode.DEADLINE_EXCEEDED,
+                    ):
+                        slot.healthy = False
                 raise
             finally:
                 slot.sema.release()

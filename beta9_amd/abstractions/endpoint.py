@@ -35,6 +35,7 @@ class EndpointInstance:
             raise KeyError(f"stub {stub_id}")
         self.cfg: StubConfig = gateway.db.get_stub_config(stub_id)
         serve_mode = self.stub_row["type"].endswith("/serve")
+        kind = self.stub_row["type"].split("/")[0]
         self.buffer = RequestBuffer(
             gateway.store,
             gateway.container_repo,
@@ -42,6 +43,7 @@ class EndpointInstance:
             concurrent_requests=self.cfg.concurrent_requests,
             keep_warm_seconds=self.cfg.keep_warm_seconds,
             request_timeout=self.cfg.timeout,
+            protocol="http" if kind == "asgi" else "grpc",
         )
 
         async def sample() -> float:
@@ -91,7 +93,7 @@ class EndpointInstance:
 
     async def forward(self, method: str, path: str, headers=None, content: bytes = b"", json_body=None):
         self.instance.autoscaler.trigger()
-        return await self.buffer.forward(
+        return await self.buffer.dispatch(
             method, path, headers=headers, content=content, json_body=json_body,
             wait_timeout=min(self.cfg.timeout, 120.0),
         )
@@ -119,15 +121,10 @@ class EndpointExecutor:
                 json_body=payload.get("json"),
                 content=payload.get("content", b""),
             )
-            body: Any
-            try:
-                body = resp.json()
-            except Exception:
-                body = resp.content
             await self.gw.dispatcher.complete(
                 task.task_id,
-                {"status_code": resp.status_code, "body": body},
-                status=TaskStatus.complete if resp.status_code < 500 else TaskStatus.error,
+                resp,
+                status=TaskStatus.complete if resp["status_code"] < 500 else TaskStatus.error,
             )
         except Exception as e:
             log.warning("endpoint task %s failed: %s", task.task_id, e)

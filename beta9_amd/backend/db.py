@@ -130,10 +130,54 @@ class BackendDB:
         self._conn.row_factory = sqlite3.Row
         self._lock = threading.Lock()
         with self._lock:
+            self._conn.execute("PRAGMA journal_mode=WAL")
+            self._conn.execute("PRAGMA synchronous=NORMAL")
             self._conn.executescript(_SCHEMA)
             self._conn.commit()
+        # write-behind queue for hot-path task rows (authoritative task
+        # state lives in the state store; rows are the durable audit log)
+        import queue as _queue
+
+        self._wq: "_queue.Queue" = _queue.Queue()
+        self._writer = threading.Thread(target=self._write_loop, daemon=True)
+        self._writer_stop = threading.Event()
+        self._writer.start()
+
+    def _write_loop(self) -> None:
+        import queue as _queue
+
+        while not self._writer_stop.is_set():
+            try:
+                item = self._wq.get(timeout=0.2)
+            except _queue.Empty:
+                continue
+            batch = [item]
+            try:
+                while len(batch) < 256:
+                    batch.append(self._wq.get_nowait())
+            except _queue.Empty:
+                pass
+            with self._lock:
+                for sql, args in batch:
+                    try:
+                        self._conn.execute(sql, args)
+                    except sqlite3.Error:
+                        pass
+                try:
+                    self._conn.commit()
+                except sqlite3.Error:
+                    pass
+
+    def flush(self, timeout: float = 5.0) -> None:
+        """Drain the write-behind queue (tests / shutdown)."""
+        deadline = time.time() + timeout
+        while not self._wq.empty() and time.time() < deadline:
+            time.sleep(0.005)
 
     def close(self) -> None:
+        self.flush()
+        self._writer_stop.set()
+        self._writer.join(timeout=2.0)
         self._conn.close()
 
     def _exec(self, sql: str, args: tuple = ()) -> sqlite3.Cursor:
@@ -309,9 +353,21 @@ class BackendDB:
             (task_id, workspace_id, stub_id, status, time.time()),
         )
 
+    def create_task_async(self, task_id: str, workspace_id: str, stub_id: str, status: str) -> None:
+        self._wq.put(
+            (
+                "INSERT INTO tasks (id, workspace_id, stub_id, status, created_at) VALUES (?,?,?,?,?)",
+                (task_id, workspace_id, stub_id, status, time.time()),
+            )
+        )
+
     def update_task(self, task_id: str, **fields: Any) -> None:
         sets = ", ".join(f"{k}=?" for k in fields)
         self._exec(f"UPDATE tasks SET {sets} WHERE id=?", (*fields.values(), task_id))
+
+    def update_task_async(self, task_id: str, **fields: Any) -> None:
+        sets = ", ".join(f"{k}=?" for k in fields)
+        self._wq.put((f"UPDATE tasks SET {sets} WHERE id=?", (*fields.values(), task_id)))
 
     def get_task(self, task_id: str) -> Optional[Dict[str, Any]]:
         rows = self._query("SELECT * FROM tasks WHERE id=?", (task_id,))

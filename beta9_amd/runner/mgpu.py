@@ -129,18 +129,18 @@ def _serve_rank0(world: int, procs: List[subprocess.Popen]) -> int:
     _rendezvous = rdzv
     gpu = GPUBoot(env)
     gpu.boot()
-    from .endpoint import build_app
+    from .endpoint import run_rpc_server
 
-    app = build_app(env, gpu, rdzv=rdzv)
-    import uvicorn
-
-    uvicorn.run(app, host="0.0.0.0", port=env.bind_port, log_level="warning", lifespan="on", access_log=False)
     try:
-        rdzv.broadcast(_STOP)
-    except Exception:
-        pass
-    rdzv.shutdown()
-    return 0
+        return asyncio.run(run_rpc_server(env, gpu, rdzv=rdzv))
+    except KeyboardInterrupt:
+        return 0
+    finally:
+        try:
+            rdzv.broadcast(_STOP)
+        except Exception:
+            pass
+        rdzv.shutdown()
 
 
 def _rank_worker(rank: int, world: int) -> int:

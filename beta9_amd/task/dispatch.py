@@ -33,6 +33,9 @@ class Dispatcher:
         self._monitor_interval_s = monitor_interval_s
         self._executors: Dict[str, Executor] = {}
         self._monitor_task: Optional[asyncio.Task] = None
+        # event-driven completion (hot path: no polling between a runner's
+        # EndTask and the caller's wait())
+        self._done_events: Dict[str, asyncio.Event] = {}
 
     def register(self, executor_name: str, fn: Executor) -> None:
         self._executors[executor_name] = fn
@@ -68,7 +71,7 @@ class Dispatcher:
             policy=policy or TaskPolicy(),
             executor=executor_name,
         )
-        self.db.create_task(task.task_id, workspace_id, stub_id, task.status.value)
+        self.db.create_task_async(task.task_id, workspace_id, stub_id, task.status.value)
         await self.store.set(keys.task_state(task.task_id), task.model_dump())
         await self.store.set(keys.task_args(task.task_id), payload)
         await self._execute(task, payload)
@@ -96,7 +99,7 @@ class Dispatcher:
         if status.is_terminal:
             task.ended_at = time.time()
         await self.store.set(keys.task_state(task_id), task.model_dump())
-        self.db.update_task(
+        self.db.update_task_async(
             task_id,
             status=status.value,
             container_id=task.container_id,
@@ -104,6 +107,10 @@ class Dispatcher:
             ended_at=task.ended_at,
             retries=task.retries,
         )
+        if status.is_terminal:
+            ev = self._done_events.pop(task_id, None)
+            if ev is not None:
+                ev.set()
 
     async def claim(self, task_id: str, container_id: str) -> bool:
         """A runner claims a task before executing it; the claim plus its
@@ -134,13 +141,27 @@ class Dispatcher:
         return await self.store.get(keys.task_cancel(task_id)) is not None
 
     async def wait(self, task_id: str, timeout: float = 300.0, poll: float = 0.005) -> Task:
-        deadline = time.monotonic() + timeout
-        while time.monotonic() < deadline:
+        task = await self.get_task(task_id)
+        if task is not None and task.status.is_terminal:
+            return task
+        ev = self._done_events.get(task_id)
+        if ev is None:
+            ev = self._done_events[task_id] = asyncio.Event()
+            # re-check after publishing the event (complete() may have run
+            # between the status read and the event registration)
             task = await self.get_task(task_id)
-            if task and task.status.is_terminal:
+            if task is not None and task.status.is_terminal:
+                self._done_events.pop(task_id, None)
                 return task
-            await asyncio.sleep(poll)
-        raise TimeoutError(f"task {task_id} did not finish in {timeout}s")
+        try:
+            await asyncio.wait_for(ev.wait(), timeout=timeout)
+        except asyncio.TimeoutError:
+            self._done_events.pop(task_id, None)
+            raise TimeoutError(f"task {task_id} did not finish in {timeout}s")
+        task = await self.get_task(task_id)
+        if task is None:
+            raise TimeoutError(f"task {task_id} vanished")
+        return task
 
     # -- retry monitor ------------------------------------------------------
     async def _monitor_loop(self) -> None:

@@ -57,6 +57,17 @@ def test_mgpu_endpoint_gloo_world2():
         stderr=subprocess.STDOUT,
         start_new_session=True,
     )
+    import asyncio
+
+    from beta9_amd.rpc.channel import RpcClient
+
+    async def call(method, payload, timeout=60.0):
+        c = RpcClient(f"127.0.0.1:{bind}")
+        try:
+            return await c.call("Runner", method, payload, timeout=timeout)
+        finally:
+            await c.close()
+
     try:
         deadline = time.monotonic() + 120
         up = False
@@ -65,27 +76,27 @@ def test_mgpu_endpoint_gloo_world2():
                 out = proc.stdout.read().decode(errors="replace")
                 raise AssertionError(f"runner exited early:\n{out[-3000:]}")
             try:
-                r = httpx.get(f"http://127.0.0.1:{bind}/health", timeout=2.0)
-                if r.status_code == 200:
+                h = asyncio.run(call("Health", {}, timeout=2.0))
+                if h.get("ok"):
                     up = True
                     break
-            except httpx.HTTPError:
+            except Exception:
                 pass
             time.sleep(0.2)
         assert up, "mgpu endpoint never became healthy"
 
-        r = httpx.post(f"http://127.0.0.1:{bind}/", json={"tokens": 8}, timeout=60.0)
-        assert r.status_code == 200, r.text
-        result = r.json()["result"]
+        r = asyncio.run(call("Invoke", {"args": {"tokens": 8}}))
+        assert r["status_code"] == 200, str(r)
+        result = r["body"]["result"]
         assert result["world"] == 2
         assert result["load"]["world"] == 2
         assert result["load"]["backend"] == "gloo"
         first_checksum = result["checksum"]
 
         # second request exercises the broadcast loop again
-        r2 = httpx.post(f"http://127.0.0.1:{bind}/", json={"tokens": 8}, timeout=60.0)
-        assert r2.status_code == 200
-        assert r2.json()["result"]["checksum"] != 0.0
+        r2 = asyncio.run(call("Invoke", {"args": {"tokens": 8}}))
+        assert r2["status_code"] == 200
+        assert r2["body"]["result"]["checksum"] != 0.0
         assert first_checksum != 0.0
     finally:
         import signal
