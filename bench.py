@@ -134,6 +134,10 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
         cold_ms = (time.perf_counter() - t0) * 1000.0
         if not first["ok"]:
             raise RuntimeError(f"cold-start request failed: {str(first)[:400]}")
+        try:
+            first_load = first["result"]["body"]["result"]["load"]
+        except (KeyError, TypeError):
+            first_load = {}
 
         async def step() -> int:
             results = await asyncio.gather(
@@ -201,6 +205,7 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
             "cold_list": cold_list,
             "native_gpu": has_gpu,
             "spans": spans,
+            "first_load": first_load,
         }
 
 
@@ -246,6 +251,7 @@ def main() -> int:
             "cold_start_p50_ms": round(p50, 1),
             "cold_start_all_ms": [round(x, 1) for x in cold_sorted],
             "cold_start_phases_ms": r["spans"],
+            "first_container_load": r["first_load"],
             "native_gpu_path": r["native_gpu"],
             "config": {
                 "model": f"synthetic-mlp-{args.layers}L-{args.hidden}h-{args.inter}i",

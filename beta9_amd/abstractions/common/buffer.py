@@ -63,7 +63,9 @@ class RequestBuffer:
             timeout=httpx.Timeout(request_timeout, connect=5.0),
             limits=httpx.Limits(max_connections=512, max_keepalive_connections=256),
         )
-        self._probe_interval = 0.25
+        # unhealthy->healthy detection granularity; probes are a local gRPC
+        # call, so keep this tight — it bounds cold-start detection latency
+        self._probe_interval = 0.02
 
     async def close(self) -> None:
         await self._client.aclose()

@@ -56,6 +56,30 @@ def _load_ext():
 DEFAULT_ARENA_BYTES = int(os.environ.get("B9_ARENA_BYTES", str(4 << 30)))
 DEFAULT_STAGING_BYTES = int(os.environ.get("B9_STAGING_BYTES", str(256 << 20)))
 
+# Arena pre-created by the prewarm shell (worker/prewarm pool): hipMalloc of
+# the slab + pinned staging cost ~0.5 s and move off the cold-start path.
+_prewarmed: Optional["Arena"] = None
+
+
+def prewarm_arena(device: int = 0, size_bytes: int = 0) -> Optional["Arena"]:
+    global _prewarmed
+    try:
+        _prewarmed = Arena(device=device, size_bytes=size_bytes)
+        return _prewarmed
+    except Exception:
+        _prewarmed = None
+        return None
+
+
+def take_prewarmed(min_size: int = 0) -> Optional["Arena"]:
+    """Claim the shell's pre-created arena if it is big enough."""
+    global _prewarmed
+    a = _prewarmed
+    if a is not None and a.native and (min_size == 0 or a.size_bytes >= min_size):
+        _prewarmed = None
+        return a
+    return None
+
 
 class Arena:
     """One reserved HBM slab per GPU process, sub-allocated bump-style.

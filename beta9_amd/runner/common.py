@@ -149,7 +149,13 @@ class GPUBoot:
                 if meta is not None:
                     total = sum(r["nbytes"] for r in meta["regions"])
                     size = int(total * 1.25) + (64 << 20)
-            self.arena = hipsnap.Arena(device=0, size_bytes=size)
+            # prefer the prewarm shell's pre-created arena (hipMalloc +
+            # pinned staging already paid before the container started)
+            self.arena = hipsnap.take_prewarmed(min_size=size)
+            if self.arena is None:
+                self.arena = hipsnap.Arena(device=0, size_bytes=size)
+            else:
+                self.phases["arena_prewarmed"] = 1.0
             self.phases["hbm_prealloc_ms"] = (time.perf_counter() - t1) * 1000
             if snap_path:
                 t2 = time.perf_counter()

@@ -35,6 +35,11 @@ def main() -> int:
         try:
             torch.cuda.init()
             torch.zeros(1, device="cuda:0")  # force full HIP context + pools
+            # pre-create the HBM arena + pinned staging (hipMalloc slab and
+            # hipHostMalloc cost ~0.5 s — off the cold-start path)
+            from beta9_amd.ops import hipsnap
+
+            hipsnap.prewarm_arena(device=0)
         except Exception:
             pass
     for mod in ("fastapi", "uvicorn", "grpc", "msgpack", "httpx"):
