@@ -132,8 +132,13 @@ class ContainerRepo:
         self.store = store
 
     async def set_state(self, st: ContainerState) -> None:
+        is_new = await self.store.get(keys.container_state(st.container_id)) is None
         await self.store.set(keys.container_state(st.container_id), st.model_dump())
         await self.store.hset(keys.containers_by_stub(st.stub_id), {st.container_id: st.status.value})
+        if is_new and st.workspace_id:
+            await self.store.hincr("containers:count", st.workspace_id, 1)
+            if st.gpu_count > 0:
+                await self.store.hincr("containers:gpu_count", st.workspace_id, st.gpu_count)
 
     async def get_state(self, container_id: str) -> Optional[ContainerState]:
         d = await self.store.get(keys.container_state(container_id))
@@ -149,8 +154,18 @@ class ContainerRepo:
         st = await self.get_state(container_id)
         if st:
             await self.store.hdel(keys.containers_by_stub(st.stub_id), container_id)
+            if st.workspace_id:
+                await self.store.hincr("containers:count", st.workspace_id, -1)
+                if st.gpu_count > 0:
+                    await self.store.hincr("containers:gpu_count", st.workspace_id, -st.gpu_count)
         await self.store.delete(keys.container_state(container_id))
         await self.store.delete(keys.container_address(container_id))
+
+    async def workspace_counts(self, workspace_id: str) -> Dict[str, int]:
+        return {
+            "containers": int(await self.store.hget("containers:count", workspace_id) or 0),
+            "gpus": int(await self.store.hget("containers:gpu_count", workspace_id) or 0),
+        }
 
     async def containers_for_stub(self, stub_id: str) -> Dict[str, str]:
         return await self.store.hgetall(keys.containers_by_stub(stub_id))

@@ -127,6 +127,13 @@ class Gateway:
             return
         os.makedirs(self.cfg.storage.root, exist_ok=True)
         await self.store.start()
+        from ..common.events import EventSink
+        from ..common.metrics import REGISTRY
+
+        self.events = EventSink(self.store, jsonl_path=self.cfg.monitoring.span_sink_path)
+        self.metrics = REGISTRY
+        self.scheduler.events = self.events
+        self.scheduler.metrics_registry = self.metrics
         await self.dispatcher.start()
         await self.scheduler.start()
         self._register_services()
@@ -141,6 +148,12 @@ class Gateway:
             )
             self.cache_service.transport_port = await self.cache_transport.start()
         await self._start_http()
+        from .loops import CronLoop, InstanceController, PoolCleaner, PoolSizer
+
+        self._loops = [CronLoop(self), PoolCleaner(self), PoolSizer(self)]
+        for loop in self._loops:
+            loop.start()
+        await InstanceController(self).load()
         self._started = True
         log.info("gateway up: grpc=%d http=%d", self.grpc_port, self.http_port)
 
@@ -176,6 +189,23 @@ class Gateway:
         self.scheduler.checkpoint_lookup = (
             lambda ws, stub_id: self.db.latest_checkpoint(ws, stub_id)
         )
+
+        async def concurrency_check(request) -> bool:
+            ws = self.db.get_workspace(request.workspace_id)
+            if ws is None:
+                return True
+            counts = await self.container_repo.workspace_counts(request.workspace_id)
+            cpu_limit = ws.get("concurrency_limit_cpu") or 0
+            gpu_limit = ws.get("concurrency_limit_gpu") or 0
+            if cpu_limit and counts["containers"] >= cpu_limit:
+                return False
+            if gpu_limit and request.gpu_count > 0 and counts["gpus"] + request.gpu_count > gpu_limit:
+                return False
+            if request.gpu_count > 1 and not ws.get("multi_gpu_enabled", 1):
+                return False
+            return True
+
+        self.scheduler.concurrency_check = concurrency_check
         repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo)
 
         fn_exec = FunctionExecutor(self.db, self.scheduler, self.store, self.runner_env)
@@ -251,6 +281,8 @@ class Gateway:
     async def shutdown(self) -> None:
         if not self._started:
             return
+        for loop in getattr(self, "_loops", []):
+            await loop.stop()
         for inst in list(self._instances.values()):
             await inst.shutdown()
         await self.scheduler.shutdown()

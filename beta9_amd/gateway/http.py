@@ -136,7 +136,20 @@ def build_http_app(gateway) -> FastAPI:
         return {
             "scheduler": gateway.scheduler.metrics.snapshot(),
             "backlog": await gateway.scheduler.backlog_len(),
+            "registry": gateway.metrics.snapshot() if getattr(gateway, "metrics", None) else {},
         }
+
+    @app.get("/metrics")
+    async def metrics_prometheus():
+        if getattr(gateway, "metrics", None) is None:
+            return Response(content="", media_type="text/plain")
+        return Response(content=gateway.metrics.prometheus_text(), media_type="text/plain")
+
+    @app.get("/api/v1/events")
+    async def events(limit: int = 100, prefix: str = "", auth: dict = Depends(auth_ctx)):
+        if getattr(gateway, "events", None) is None:
+            return []
+        return await gateway.events.recent(limit=limit, name_prefix=prefix)
 
     @app.get("/api/v1/outputs/{rel:path}")
     async def get_output(rel: str, auth: dict = Depends(auth_ctx)):

@@ -54,6 +54,10 @@ class PoolController(Protocol):
     async def shutdown(self) -> None: ...
 
 
+class ConcurrencyLimitError(Exception):
+    pass
+
+
 class SchedulerMetrics:
     def __init__(self) -> None:
         self.scheduled = 0
@@ -81,6 +85,11 @@ class Scheduler:
         self.metrics = SchedulerMetrics()
         # set by the gateway: (workspace_id, stub_id) -> checkpoint row dict
         self.checkpoint_lookup = None
+        # set by the gateway: async (request) -> bool (workspace concurrency
+        # limits, reference: CheckConcurrencyLimit scheduler.go:391-505)
+        self.concurrency_check = None
+        self.events = None  # EventSink
+        self.metrics_registry = None  # MetricsRegistry
         self._loop_task: Optional[asyncio.Task] = None
         self._wake = asyncio.Event()
 
@@ -91,6 +100,10 @@ class Scheduler:
     async def run(self, request: ContainerRequest) -> None:
         """Accept a container request: normalize, record state, enqueue."""
         self._normalize_gpu(request)
+        if self.concurrency_check is not None and not await self.concurrency_check(request):
+            raise ConcurrencyLimitError(
+                f"workspace {request.workspace_id} at its concurrency limit"
+            )
         if request.checkpoint_enabled and self.checkpoint_lookup is not None:
             cp = self.checkpoint_lookup(request.workspace_id, request.stub_id)
             if cp is not None and (not cp.get("accelerator") or not request.gpu or cp["accelerator"] == request.gpu):
@@ -203,16 +216,33 @@ class Scheduler:
         candidates.sort(key=lambda w: (-self._score(request, w), -self._fullness(w)))
         for w in candidates:
             if await self.workers.schedule_container_request(w, request):
-                await self.store.delete(keys.scheduler_request(request.container_id))
-                self.metrics.scheduled += 1
+                await self._committed(request, w)
                 return
             # capacity raced; refresh this worker snapshot and continue
             fresh = await self.workers.get(w.id)
             if fresh and await self.workers.schedule_container_request(fresh, request):
-                await self.store.delete(keys.scheduler_request(request.container_id))
-                self.metrics.scheduled += 1
+                await self._committed(request, fresh)
                 return
         await self._no_fit(request, workers)
+
+    async def _committed(self, request: ContainerRequest, worker: Worker) -> None:
+        await self.store.delete(keys.scheduler_request(request.container_id))
+        self.metrics.scheduled += 1
+        wait_ms = (time.time() - request.timestamp) * 1000.0
+        if self.metrics_registry is not None:
+            self.metrics_registry.inc("scheduler_containers_scheduled")
+            self.metrics_registry.observe("scheduler_schedule_wait_ms", wait_ms)
+        if self.events is not None:
+            await self.events.emit(
+                "container.scheduled",
+                container_id=request.container_id,
+                worker_id=worker.id,
+                stub_id=request.stub_id,
+                gpu=request.gpu,
+                gpu_count=request.gpu_count,
+                retries=request.retry_count,
+                wait_ms=round(wait_ms, 2),
+            )
 
     def _filter(self, request: ContainerRequest, workers: List[Worker]) -> List[Worker]:
         """Filter chain (reference order: scheduler.go:1306-1316)."""
