@@ -45,6 +45,20 @@ def main():
         dt = bench_gpu(lambda: a @ w.t())
         out[f"gemm_{M}x{N}x{K}_torch"] = round(2 * M * N * K / dt / 1e12, 1)
 
+    # -- fused gate+up+silu vs unfused (2 gemm + silu_mul) -----------------
+    for (M, N, K) in [(256, 14336, 4096), (2048, 14336, 4096)]:
+        a = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
+        wg = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
+        wu = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
+        dt = bench_gpu(lambda: ops.mlp_gate_up(a, wg, wu))
+        out[f"fused_gateup_{M}x{N}x{K}_TF"] = round(4 * M * N * K / dt / 1e12, 1)
+        dt = bench_gpu(lambda: ops.silu_mul(ops.gemm_nt(a, wg), ops.gemm_nt(a, wu)))
+        out[f"unfused_gateup_{M}x{N}x{K}_TF"] = round(4 * M * N * K / dt / 1e12, 1)
+        dt = bench_gpu(
+            lambda: torch.nn.functional.silu(a @ wg.t()) * (a @ wu.t())
+        )
+        out[f"torch_gateup_{M}x{N}x{K}_TF"] = round(4 * M * N * K / dt / 1e12, 1)
+
     # -- elementwise bandwidth ---------------------------------------------
     n = 64 << 20
     g = torch.randn(n, device="cuda").to(torch.bfloat16)

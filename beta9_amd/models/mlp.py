@@ -44,11 +44,18 @@ class MLPBlock:
         self.norm_w = make("norm_w", (hidden,))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        import os
+
         h = ops.rmsnorm(x, self.norm_w)
-        gate = ops.gemm_nt(h, self.w_gate)
-        up = ops.gemm_nt(h, self.w_up)
-        act = ops.silu_mul(gate, up)
-        return x + ops.gemm_nt(act, self.w_down)
+        # fused hand-written gate+up+silu (one A pass, no intermediates)
+        act = ops.mlp_gate_up(h, self.w_gate, self.w_up)
+        backend = os.environ.get("B9_GEMM_BACKEND", "hybrid")
+        if backend == "native":
+            out = ops.gemm_nt(act, self.w_down)
+        else:
+            # plain library GEMM for the unfused projection (hipBLASLt)
+            out = act @ self.w_down.t()
+        return x + out
 
 
 class SyntheticServingModel:

@@ -41,11 +41,14 @@ class TPMLPBlock:
         self.norm_w = make("norm_w", (hidden,))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        import os
+
         h = ops.rmsnorm(x, self.norm_w)
-        gate = ops.gemm_nt(h, self.w_gate)   # [tokens, shard]
-        up = ops.gemm_nt(h, self.w_up)
-        act = ops.silu_mul(gate, up)
-        out = ops.gemm_nt(act, self.w_down)  # [tokens, hidden] partial
+        act = ops.mlp_gate_up(h, self.w_gate, self.w_up)  # [tokens, shard]
+        if os.environ.get("B9_GEMM_BACKEND", "hybrid") == "native":
+            out = ops.gemm_nt(act, self.w_down)  # [tokens, hidden] partial
+        else:
+            out = act @ self.w_down.t()
         if dist.get_backend() == "gloo":
             # CPU test path: gloo has no bf16 reduction
             out32 = out.float()

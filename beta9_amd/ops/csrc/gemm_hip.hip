@@ -194,6 +194,136 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused MLP gate: act = silu(A @ Wg^T) * (A @ Wu^T), bf16.
+// One kernel instead of {gemm, gemm, silu_mul}: the A tile is staged once
+// and feeds BOTH projections' MFMAs; gate/up intermediates never touch HBM
+// (saves 2 writes + 2 reads of [M, N] plus one full A re-read).
+// LDS: A + Wg + Wu tiles, double-buffered = 96 KiB -> 1 block/CU, which is
+// exactly the regime where the counted-vmcnt prefetch pipeline pays
+// (guide §5: glds overlap matters at ~1 block/CU).
+// ---------------------------------------------------------------------------
+
+template <bool XCD_SWIZZLE>
+__global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ Wg,
+    const bf16_t* __restrict__ Wu, bf16_t* __restrict__ Out, int M, int N,
+    int K) {
+  __shared__ bf16_t lds[2 * 3 * BM * BK];  // [buf][A|Wg|Wu][128][64]
+
+  int nwg_n = N / BN;
+  int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = (wg / nwg_n) * BM;
+  int tile_n = (wg % nwg_n) * BN;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 1, wc = wave & 1;
+
+  int ld_row = lane >> 3;
+  int ld_col = (lane & 7) * 8;
+  auto stage = [&](int buf, int kt) {
+    bf16_t* Ab = lds + buf * 3 * BM * BK;
+    bf16_t* Gb = Ab + BM * BK;
+    bf16_t* Ub = Gb + BM * BK;
+    const bf16_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
+    const bf16_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK;
+    const bf16_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK;
+#pragma unroll
+    for (int call = 0; call < 4; call++) {
+      int p = wave + WAVES * call;
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ld_col),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ld_col),
+          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 512), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ld_col),
+          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 512), 16, 0, 0);
+    }
+  };
+
+  f32x4 acc_g[4][4] = {};
+  f32x4 acc_u[4][4] = {};
+
+  int ntiles = K / BK;
+  stage(0, 0);
+
+  int fr_row = lane & 15;
+  int fr_k = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < ntiles; kt++) {
+    int buf = kt & 1;
+    bf16_t* Ab = lds + buf * 3 * BM * BK;
+    bf16_t* Gb = Ab + BM * BK;
+    bf16_t* Ub = Gb + BM * BK;
+    if (kt + 1 < ntiles) {
+      stage(buf ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");  // 12 glds per tile
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ks++) {
+      bf16x8 a_frag[4], g_frag[4], u_frag[4];
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        int row = wr * 64 + i * 16 + fr_row;
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + ks * 32 + fr_k);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        int col = wc * 64 + j * 16 + fr_row;
+        g_frag[j] = *(const bf16x8*)(Gb + col * BK + ks * 32 + fr_k);
+        u_frag[j] = *(const bf16x8*)(Ub + col * BK + ks * 32 + fr_k);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          acc_g[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], g_frag[j], acc_g[i][j], 0, 0, 0);
+          acc_u[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], u_frag[j], acc_u[i][j], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // fused epilogue: silu(gate) * up
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int col = tile_n + wc * 64 + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 64 + i * 16 + c_row0 + r;
+        float g = acc_g[i][j][r];
+        float u = acc_u[i][j][r];
+        float s = g / (1.0f + __expf(-g));
+        Out[(int64_t)row * N + col] = (bf16_t)(s * u);
+      }
+    }
+  }
+}
+
 // combine split-K fp32 partials -> bf16 (8-wide)
 __global__ __launch_bounds__(256) void splitk_reduce_kernel(
     const float* __restrict__ Cpart, bf16_t* __restrict__ C, int64_t mn,
@@ -378,6 +508,31 @@ torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
   return C;
 }
 
+torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
+                                torch::Tensor Wu, bool xcd_swizzle) {
+  check_bf16_2d(A, "A");
+  check_bf16_2d(Wg, "Wg");
+  check_bf16_2d(Wu, "Wu");
+  int64_t M = A.size(0), K = A.size(1), N = Wg.size(0);
+  TORCH_CHECK(Wu.size(0) == N && Wu.size(1) == K && Wg.size(1) == K);
+  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
+  auto Out = torch::empty({M, N}, A.options());
+  dim3 grid((M / BM) * (N / BN));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (xcd_swizzle) {
+    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<true>), grid, dim3(THREADS),
+                       0, stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Wg.data_ptr(), (const bf16_t*)Wu.data_ptr(),
+                       (bf16_t*)Out.data_ptr(), (int)M, (int)N, (int)K);
+  } else {
+    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<false>), grid, dim3(THREADS),
+                       0, stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Wg.data_ptr(), (const bf16_t*)Wu.data_ptr(),
+                       (bf16_t*)Out.data_ptr(), (int)M, (int)N, (int)K);
+  }
+  return Out;
+}
+
 torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
   TORCH_CHECK(gate.is_cuda() && gate.dtype() == torch::kBFloat16);
   TORCH_CHECK(gate.sizes() == up.sizes());
@@ -411,6 +566,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "hand-written gfx950 MFMA kernels: bf16 GEMM, silu_mul, rmsnorm";
   m.def("gemm_bf16_nt", &gemm_bf16_nt, py::arg("A"), py::arg("Bt"),
         py::arg("xcd_swizzle") = true, py::arg("splitk") = -1);
+  m.def("gemm_gate_up_silu", &gemm_gate_up_silu, py::arg("A"), py::arg("Wg"),
+        py::arg("Wu"), py::arg("xcd_swizzle") = true);
   m.def("silu_mul", &silu_mul);
   m.def("rmsnorm", &rmsnorm, py::arg("x"), py::arg("w"),
         py::arg("eps") = 1e-6);

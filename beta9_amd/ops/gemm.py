@@ -56,6 +56,22 @@ def gemm_nt(a: torch.Tensor, w: torch.Tensor, xcd_swizzle: bool = True) -> torch
     return a @ w.t()
 
 
+def mlp_gate_up(a: torch.Tensor, w_gate: torch.Tensor, w_up: torch.Tensor) -> torch.Tensor:
+    """Fused act = silu(a @ w_gate.T) * (a @ w_up.T) — one hand-written
+    kernel; A staged once, no gate/up intermediates in HBM."""
+    ext = _load()
+    if ext is not None and a.is_cuda:
+        M, K = a.shape
+        N = w_gate.shape[0]
+        if N % 128 == 0 and K % 64 == 0:
+            if M % 128 != 0:
+                pad = (128 - M % 128) % 128
+                a2 = torch.nn.functional.pad(a, (0, 0, 0, pad))
+                return ext.gemm_gate_up_silu(a2.contiguous(), w_gate.contiguous(), w_up.contiguous(), True)[:M]
+            return ext.gemm_gate_up_silu(a.contiguous(), w_gate.contiguous(), w_up.contiguous(), True)
+    return torch.nn.functional.silu(a @ w_gate.t()) * (a @ w_up.t())
+
+
 def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     ext = _load()
     if ext is not None and gate.is_cuda:

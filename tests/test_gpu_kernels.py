@@ -66,6 +66,32 @@ class TestGemm:
             ref = a.float() @ w.float().t()
             assert _rel_err(out, ref) < 0.02
 
+    def test_fused_gate_up_silu(self):
+        from beta9_amd.ops import gemm as ops
+
+        torch.manual_seed(1)
+        M, N, K = 256, 384, 512
+        a = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.3
+        wg = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.3
+        wu = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.3
+        out = ops.mlp_gate_up(a, wg, wu)
+        g = a.float() @ wg.float().t()
+        u = a.float() @ wu.float().t()
+        ref = torch.nn.functional.silu(g) * u
+        assert out.shape == (M, N)
+        assert _rel_err(out, ref) < 0.03
+
+    def test_fused_gate_up_padded_m(self):
+        from beta9_amd.ops import gemm as ops
+
+        a = torch.randn(100, 256, dtype=torch.bfloat16, device="cuda") * 0.3
+        wg = torch.randn(128, 256, dtype=torch.bfloat16, device="cuda") * 0.3
+        wu = torch.randn(128, 256, dtype=torch.bfloat16, device="cuda") * 0.3
+        out = ops.mlp_gate_up(a, wg, wu)
+        ref = torch.nn.functional.silu(a.float() @ wg.float().t()) * (a.float() @ wu.float().t())
+        assert out.shape == (100, 128)
+        assert _rel_err(out, ref) < 0.03
+
     def test_silu_mul(self):
         from beta9_amd.ops import gemm as ops
 
