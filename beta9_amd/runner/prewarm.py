@@ -35,11 +35,23 @@ def main() -> int:
         try:
             torch.cuda.init()
             torch.zeros(1, device="cuda:0")  # force full HIP context + pools
+            # warm hipBLASLt (library init ~1 s on first GEMM) and our own
+            # extension modules
+            x = torch.randn(256, 256, dtype=torch.bfloat16, device="cuda:0")
+            (x @ x.t()).sum().item()
+            try:
+                from beta9_amd.ops import gemm as _g
+
+                if _g.has_native():
+                    _g.gemm_nt(x.repeat(1, 1), x)
+            except Exception:
+                pass
             # pre-create the HBM arena + pinned staging (hipMalloc slab and
             # hipHostMalloc cost ~0.5 s — off the cold-start path)
             from beta9_amd.ops import hipsnap
 
             hipsnap.prewarm_arena(device=0)
+            torch.cuda.synchronize()
         except Exception:
             pass
     for mod in ("fastapi", "uvicorn", "grpc", "msgpack", "httpx"):
