@@ -46,8 +46,23 @@ class EndpointInstance:
             protocol="http" if kind == "asgi" else "grpc",
         )
 
-        async def sample() -> float:
-            return float(self.buffer.inflight)
+        from ..types import AutoscalerType
+
+        if self.cfg.autoscaler.type == AutoscalerType.llm_token_pressure:
+            # LLM-aware scaling (reference: pod/llm.go LLMTokenPressure —
+            # vLLM-style token pressure instead of request depth): containers
+            # report pressure via RunnerService.ReportPressure; sample = sum.
+            async def sample() -> float:
+                total = 0.0
+                for cid in (await gateway.container_repo.containers_for_stub(stub_id)).keys():
+                    v = await gateway.store.get(f"llm:pressure:{stub_id}:{cid}")
+                    total += float(v or 0.0)
+                return max(total, float(self.buffer.inflight))
+
+        else:
+
+            async def sample() -> float:
+                return float(self.buffer.inflight)
 
         scale_fn = deployment_scale_fn(self.cfg, gateway.cfg.gateway.stub_limits_max_replicas)
         self.instance = AutoscaledInstance(
@@ -72,6 +87,10 @@ class EndpointInstance:
                 "TIMEOUT": str(cfg.timeout),
             }
         )
+        if cfg.inputs:
+            import json as _json
+
+            env["B9_INPUTS"] = _json.dumps(cfg.inputs)
         env.update(cfg.env)
         return ContainerRequest(
             container_id=f"endpoint-{self.stub_id[-8:]}-{new_id()}",

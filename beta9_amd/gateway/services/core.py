@@ -185,6 +185,27 @@ class GatewayService:
         result = await self.dispatcher.get_result(req["task_id"])
         return {"status": t.status.value, "result": result}
 
+    @rpc()
+    async def SyncServeWorkspace(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Serve-mode hot reload: push changed files to every running
+        container of a stub (reference: ContainerSyncWorkspace,
+        worker.proto:19 + sdk sync.py)."""
+        from ...rpc.channel import RpcClient
+
+        stub_id = req["stub_id"]
+        results = {}
+        for cid, addr in (await self.containers.addresses_for_stub(stub_id)).items():
+            client = RpcClient(addr)
+            try:
+                results[cid] = await client.call(
+                    "Runner", "SyncFiles", {"files": req.get("files", [])}, timeout=30.0
+                )
+            except Exception as e:
+                results[cid] = {"error": str(e)}
+            finally:
+                await client.close()
+        return results
+
     # -- workers (ops surface) -------------------------------------------------
     @rpc()
     async def ListWorkers(self, req: Dict[str, Any], context, auth) -> list:
@@ -243,6 +264,18 @@ class RunnerService:
             await self.store.set(
                 keys.container_keep_warm(req["stub_id"], req["container_id"]), 1, ttl=ttl
             )
+        return True
+
+    @rpc()
+    async def ReportPressure(self, req: Dict[str, Any], context, auth) -> bool:
+        """LLM serving containers report token pressure (0..N, ~queue depth
+        in units of tasks_per_container) for the llm_token_pressure
+        autoscaler (reference: pod/llm.go + LLMTokenPressureAutoscaler)."""
+        await self.store.set(
+            f"llm:pressure:{req['stub_id']}:{req['container_id']}",
+            float(req.get("pressure", 0.0)),
+            ttl=15.0,
+        )
         return True
 
     @rpc()

@@ -256,6 +256,19 @@ class RunnerClient:
             },
         )
 
+    async def report_pressure(self, pressure: float) -> None:
+        """LLM containers report token pressure for the
+        llm_token_pressure autoscaler."""
+        await self.rpc.call(
+            "RunnerService",
+            "ReportPressure",
+            {
+                "stub_id": self.env.stub_id,
+                "container_id": self.env.container_id,
+                "pressure": pressure,
+            },
+        )
+
     async def checkpoint_ready(self) -> Dict[str, Any]:
         """Signal on_start done; gateway may grant this container the
         checkpoint (single-flight per stub)."""

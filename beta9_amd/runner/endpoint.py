@@ -114,6 +114,14 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
     from ..rpc.channel import RpcServer, rpc as rpc_deco
 
     state: dict = {"ready": False, "handler": None, "dispatcher": None}
+    _inputs_raw = os.environ.get("B9_INPUTS", "")
+    if _inputs_raw:
+        import json as _json
+
+        try:
+            state["inputs_spec"] = _json.loads(_inputs_raw)
+        except ValueError:
+            state["inputs_spec"] = None
     if rdzv is not None:
         from .mgpu import Rank0Dispatcher
 
@@ -131,10 +139,46 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
             }
 
         @rpc_deco()
+        async def SyncFiles(self, req, context, auth):
+            """Serve-mode hot reload: write pushed files into the workspace
+            and reload the handler (reference: ContainerSyncWorkspace +
+            sdk sync.py watchdog loop)."""
+            base = env.workspace_dir or os.getcwd()
+            written = 0
+            for f in (req or {}).get("files", []):
+                path = os.path.normpath(os.path.join(base, f["path"].lstrip("/")))
+                if not path.startswith(os.path.realpath(base)) and not path.startswith(base):
+                    continue
+                os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+                with open(path, "wb") as fh:
+                    fh.write(f.get("data", b""))
+                written += 1
+            # reload handler module so the next request sees the new code
+            try:
+                import importlib
+
+                mod_name = env.handler.split(":")[0]
+                import sys as _sys
+
+                if mod_name in _sys.modules:
+                    importlib.reload(_sys.modules[mod_name])
+                state["handler"] = load_callable(env.handler)
+            except Exception as e:
+                return {"written": written, "reloaded": False, "error": str(e)}
+            return {"written": written, "reloaded": True}
+
+        @rpc_deco()
         async def Invoke(self, req, context, auth):
             if not state["ready"]:
                 return {"status_code": 503, "body": {"error": "starting"}}
             args = (req or {}).get("args") or {}
+            if state.get("inputs_spec"):
+                from ..schema import Schema
+
+                try:
+                    Schema.validate_payload(state["inputs_spec"], args)
+                except ValueError as ve:
+                    return {"status_code": 422, "body": {"error": str(ve)}}
             try:
                 if state["dispatcher"] is not None:
                     result = await asyncio.wait_for(

@@ -242,6 +242,25 @@ class QueueDepthAutoscaler:
         )
 
 
+@dataclass
+class LLMTokenPressureAutoscaler:
+    """Scale LLM serving containers by reported token pressure instead of
+    request depth (reference: SDK type.py LLMTokenPressureAutoscaler +
+    pod/llm.go). Containers report via RunnerClient.report_pressure()."""
+
+    max_containers: int = 1
+    pressure_per_container: float = 1.0
+    min_containers: int = 0
+
+    def to_config(self) -> AutoscalerConfig:
+        return AutoscalerConfig(
+            type=AutoscalerType.llm_token_pressure,
+            max_containers=self.max_containers,
+            tasks_per_container=max(1, int(self.pressure_per_container)),
+            min_containers=self.min_containers,
+        )
+
+
 # ---------------------------------------------------------------------------
 # runtime prep shared by all decorators
 # ---------------------------------------------------------------------------
@@ -301,6 +320,8 @@ class RunnerAbstraction:
         name: str = "",
         sync_workspace: bool = False,
         client: Optional[Client] = None,
+        inputs=None,  # Schema
+        outputs=None,  # Schema
     ):
         self.func = func
         self.cpu = cpu
@@ -323,6 +344,8 @@ class RunnerAbstraction:
         self.pool = pool
         self.name = name
         self.sync_workspace = sync_workspace
+        self.inputs = inputs
+        self.outputs = outputs
         self._client = client
         self._stub_id: Optional[str] = None
         if func is not None:
@@ -369,6 +392,8 @@ class RunnerAbstraction:
             checkpoint_enabled=self.checkpoint_enabled,
             pool=self.pool,
             timeout=self.timeout,
+            inputs=self.inputs.to_spec() if self.inputs is not None else None,
+            outputs=self.outputs.to_spec() if self.outputs is not None else None,
         )
         return cfg
 
