@@ -87,7 +87,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
   // Piece p covers rows p*8 .. p*8+7 (128 B per row); lane l handles
   // row p*8 + l/8, 16-byte chunk l%8.
   int ld_row = lane >> 3;          // 0..7 within piece
-  int ld_col = (lane & 7) * 8;     // bf16 column of 16B chunk
+  int ld_chunk = lane & 7;         // 16B chunk index within the 128B row
   auto stage = [&](int buf, int kt) {
     bf16_t* Ab = lds + buf * 2 * BM * BK;
     bf16_t* Bb = Ab + BM * BK;
@@ -100,13 +100,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(a_src +
                                                              (int64_t)row * K +
-                                                             ld_col),
+                                                             ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512),
           16, 0, 0);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(b_src +
                                                              (int64_t)row * K +
-                                                             ld_col),
+                                                             ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Bb + p * 512),
           16, 0, 0);
     }
@@ -124,7 +124,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
 
   // fragment read offsets: lane l -> row (l&15), k (l>>4)*8
   int fr_row = lane & 15;
-  int fr_k = (lane >> 4) * 8;
+  int fr_q = lane >> 4;  // linear 16B chunk sub-index (0..3) within k-step
 
   // Pipelined K-loop with COUNTED vmcnt + raw barriers (guide §5: a
   // __syncthreads() with a glds in flight emits vmcnt(0) and drains the
@@ -151,12 +151,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
 #pragma unroll
       for (int i = 0; i < 4; i++) {
         int row = wr * 64 + i * 16 + fr_row;
-        a_frag[i] = *(const bf16x8*)(Ab + row * BK + ks * 32 + fr_k);
+        int sc = (ks * 4 + fr_q) ^ (row & 7);  // bank-conflict-free swizzle
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
       }
 #pragma unroll
       for (int j = 0; j < 4; j++) {
         int col = wc * 64 + j * 16 + fr_row;
-        b_frag[j] = *(const bf16x8*)(Bb + col * BK + ks * 32 + fr_k);
+        int scb = (ks * 4 + fr_q) ^ (col & 7);
+        b_frag[j] = *(const bf16x8*)(Bb + col * BK + scb * 8);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -228,7 +230,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
   int wr = wave >> 1, wc = wave & 1;
 
   int ld_row = lane >> 3;
-  int ld_col = (lane & 7) * 8;
+  int ld_chunk = lane & 7;
   auto stage = [&](int buf, int kt) {
     bf16_t* Ab = lds + buf * 3 * BM * BK;
     bf16_t* Gb = Ab + BM * BK;
@@ -241,13 +243,13 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
       int p = wave + WAVES * call;
       int row = p * 8 + ld_row;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ld_col),
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512), 16, 0, 0);
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ld_col),
+          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Gb + p * 512), 16, 0, 0);
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ld_col),
+          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Ub + p * 512), 16, 0, 0);
     }
   };
@@ -259,7 +261,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
   stage(0, 0);
 
   int fr_row = lane & 15;
-  int fr_k = (lane >> 4) * 8;
+  int fr_q = lane >> 4;  // linear 16B chunk sub-index (0..3) within k-step
 
   for (int kt = 0; kt < ntiles; kt++) {
     int buf = kt & 1;
@@ -280,13 +282,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 #pragma unroll
       for (int i = 0; i < 4; i++) {
         int row = wr * 64 + i * 16 + fr_row;
-        a_frag[i] = *(const bf16x8*)(Ab + row * BK + ks * 32 + fr_k);
+        int sc = (ks * 4 + fr_q) ^ (row & 7);  // bank-conflict-free swizzle
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
       }
 #pragma unroll
       for (int j = 0; j < 4; j++) {
         int col = wc * 64 + j * 16 + fr_row;
-        g_frag[j] = *(const bf16x8*)(Gb + col * BK + ks * 32 + fr_k);
-        u_frag[j] = *(const bf16x8*)(Ub + col * BK + ks * 32 + fr_k);
+        int scb = (ks * 4 + fr_q) ^ (col & 7);
+        g_frag[j] = *(const bf16x8*)(Gb + col * BK + scb * 8);
+        u_frag[j] = *(const bf16x8*)(Ub + col * BK + scb * 8);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
