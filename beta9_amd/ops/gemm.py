@@ -36,7 +36,7 @@ def has_native() -> bool:
     return _load() is not None
 
 
-def gemm_nt(a: torch.Tensor, w: torch.Tensor, xcd_swizzle: bool = True) -> torch.Tensor:
+def gemm_nt(a: torch.Tensor, w: torch.Tensor, xcd_swizzle: bool = False) -> torch.Tensor:
     """C = a @ w.T — a [M,K] bf16, w [N,K] bf16 (torch Linear weight layout).
 
     Hand-written MFMA path for aligned shapes (M%128, N%128, K%64 == 0);
@@ -67,8 +67,8 @@ def mlp_gate_up(a: torch.Tensor, w_gate: torch.Tensor, w_up: torch.Tensor) -> to
             if M % 128 != 0:
                 pad = (128 - M % 128) % 128
                 a2 = torch.nn.functional.pad(a, (0, 0, 0, pad))
-                return ext.gemm_gate_up_silu(a2.contiguous(), w_gate.contiguous(), w_up.contiguous(), True)[:M]
-            return ext.gemm_gate_up_silu(a.contiguous(), w_gate.contiguous(), w_up.contiguous(), True)
+                return ext.gemm_gate_up_silu(a2.contiguous(), w_gate.contiguous(), w_up.contiguous(), False)[:M]
+            return ext.gemm_gate_up_silu(a.contiguous(), w_gate.contiguous(), w_up.contiguous(), False)
     return torch.nn.functional.silu(a @ w_gate.t()) * (a @ w_up.t())
 
 
