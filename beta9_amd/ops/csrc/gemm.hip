@@ -197,6 +197,123 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// 256x256-tile GEMM: 8 waves (2M x 4N, each 128x64 output = 8x4 fragments),
+// BK=64, 128 KiB LDS (1 block/CU), same counted-vmcnt pipeline + conflict-
+// free swizzle. Twice the MFMA work per staged byte of the 128x128 kernel —
+// the step toward the guide's 256^2 deep-pipelined template.
+// ---------------------------------------------------------------------------
+
+constexpr int BM2 = 256, BN2 = 256;
+constexpr int WAVES2 = 8;  // 2M x 4N
+constexpr int THREADS2 = WAVES2 * 64;
+
+template <bool XCD_SWIZZLE>
+__global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ Bt,
+    bf16_t* __restrict__ C, int M, int N, int K) {
+  __shared__ bf16_t lds[2 * 2 * BM2 * BK];  // [buf][A|B][256][64] = 128 KiB
+
+  int nwg_n = N / BN2;
+  int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = (wg / nwg_n) * BM2;
+  int tile_n = (wg % nwg_n) * BN2;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 2, wc = wave & 3;  // wave tile: 128m x 64n
+
+  int ld_row = lane >> 3;
+  int ld_chunk = lane & 7;
+  auto stage = [&](int buf, int kt) {
+    bf16_t* Ab = lds + buf * 2 * BM2 * BK;
+    bf16_t* Bb = Ab + BM2 * BK;
+    const bf16_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
+    const bf16_t* b_src = Bt + (int64_t)(tile_n)*K + kt * BK;
+#pragma unroll
+    for (int call = 0; call < 4; call++) {
+      int p = wave + WAVES2 * call;  // piece 0..31 (each 8 rows)
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(b_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(Bb + p * 512), 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+
+  int ntiles = K / BK;
+  stage(0, 0);
+
+  int fr_row = lane & 15;
+  int fr_q = lane >> 4;
+
+  for (int kt = 0; kt < ntiles; kt++) {
+    int buf = kt & 1;
+    bf16_t* Ab = lds + buf * 2 * BM2 * BK;
+    bf16_t* Bb = Ab + BM2 * BK;
+    if (kt + 1 < ntiles) {
+      stage(buf ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ks++) {
+      bf16x8 a_frag[8], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        int row = wr * 128 + i * 16 + fr_row;
+        int sc = (ks * 4 + fr_q) ^ (row & 7);
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        int col = wc * 64 + j * 16 + fr_row;
+        int scb = (ks * 4 + fr_q) ^ (col & 7);
+        b_frag[j] = *(const bf16x8*)(Bb + col * BK + scb * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 8; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int col = tile_n + wc * 64 + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 128 + i * 16 + c_row0 + r;
+        C[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Fused MLP gate: act = silu(A @ Wg^T) * (A @ Wu^T), bf16.
 // One kernel instead of {gemm, gemm, silu_mul}: the A tile is staged once
 // and feeds BOTH projections' MFMAs; gate/up intermediates never touch HBM
@@ -495,6 +612,23 @@ torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
     hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
                        stream, (const float*)Cpart.data_ptr(),
                        (bf16_t*)C.data_ptr(), mn, (int)splitk);
+    return C;
+  }
+  // 256^2 tile when the shape divides and yields enough blocks to fill the
+  // chip at 1 block/CU (double the MFMA work per staged byte)
+  if (M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 128) {
+    dim3 grid2((M / BM2) * (N / BN2));
+    if (xcd_swizzle) {
+      hipLaunchKernelGGL((gemm_bf16_nt_256_kernel<true>), grid2,
+                         dim3(THREADS2), 0, stream, (const bf16_t*)A.data_ptr(),
+                         (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                         (int)M, (int)N, (int)K);
+    } else {
+      hipLaunchKernelGGL((gemm_bf16_nt_256_kernel<false>), grid2,
+                         dim3(THREADS2), 0, stream, (const bf16_t*)A.data_ptr(),
+                         (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                         (int)M, (int)N, (int)K);
+    }
     return C;
   }
   dim3 grid(tiles);
