@@ -214,8 +214,22 @@ def container_stop(container_id: str):
 
 
 @container_app.command("logs")
-def container_logs(container_id: str, tail: int = 200):
-    for line in _client().call("GatewayService", "GetContainerLogs", {"container_id": container_id, "tail": tail}):
+def container_logs(
+    container_id: str,
+    tail: int = 200,
+    follow: bool = typer.Option(False, "--follow", "-f"),
+):
+    client = _client()
+    if follow:
+        for msg in client.stream(
+            "GatewayService", "StreamContainerLogs", {"container_id": container_id}
+        ):
+            if "line" in msg:
+                typer.echo(msg["line"])
+            if "exit_code" in msg:
+                typer.echo(f"[container exited: {msg['exit_code']}]", err=True)
+        return
+    for line in client.call("GatewayService", "GetContainerLogs", {"container_id": container_id, "tail": tail}):
         typer.echo(line)
 
 

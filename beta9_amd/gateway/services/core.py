@@ -162,6 +162,30 @@ class GatewayService:
     async def GetContainerSpans(self, req: Dict[str, Any], context, auth) -> list:
         return await self.containers.get_spans(req["container_id"])
 
+    @rpc(kind="unary_stream")
+    async def StreamContainerLogs(self, req: Dict[str, Any], context, auth) -> AsyncIterator[Dict]:
+        """Follow a container's log ring until it exits (reference:
+        ContainerStreamLogs)."""
+        import asyncio as _asyncio
+
+        cid = req["container_id"]
+        sent = 0
+        idle = 0.0
+        while True:
+            lines = await self.containers.get_logs(cid, tail=100000)
+            for line in lines[sent:]:
+                yield {"line": line}
+            sent = len(lines)
+            st = await self.containers.get_state(cid)
+            if st is None:
+                code = await self.containers.get_exit_code(cid)
+                yield {"exit_code": code if code is not None else -1}
+                return
+            await _asyncio.sleep(0.2)
+            idle += 0.2
+            if idle > float(req.get("timeout", 3600.0)):
+                return
+
     # -- tasks ----------------------------------------------------------------
     @rpc()
     async def GetTask(self, req: Dict[str, Any], context, auth) -> Optional[Dict]:

@@ -84,6 +84,7 @@ class LocalProcessPoolController:
                     "GPU_TYPE": self.pool.gpu_type,
                     "GPU_COUNT": str(gpu_count if self.pool.gpu_type else 0),
                     "WORKER_RUNTIME": self.pool.runtime,
+                    "B9_IMAGE_STORE": self.cfg.image_service.store_path,
                 }
             )
             proc = subprocess.Popen(

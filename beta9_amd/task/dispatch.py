@@ -128,7 +128,48 @@ class Dispatcher:
     async def complete(self, task_id: str, result: Any, status: TaskStatus = TaskStatus.complete, result_ttl_s: float = 600.0) -> None:
         await self.store.set(keys.task_result(task_id), result, ttl=result_ttl_s)
         await self.store.delete(keys.task_claim(task_id))
+        task = await self.get_task(task_id)
         await self.update_status(task_id, status)
+        if task is not None:
+            await self._fire_callback(task, status, result)
+
+    async def _fire_callback(self, task: Task, status: TaskStatus, result: Any) -> None:
+        """POST the terminal result to the stub's callback_url (reference:
+        end_task_and_send_callback, sdk runner/common.py)."""
+        cfg = self._stub_config_cached(task.stub_id)
+        url = getattr(cfg, "callback_url", "") if cfg else ""
+        if not url:
+            return
+
+        async def post() -> None:
+            try:
+                import httpx
+
+                async with httpx.AsyncClient(timeout=10.0) as client:
+                    await client.post(
+                        url,
+                        json={
+                            "task_id": task.task_id,
+                            "stub_id": task.stub_id,
+                            "status": status.value,
+                            "result": result,
+                        },
+                    )
+            except Exception:
+                log.warning("callback to %s failed", url, exc_info=True)
+
+        asyncio.create_task(post())
+
+    def _stub_config_cached(self, stub_id: str):
+        cache = getattr(self, "_cfg_cache", None)
+        if cache is None:
+            cache = self._cfg_cache = {}
+        if stub_id not in cache:
+            try:
+                cache[stub_id] = self.db.get_stub_config(stub_id)
+            except Exception:
+                cache[stub_id] = None
+        return cache[stub_id]
 
     async def get_result(self, task_id: str) -> Any:
         return await self.store.get(keys.task_result(task_id))

@@ -74,6 +74,7 @@ class Lifecycle:
         gateway_grpc_port: int,
         worker_host: str = "127.0.0.1",
         prewarm=None,  # PrewarmPool
+        image_store_path: str = "",
     ):
         self.worker_id = worker_id
         self.repo = repo
@@ -83,6 +84,7 @@ class Lifecycle:
         self.gateway_grpc_port = gateway_grpc_port
         self.worker_host = worker_host
         self.prewarm = prewarm
+        self.image_store_path = image_store_path
         self.containers: Dict[str, ContainerInstance] = {}
 
     async def run_container(self, request: ContainerRequest) -> int:
@@ -191,6 +193,11 @@ class Lifecycle:
         env.setdefault("CONTAINER_HOSTNAME", request.hostname or request.container_id)
         if request.task_id:
             env.setdefault("TASK_ID", request.task_id)
+        if request.image_id and self.image_store_path:
+            # image prefix injection: site dir onto PYTHONPATH + files tree
+            from ..abstractions.image import image_env
+
+            env.update(image_env(self.image_store_path, request.image_id))
         if needs_addr:
             inst.port = free_port()
             env["BIND_PORT"] = str(inst.port)

@@ -43,6 +43,7 @@ class Worker:
         machine_id: str = "",
         prewarm: bool = True,
         prewarm_cpu_shells: int = 2,
+        image_store_path: str = "",
     ):
         self.id = worker_id
         self.repo = repo
@@ -75,6 +76,8 @@ class Worker:
             self.prewarm_pool = PrewarmPool(
                 gpu_ordinals=ordinals, cpu_shells=prewarm_cpu_shells
             )
+        if not image_store_path:
+            image_store_path = os.environ.get("B9_IMAGE_STORE", "")
         self.lifecycle = Lifecycle(
             worker_id=worker_id,
             repo=repo,
@@ -83,6 +86,7 @@ class Worker:
             gateway_host=gateway_host,
             gateway_grpc_port=gateway_grpc_port,
             prewarm=self.prewarm_pool,
+            image_store_path=image_store_path,
         )
         self._sema = asyncio.Semaphore(container_start_limit)
         self._keepalive_interval_s = keepalive_interval_s

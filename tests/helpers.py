@@ -59,6 +59,7 @@ class Cluster:
                 gpu_manager=gpu_manager,
                 gateway_host=self.cfg.gateway.host,
                 gateway_grpc_port=self.gateway.grpc_port,
+                image_store_path=self.cfg.image_service.store_path,
             )
             await w.start()
             self.workers.append(w)
