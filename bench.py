@@ -68,6 +68,9 @@ async def build_node(tmpdir: str, local_rank: int, args):
         gpu_manager=gpu_manager,
         gateway_host=cfg.gateway.host,
         gateway_grpc_port=gw.grpc_port,
+        # GPU bench: shells for the GPU only (no CPU shells; at N=8 ranks
+        # that is 16 fewer concurrent torch imports), CPU dev box: one
+        prewarm_cpu_shells=0 if (gpu_manager and gpu_manager.total) else 1,
     )
     await worker.start()
     return gw, worker, creds, gpu_manager

@@ -33,6 +33,8 @@ from .sdk import (  # noqa: F401
     schedule,
     task_queue,
 )
+from .schema import Schema  # noqa: F401
+from .sdk import LLMTokenPressureAutoscaler  # noqa: F401
 from .types import TaskPolicy  # noqa: F401
 
 __version__ = "0.1.0"
