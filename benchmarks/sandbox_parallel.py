@@ -27,8 +27,15 @@ sys.path.insert(0, REPO)
 
 
 async def main(n: int, use_gpu: bool, parallelism: int) -> dict:
+    from concurrent.futures import ThreadPoolExecutor
+
     from tests.helpers import Cluster
 
+    # to_thread's default pool is 32 threads; each blocking create() holds
+    # one, so parallelism>32 would starve the exec/terminate calls
+    asyncio.get_event_loop().set_default_executor(
+        ThreadPoolExecutor(max_workers=parallelism * 3 + 8)
+    )
     with tempfile.TemporaryDirectory() as tmp:
         c = Cluster(tmp)
         await c.start(n_workers=1, cpu_per_worker=max(64000, n * 200))
@@ -43,7 +50,7 @@ async def main(n: int, use_gpu: bool, parallelism: int) -> dict:
                 async with sem:
                     sb_def = beta9_amd.Sandbox(cpu=100, memory=192)
                     t0 = time.perf_counter()
-                    sb = await asyncio.to_thread(sb_def.create, 120.0)
+                    sb = await asyncio.to_thread(sb_def.create, 240.0)
                     dt = (time.perf_counter() - t0) * 1000
                     # evidence: unique exec output + distinct container ids
                     marker = f"sb-evidence-{i}-{os.getpid()}"

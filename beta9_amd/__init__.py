@@ -8,33 +8,40 @@ rendezvous provisioning, and a hand-written HIP cold-start path (HBM arena
 pre-alloc + pinned weight streaming + GPU snapshot/restore).
 """
 
-from .sdk import (  # noqa: F401
-    Client,
-    ClientConfig,
-    Endpoint,
-    Function,
-    Image,
-    MapClient,
-    Pod,
-    PodInstance,
-    QueueClient,
-    QueueDepthAutoscaler,
-    Sandbox,
-    SandboxInstance,
-    SignalClient,
-    TaskQueue,
-    Volume,
-    asgi,
-    configure,
-    endpoint,
-    function,
-    get_config,
-    realtime,
-    schedule,
-    task_queue,
-)
-from .schema import Schema  # noqa: F401
-from .sdk import LLMTokenPressureAutoscaler  # noqa: F401
-from .types import TaskPolicy  # noqa: F401
+import os as _os
+
+if _os.environ.get("B9_LIGHT_IMPORT"):
+    # container runners set this: skip the SDK surface so interpreter
+    # startup stays light on the 100-way-parallel cold-start path
+    __all__ = []
+else:
+    from .sdk import (  # noqa: F401
+        Client,
+        ClientConfig,
+        Endpoint,
+        Function,
+        Image,
+        MapClient,
+        Pod,
+        PodInstance,
+        QueueClient,
+        QueueDepthAutoscaler,
+        Sandbox,
+        SandboxInstance,
+        SignalClient,
+        TaskQueue,
+        Volume,
+        asgi,
+        configure,
+        endpoint,
+        function,
+        get_config,
+        realtime,
+        schedule,
+        task_queue,
+    )
+    from .schema import Schema  # noqa: F401
+    from .sdk import LLMTokenPressureAutoscaler  # noqa: F401
+    from .types import TaskPolicy  # noqa: F401
 
 __version__ = "0.1.0"

@@ -191,6 +191,7 @@ class Lifecycle:
         env.setdefault("B9_GATEWAY_HOST", self.gateway_host)
         env.setdefault("B9_GATEWAY_PORT", str(self.gateway_grpc_port))
         env.setdefault("CONTAINER_HOSTNAME", request.hostname or request.container_id)
+        env.setdefault("B9_LIGHT_IMPORT", "1")  # runners skip the SDK surface
         if request.task_id:
             env.setdefault("TASK_ID", request.task_id)
         if request.image_id and self.image_store_path:
