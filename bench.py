@@ -141,6 +141,18 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
             first_load = first["result"]["body"]["result"]["load"]
         except (KeyError, TypeError):
             first_load = {}
+        boot_phases = {}
+        try:
+            addrs = await gw.container_repo.addresses_for_stub(stub_id)
+            if addrs:
+                from beta9_amd.rpc.channel import RpcClient
+
+                rc = RpcClient(list(addrs.values())[0])
+                h = await rc.call("Runner", "Health", {}, timeout=5.0)
+                boot_phases = h.get("gpu_phases", {})
+                await rc.close()
+        except Exception:
+            pass
 
         async def step() -> int:
             results = await asyncio.gather(
@@ -209,6 +221,7 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
             "native_gpu": has_gpu,
             "spans": spans,
             "first_load": first_load,
+            "boot_phases": boot_phases,
         }
 
 
@@ -255,6 +268,7 @@ def main() -> int:
             "cold_start_all_ms": [round(x, 1) for x in cold_sorted],
             "cold_start_phases_ms": r["spans"],
             "first_container_load": r["first_load"],
+            "container_boot_phases_ms": r["boot_phases"],
             "native_gpu_path": r["native_gpu"],
             "config": {
                 "model": f"synthetic-mlp-{args.layers}L-{args.hidden}h-{args.inter}i",

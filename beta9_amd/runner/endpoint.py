@@ -196,13 +196,20 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
                 log.exception("handler error")
                 return {"status_code": 500, "body": {"error": str(e)}}
 
+    boot = {}
+    t0 = time.perf_counter()
     server = RpcServer(host="0.0.0.0", port=env.bind_port)
     server.register_service(RunnerSvc())
     await server.start()
+    boot["rpc_bind_ms"] = round((time.perf_counter() - t0) * 1000, 2)
 
     client = RunnerClient(env)
+    t1 = time.perf_counter()
     await run_on_start(env)
+    boot["on_start_ms"] = round((time.perf_counter() - t1) * 1000, 2)
     state["handler"] = load_callable(env.handler)
+    boot["total_ms"] = round((time.perf_counter() - t0) * 1000, 2)
+    gpu.phases.update(boot)
     state["ready"] = True
     if env.checkpoint_enabled and not gpu.restored:
         try:
