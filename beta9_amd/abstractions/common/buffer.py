@@ -59,16 +59,25 @@ class RequestBuffer:
         self.protocol = protocol
         self.inflight = 0
         self._slots: Dict[str, _ContainerSlot] = {}
-        self._client = httpx.AsyncClient(
-            timeout=httpx.Timeout(request_timeout, connect=5.0),
-            limits=httpx.Limits(max_connections=512, max_keepalive_connections=256),
-        )
+        self._httpx: Optional[httpx.AsyncClient] = None
         # unhealthy->healthy detection granularity; probes are a local gRPC
         # call, so keep this tight — it bounds cold-start detection latency
         self._probe_interval = 0.02
 
+    @property
+    def _client(self) -> httpx.AsyncClient:
+        # lazy: constructing an AsyncClient builds an SSL context (~60 ms);
+        # the grpc dispatch protocol never needs it
+        if self._httpx is None:
+            self._httpx = httpx.AsyncClient(
+                timeout=httpx.Timeout(self.request_timeout, connect=5.0),
+                limits=httpx.Limits(max_connections=512, max_keepalive_connections=256),
+            )
+        return self._httpx
+
     async def close(self) -> None:
-        await self._client.aclose()
+        if self._httpx is not None:
+            await self._httpx.aclose()
         for slot in self._slots.values():
             if slot.rpc is not None:
                 await slot.rpc.close()

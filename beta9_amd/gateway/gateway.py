@@ -42,6 +42,7 @@ class Gateway:
             heartbeat_ttl_s=self.cfg.task.heartbeat_ttl_s,
         )
         self.scheduler = Scheduler(self.cfg, self.store, self.worker_repo, self.container_repo)
+        self.scheduler._gateway = self  # backref for services needing the gateway
         self.rpc_server = RpcServer(
             host=self.cfg.gateway.host, port=self.cfg.gateway.grpc_port, auth_fn=self._auth
         )

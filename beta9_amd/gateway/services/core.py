@@ -121,7 +121,28 @@ class GatewayService:
             raise not_found(f"stub {stub_id}")
         dep = self.db.create_deployment(ws, name, stub_id, stub["type"], app_id=stub["app_id"])
         await self.store.set(f"deploy:announce:{stub_id}", dep)
+        # warm the instance now (reference: InstanceController warms min
+        # containers at deploy) so the first request pays no instance setup
+        kind = stub["type"].split("/")[0]
+        try:
+            if kind in ("endpoint", "asgi"):
+                from ...abstractions.endpoint import get_instance
+
+                inst = await get_instance(self._gateway(), stub_id)
+                inst.instance.autoscaler.trigger()
+            elif kind == "taskqueue":
+                from ...abstractions.taskqueue import get_tq_instance
+
+                inst = await get_tq_instance(self._gateway(), stub_id)
+                inst.instance.autoscaler.trigger()
+        except Exception:
+            pass
         return {"deployment_id": dep["id"], "version": dep["version"], "invoke_url": f"/{stub['type'].split('/')[0]}/id/{stub_id}"}
+
+    def _gateway(self):
+        # the service only holds sub-objects; recover the gateway through
+        # the scheduler backref set at registration
+        return self.scheduler._gateway
 
     @rpc()
     async def GetStubConfig(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:

@@ -210,6 +210,12 @@ class RpcClient:
                 options=[
                     ("grpc.max_send_message_length", 256 * 1024 * 1024),
                     ("grpc.max_receive_message_length", 256 * 1024 * 1024),
+                    # cold-start path: a probe may race the runner's bind;
+                    # the default ~1 s reconnect backoff would become the
+                    # container's observed cold start
+                    ("grpc.initial_reconnect_backoff_ms", 20),
+                    ("grpc.min_reconnect_backoff_ms", 20),
+                    ("grpc.max_reconnect_backoff_ms", 500),
                 ],
             )
 
