@@ -152,10 +152,16 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
                 os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
                 with open(path, "wb") as fh:
                     fh.write(f.get("data", b""))
+                # same-size writes within one mtime second would pass the
+                # bytecode cache's (mtime, size) check — force a newer stamp
+                bump = time.time() + 1.5
+                os.utime(path, (bump, bump))
                 written += 1
             # reload handler module so the next request sees the new code
             try:
                 import importlib
+
+                importlib.invalidate_caches()
 
                 mod_name = env.handler.split(":")[0]
                 import sys as _sys
