@@ -86,6 +86,9 @@ class Lifecycle:
         self.prewarm = prewarm
         self.image_store_path = image_store_path
         self.containers: Dict[str, ContainerInstance] = {}
+        from .usage import UsageTracker
+
+        self.usage = UsageTracker(repo)
 
     async def run_container(self, request: ContainerRequest) -> int:
         cid = request.container_id
@@ -146,6 +149,10 @@ class Lifecycle:
                 inst.started_at = time.time()
 
             await self.repo.update_container_status(cid, ContainerStatus.running.value)
+            self.usage.container_started(
+                cid, gpu=request.gpu, gpu_count=request.gpu_count,
+                cpu=request.cpu, memory_mb=request.memory_mb,
+            )
             if needs_addr:
                 with rec.span("worker.publish_address"):
                     addr = f"{self.worker_host}:{inst.port}"
@@ -169,6 +176,9 @@ class Lifecycle:
             inst.exit_code = 1
             return 1
         finally:
+            usage_rec = self.usage.container_stopped(cid, inst.exit_code)
+            if usage_rec is not None:
+                rec.record("worker.usage", start_total, time.time(), **usage_rec)
             with rec.span("worker.cleanup"):
                 if gpu_assignment is not None and self.gpu is not None:
                     self.gpu.unassign(cid)
