@@ -323,14 +323,20 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256_kernel(
 // (guide §5: glds overlap matters at ~1 block/CU).
 // ---------------------------------------------------------------------------
 
-template <bool XCD_SWIZZLE>
+// BNJ: output-column fragments per wave (4 -> BN=128 tile at 1 block/CU;
+// 2 -> BN=64 tile, 64 KiB LDS, 2 blocks/CU — fills the chip for M=256
+// serving shapes where 128-wide tiles leave CUs idle).
+template <bool XCD_SWIZZLE, int BNJ>
 __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ Wg,
     const bf16_t* __restrict__ Wu, bf16_t* __restrict__ Out, int M, int N,
     int K) {
-  __shared__ bf16_t lds[2 * 3 * BM * BK];  // [buf][A|Wg|Wu][128][64]
+  constexpr int BNW = BNJ * 16;       // wave's output columns
+  constexpr int BNT = BNW * 2;        // block tile columns (2 waves in N)
+  constexpr int WROWS = BNT;          // weight-image rows per tile
+  __shared__ bf16_t lds[2 * (BM + 2 * BNT) * BK];  // [buf][A|Wg|Wu]
 
-  int nwg_n = N / BN;
+  int nwg_n = N / BNT;
   int wg = blockIdx.x;
   if (XCD_SWIZZLE) {
     int nwg = gridDim.x;
@@ -339,7 +345,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
   int tile_m = (wg / nwg_n) * BM;
-  int tile_n = (wg % nwg_n) * BN;
+  int tile_n = (wg % nwg_n) * BNT;
 
   int tid = threadIdx.x;
   int wave = tid >> 6;
@@ -349,12 +355,13 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
   int ld_row = lane >> 3;
   int ld_chunk = lane & 7;
   auto stage = [&](int buf, int kt) {
-    bf16_t* Ab = lds + buf * 3 * BM * BK;
+    bf16_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
     bf16_t* Gb = Ab + BM * BK;
-    bf16_t* Ub = Gb + BM * BK;
+    bf16_t* Ub = Gb + BNT * BK;
     const bf16_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
     const bf16_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK;
     const bf16_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK;
+    // A image: 16 pieces (128 rows); W images: WROWS/8 pieces each
 #pragma unroll
     for (int call = 0; call < 4; call++) {
       int p = wave + WAVES * call;
@@ -362,6 +369,11 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512), 16, 0, 0);
+    }
+#pragma unroll
+    for (int call = 0; call < WROWS / 8 / WAVES; call++) {
+      int p = wave + WAVES * call;
+      int row = p * 8 + ld_row;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(Gb + p * 512), 16, 0, 0);
@@ -371,8 +383,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
     }
   };
 
-  f32x4 acc_g[4][4] = {};
-  f32x4 acc_u[4][4] = {};
+  f32x4 acc_g[4][BNJ] = {};
+  f32x4 acc_u[4][BNJ] = {};
 
   int ntiles = K / BK;
   stage(0, 0);
@@ -382,12 +394,13 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 
   for (int kt = 0; kt < ntiles; kt++) {
     int buf = kt & 1;
-    bf16_t* Ab = lds + buf * 3 * BM * BK;
+    bf16_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
     bf16_t* Gb = Ab + BM * BK;
-    bf16_t* Ub = Gb + BM * BK;
+    bf16_t* Ub = Gb + BNT * BK;
     if (kt + 1 < ntiles) {
       stage(buf ^ 1, kt + 1);
-      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");  // 12 glds per tile
+      constexpr int GLDS_PER_TILE = 4 + 2 * (WROWS / 8 / WAVES);
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(GLDS_PER_TILE) : "memory");
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
@@ -395,7 +408,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 
 #pragma unroll
     for (int ks = 0; ks < 2; ks++) {
-      bf16x8 a_frag[4], g_frag[4], u_frag[4];
+      bf16x8 a_frag[4], g_frag[BNJ], u_frag[BNJ];
 #pragma unroll
       for (int i = 0; i < 4; i++) {
         int row = wr * 64 + i * 16 + fr_row;
@@ -403,8 +416,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
         a_frag[i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
       }
 #pragma unroll
-      for (int j = 0; j < 4; j++) {
-        int col = wc * 64 + j * 16 + fr_row;
+      for (int j = 0; j < BNJ; j++) {
+        int col = wc * BNW + j * 16 + fr_row;
         int scb = (ks * 4 + fr_q) ^ (col & 7);
         g_frag[j] = *(const bf16x8*)(Gb + col * BK + scb * 8);
         u_frag[j] = *(const bf16x8*)(Ub + col * BK + scb * 8);
@@ -413,7 +426,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 #pragma unroll
       for (int i = 0; i < 4; i++)
 #pragma unroll
-        for (int j = 0; j < 4; j++) {
+        for (int j = 0; j < BNJ; j++) {
           acc_g[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], g_frag[j], acc_g[i][j], 0, 0, 0);
           acc_u[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -431,8 +444,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 #pragma unroll
   for (int i = 0; i < 4; i++) {
 #pragma unroll
-    for (int j = 0; j < 4; j++) {
-      int col = tile_n + wc * 64 + j * 16 + c_col;
+    for (int j = 0; j < BNJ; j++) {
+      int col = tile_n + wc * BNW + j * 16 + c_col;
 #pragma unroll
       for (int r = 0; r < 4; r++) {
         int row = tile_m + wr * 64 + i * 16 + c_row0 + r;
@@ -655,16 +668,20 @@ torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(Wu.size(0) == N && Wu.size(1) == K && Wg.size(1) == K);
   TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
   auto Out = torch::empty({M, N}, A.options());
-  dim3 grid((M / BM) * (N / BN));
   auto stream = at::cuda::getCurrentCUDAStream();
-  if (xcd_swizzle) {
-    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<true>), grid, dim3(THREADS),
-                       0, stream, (const bf16_t*)A.data_ptr(),
+  // BN=128 tile when it already fills the chip; BN=64 (2 blocks/CU)
+  // otherwise (M=256 serving shapes)
+  int64_t tiles128 = (M / BM) * (N / 128);
+  if (tiles128 >= 256 || N % 64 != 0) {
+    dim3 grid((M / BM) * (N / 128));
+    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<false, 4>), grid,
+                       dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),
                        (const bf16_t*)Wg.data_ptr(), (const bf16_t*)Wu.data_ptr(),
                        (bf16_t*)Out.data_ptr(), (int)M, (int)N, (int)K);
   } else {
-    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<false>), grid, dim3(THREADS),
-                       0, stream, (const bf16_t*)A.data_ptr(),
+    dim3 grid((M / BM) * (N / 64));
+    hipLaunchKernelGGL((gemm_gate_up_silu_kernel<false, 2>), grid,
+                       dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),
                        (const bf16_t*)Wg.data_ptr(), (const bf16_t*)Wu.data_ptr(),
                        (bf16_t*)Out.data_ptr(), (int)M, (int)N, (int)K);
   }
