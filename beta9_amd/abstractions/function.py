@@ -87,12 +87,22 @@ class FunctionService:
             EXECUTOR, auth["workspace_id"], stub_id, req.get("args"), policy=cfg.task_policy
         )
         yield {"type": "task", "task_id": task.task_id}
+        container_id = f"function-{task.task_id}"
+        stream_logs = bool(req.get("stream_logs", True))
+        sent_logs = 0
         last_status = None
         deadline = time.monotonic() + cfg.timeout + 60.0
+        from ..backend.repos import ContainerRepo
+
         while time.monotonic() < deadline:
             t = await self.dispatcher.get_task(task.task_id)
             if t is None:
                 break
+            if stream_logs:
+                lines = await self.store.lrange(keys.container_logs(container_id), 0, -1)
+                for line in lines[sent_logs:]:
+                    yield {"type": "log", "line": line}
+                sent_logs = len(lines)
             if t.status != last_status:
                 last_status = t.status
                 yield {"type": "status", "status": t.status.value}

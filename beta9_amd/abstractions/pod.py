@@ -166,10 +166,21 @@ class PodService:
 
     @rpc()
     async def SandboxExposePort(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
-        """Single-node: the sandbox PM binds the port directly; return the
-        address."""
-        addr = await self.gw.container_repo.get_address(req["container_id"])
+        """Expose a container port through the gateway TCP ingress
+        (reference: pod tcp.go raw TCP ingress)."""
+        cid = req["container_id"]
+        addr = await self.gw.container_repo.get_address(cid)
         if not addr:
-            raise not_found(f"container {req['container_id']}")
+            raise not_found(f"container {cid}")
         host = addr.split(":")[0]
-        return {"url": f"http://{host}:{req['port']}"}
+        if getattr(self.gw, "tcp_ingress", None) is None:
+            from ..gateway.tcp_ingress import TcpIngress
+
+            self.gw.tcp_ingress = TcpIngress()
+        port = await self.gw.tcp_ingress.expose(cid, host, int(req["port"]))
+        return {
+            "url": f"http://{self.gw.cfg.gateway.host}:{port}",
+            "host": self.gw.cfg.gateway.host,
+            "port": port,
+            "target": f"{host}:{req['port']}",
+        }

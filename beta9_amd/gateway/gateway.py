@@ -301,6 +301,8 @@ class Gateway:
                 await asyncio.wait_for(self._http_task, timeout=5.0)
             except (asyncio.TimeoutError, asyncio.CancelledError):
                 pass
+        if getattr(self, "tcp_ingress", None) is not None:
+            await self.tcp_ingress.shutdown()
         if getattr(self, "cache_transport", None) is not None:
             await self.cache_transport.stop()
         await self.rpc_server.stop()

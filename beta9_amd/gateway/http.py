@@ -78,7 +78,17 @@ def build_http_app(gateway) -> FastAPI:
     async def invoke_asgi(stub_id: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
         return await _invoke(stub_id, request, auth, path="/" + path)
 
-    # -- deployment-name routing (reference: subdomain middleware A.6) -------
+    # -- deployment-name routing (reference: subdomain middleware A.6:
+    # {name}-latest / {name}-v{N} hosts become /deployment/{name}[/v{N}]) ---
+    @app.api_route("/deployment/{name}/v{version}/{path:path}", methods=["GET", "POST"])
+    async def invoke_deployment_version(
+        name: str, version: int, path: str, request: Request, auth: dict = Depends(auth_ctx)
+    ):
+        dep = gateway.db.get_deployment(auth["workspace_id"], name, version=version)
+        if dep is None:
+            raise HTTPException(status_code=404, detail=f"deployment {name} v{version}")
+        return await _invoke(dep["stub_id"], request, auth, path="/" + path)
+
     @app.api_route("/deployment/{name}/{path:path}", methods=["GET", "POST"])
     async def invoke_deployment(name: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
         dep = gateway.db.get_deployment(auth["workspace_id"], name)

@@ -39,6 +39,10 @@ def unpack(data: bytes) -> Any:
     return msgpack.unpackb(data, raw=False, strict_map_key=False)
 
 
+def _identity(data: bytes) -> bytes:
+    return data
+
+
 class RpcError(Exception):
     def __init__(self, code: grpc.StatusCode, message: str):
         super().__init__(message)
@@ -76,14 +80,18 @@ class _GenericService(grpc.GenericRpcHandler):
             return None
         kind, fn = entry
         wrapped = self._wrap(full, kind, fn)
+        # responses are packed INSIDE the wrapper (identity serializer):
+        # grpc.aio str-encodes bare-str responses before a serializer runs,
+        # which would corrupt msgpack str payloads into bins
+        ident = _identity
         if kind == "unary_unary":
-            return grpc.unary_unary_rpc_method_handler(wrapped, unpack, pack)
+            return grpc.unary_unary_rpc_method_handler(wrapped, unpack, ident)
         if kind == "unary_stream":
-            return grpc.unary_stream_rpc_method_handler(wrapped, unpack, pack)
+            return grpc.unary_stream_rpc_method_handler(wrapped, unpack, ident)
         if kind == "stream_unary":
-            return grpc.stream_unary_rpc_method_handler(wrapped, unpack, pack)
+            return grpc.stream_unary_rpc_method_handler(wrapped, unpack, ident)
         if kind == "stream_stream":
-            return grpc.stream_stream_rpc_method_handler(wrapped, unpack, pack)
+            return grpc.stream_stream_rpc_method_handler(wrapped, unpack, ident)
         return None
 
     def _wrap(self, full: str, kind: str, fn: Handler):
@@ -107,7 +115,7 @@ class _GenericService(grpc.GenericRpcHandler):
             async def handler(request, context):
                 auth = await _authctx(context)
                 try:
-                    return await fn(request, context, auth)
+                    return pack(await fn(request, context, auth))
                 except RpcError as e:
                     await context.abort(e.code, e.message)
 
@@ -117,7 +125,7 @@ class _GenericService(grpc.GenericRpcHandler):
             auth = await _authctx(context)
             try:
                 async for item in fn(request, context, auth):
-                    yield item
+                    yield pack(item)
             except RpcError as e:
                 await context.abort(e.code, e.message)
 
