@@ -119,21 +119,29 @@ async def test_sandbox_expose_port_e2e():
         try:
             # run a tiny TCP server inside the sandbox
             code = (
-                "import socket,threading\n"
-                "s=socket.socket(); s.bind((\"0.0.0.0\", 47654)); s.listen(1)\n"
+                "import socket\n"
+                "s=socket.socket(); s.bind((\"0.0.0.0\", 47654)); s.listen(4)\n"
                 "print(\"listening\", flush=True)\n"
-                "c,_=s.accept(); c.sendall(b\"from-sandbox\"); c.close()\n"
+                "while True:\n"
+                "    c,_=s.accept(); c.sendall(b\"from-sandbox\"); c.close()\n"
             )
             await asyncio.to_thread(
                 sb.exec, ["python3", "-c", code], 60.0, True  # background
             )
-            await asyncio.sleep(1.0)
             url = await asyncio.to_thread(sb.expose_port, 47654)
             host, port = url.replace("http://", "").split(":")
-            r, w = await asyncio.open_connection(host, int(port))
-            data = await r.read(100)
+            data = b""
+            for _ in range(50):
+                try:
+                    r, w = await asyncio.open_connection(host, int(port))
+                    data = await r.read(100)
+                    w.close()
+                    if data:
+                        break
+                except OSError:
+                    pass
+                await asyncio.sleep(0.2)
             assert data == b"from-sandbox"
-            w.close()
         finally:
             await asyncio.to_thread(sb.terminate)
 
