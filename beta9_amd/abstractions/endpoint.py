@@ -105,7 +105,7 @@ class EndpointInstance:
             workspace_id=self.stub_row["workspace_id"],
             stub_type=self.stub_row["type"],
             pool_selector=cfg.pool,
-            mounts=list(cfg.volumes),
+            mounts=self.gw.resolve_mounts(self.stub_row["workspace_id"], cfg.volumes),
             checkpoint_enabled=cfg.checkpoint_enabled,
             timestamp=time.time(),
         )

@@ -63,7 +63,7 @@ class FunctionExecutor:
             stub_type=stub["type"],
             task_id=task.task_id,
             pool_selector=cfg.pool,
-            mounts=list(cfg.volumes),
+            mounts=self.scheduler._gateway.resolve_mounts(task.workspace_id, cfg.volumes),
             timestamp=time.time(),
         )
         await self.scheduler.run(request)

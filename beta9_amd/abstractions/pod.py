@@ -54,7 +54,7 @@ class PodService:
             workspace_id=auth["workspace_id"],
             stub_type=stub["type"],
             pool_selector=cfg.pool,
-            mounts=list(cfg.volumes),
+            mounts=self.gw.resolve_mounts(auth["workspace_id"], cfg.volumes),
             ports=list(cfg.ports),
             timestamp=time.time(),
         )

@@ -64,7 +64,7 @@ class TaskQueueInstance:
             workspace_id=self.stub_row["workspace_id"],
             stub_type=self.stub_row["type"],
             pool_selector=cfg.pool,
-            mounts=list(cfg.volumes),
+            mounts=self.gw.resolve_mounts(self.stub_row["workspace_id"], cfg.volumes),
             timestamp=time.time(),
         )
 

@@ -106,6 +106,21 @@ class Gateway:
                 env[name] = val
         return env
 
+    def resolve_mounts(self, workspace_id: str, mounts) -> list:
+        """Volume mounts name host-relative volumes in the SDK; resolve them
+        to absolute host paths for the worker (reference: mounts from
+        volumes, pkg/abstractions/common/mount.go)."""
+        out = []
+        for m in mounts:
+            m = m.model_copy() if hasattr(m, "model_copy") else m
+            local = m.local_path
+            if local and not os.path.isabs(local):
+                path = os.path.join(self.cfg.storage.root, workspace_id, "volumes", local)
+                os.makedirs(path, exist_ok=True)
+                m.local_path = path
+            out.append(m)
+        return out
+
     def _extracted_dir(self, obj: Dict) -> str:
         """Extract an uploaded workspace zip once, content-addressed."""
         import zipfile

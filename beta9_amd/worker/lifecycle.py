@@ -89,6 +89,28 @@ class Lifecycle:
         from .usage import UsageTracker
 
         self.usage = UsageTracker(repo)
+        # native runtime for containers needing mount/network isolation
+        # (volumes bind-mounted into an overlay rootfs; --block-network)
+        self.native_runtime = None
+        if type(runtime).__name__ != "NativeRuntime":
+            try:
+                from .runtime.native import NativeRuntime, namespaces_available
+
+                if namespaces_available():
+                    self.native_runtime = NativeRuntime()
+            except Exception:
+                self.native_runtime = None
+        self._container_runtime: Dict[str, object] = {}
+
+    def _runtime_for(self, spec) -> "Runtime":
+        if self.native_runtime is not None and (spec.mounts or spec.block_network):
+            self._container_runtime[spec.container_id] = self.native_runtime
+            return self.native_runtime
+        self._container_runtime[spec.container_id] = self.runtime
+        return self.runtime
+
+    def runtime_of(self, container_id: str):
+        return self._container_runtime.get(container_id, self.runtime)
 
     async def run_container(self, request: ContainerRequest) -> int:
         cid = request.container_id
@@ -126,9 +148,14 @@ class Lifecycle:
             with rec.span("worker.runtime_start"):
                 on_log = self._make_log_cb(cid)
                 handle = None
+                runtime = self._runtime_for(spec)
                 # Prewarmed shell path: torch+HIP context already up on the
                 # assigned GPU; the shell becomes the container.
-                if self.prewarm is not None and hasattr(self.runtime, "adopt"):
+                if (
+                    self.prewarm is not None
+                    and runtime is self.runtime
+                    and hasattr(runtime, "adopt")
+                ):
                     from .prewarm import runner_module_from_entry
 
                     module = runner_module_from_entry(spec.entry_point)
@@ -137,7 +164,7 @@ class Lifecycle:
                         shell = await self.prewarm.acquire(ordinals)
                         if shell is not None:
                             if await shell.activate(spec.env, module):
-                                handle = await self.runtime.adopt(spec, shell.proc, on_log)
+                                handle = await runtime.adopt(spec, shell.proc, on_log)
                                 rec.record("worker.prewarm_hit", time.time(), time.time())
                             else:
                                 try:
@@ -145,7 +172,7 @@ class Lifecycle:
                                 except ProcessLookupError:
                                     pass
                 if handle is None:
-                    handle = await self.runtime.run(spec, on_log=on_log)
+                    handle = await runtime.run(spec, on_log=on_log)
                 inst.started_at = time.time()
 
             await self.repo.update_container_status(cid, ContainerStatus.running.value)
@@ -162,7 +189,7 @@ class Lifecycle:
 
             stop_watch = asyncio.create_task(self._watch_stop(cid))
             try:
-                exit_code = await self.runtime.wait(cid)
+                exit_code = await runtime.wait(cid)
             finally:
                 stop_watch.cancel()
             inst.exit_code = exit_code
@@ -182,7 +209,8 @@ class Lifecycle:
             with rec.span("worker.cleanup"):
                 if gpu_assignment is not None and self.gpu is not None:
                     self.gpu.unassign(cid)
-                await self.runtime.delete(cid)
+                await self.runtime_of(cid).delete(cid)
+                self._container_runtime.pop(cid, None)
                 if inst.port:
                     await self.repo.remove_container_address(cid, request.stub_id)
                 await self.repo.set_exit_code(cid, inst.exit_code if inst.exit_code is not None else 1)
@@ -224,6 +252,7 @@ class Lifecycle:
             memory_mb=request.memory_mb,
             hostname=request.hostname or request.container_id,
             block_network=request.block_network,
+            mounts=[m.model_dump() for m in request.mounts],
         )
         return spec, needs_addr
 
@@ -241,9 +270,10 @@ class Lifecycle:
                 await asyncio.sleep(0.2)
                 if await self.repo.stop_requested(cid):
                     log.info("stop requested for %s", cid)
-                    await self.runtime.kill(cid, 15)
+                    rt = self.runtime_of(cid)
+                    await rt.kill(cid, 15)
                     await asyncio.sleep(5.0)
-                    await self.runtime.kill(cid, 9)
+                    await rt.kill(cid, 9)
                     return
         except asyncio.CancelledError:
             pass

@@ -186,10 +186,10 @@ class Worker:
         if self.prewarm_pool is not None:
             await self.prewarm_pool.shutdown()
         for cid in list(self.lifecycle.containers):
-            await self.runtime.kill(cid, 15)
+            await self.lifecycle.runtime_of(cid).kill(cid, 15)
         await self.drain(timeout=10.0)
         for cid in list(self.lifecycle.containers):
-            await self.runtime.kill(cid, 9)
+            await self.lifecycle.runtime_of(cid).kill(cid, 9)
         for t in list(self._tasks):
             t.cancel()
         if self._run_task is not None and self._run_task is not asyncio.current_task():
