@@ -48,3 +48,51 @@ async def test_volume_visible_at_mount_path():
         )
         names = {e["name"] for e in ls}
         assert "from-container.txt" in names
+
+
+@pytest.mark.gpu
+@pytest.mark.asyncio
+async def test_gpu_function_in_native_runtime():
+    """GPU container under b9launch: overlay rootfs, minimal /dev with only
+    /dev/kfd + the assigned renderD node, sysfs mounted, volume bound."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    if not namespaces_available():
+        pytest.skip("no namespace privileges")
+    async with cluster(n_workers=0) as c:
+        from beta9_amd.worker.gpu import GPUManager
+        from beta9_amd.worker.repo_client import DirectRepoClient
+        from beta9_amd.worker.worker import Worker
+
+        gm = GPUManager()
+        assert gm.total >= 1
+        repo = DirectRepoClient(c.gateway.worker_repo, c.gateway.container_repo)
+        w = Worker(
+            worker_id="gpunat", repo=repo, cpu_limit=32000, memory_limit_mb=131072,
+            gpu_type=gm.gpu_type, gpu_manager=gm,
+            gateway_host="127.0.0.1", gateway_grpc_port=c.gateway.grpc_port,
+            prewarm=False,
+        )
+        await w.start()
+        try:
+            client = beta9_amd.Client()
+            await asyncio.to_thread(
+                client.client_stream, "VolumeService", "UploadFile",
+                [{"volume_name": "data", "path": "models/weights.txt", "data": b"W" * 16}],
+            )
+            import handlers_volume  # noqa
+
+            fn = beta9_amd.Function(
+                handlers_volume.gpu_in_native,
+                cpu=2000, memory=16384, gpu=gm.gpu_type,
+                volumes=[beta9_amd.Volume(name="data", mount_path="/vol/data")],
+            )
+            r = await asyncio.to_thread(fn.remote)
+            assert r["cuda"] is True, r
+            assert r["kfd"] is True and r["dri"] is True
+            assert r["sum"] == 128.0
+            assert r["vol"] is True
+        finally:
+            await w.shutdown()
