@@ -254,6 +254,35 @@ def deployment_stop(deployment_id: str):
     typer.echo("stopped")
 
 
+@deployment_app.command("rollout")
+def deployment_rollout(name: str, version: int):
+    """Point a deployment at a specific version (rollback/rollforward)."""
+    r = _client().call("GatewayService", "RolloutDeployment", {"name": name, "version": version})
+    typer.echo(json.dumps(r))
+
+
+token_app = typer.Typer(help="workspace tokens")
+app.add_typer(token_app, name="token")
+
+
+@token_app.command("create")
+def token_create(workspace_id: str = typer.Option("", help="defaults to the caller's workspace")):
+    r = _client().call("GatewayService", "CreateToken", {"workspace_id": workspace_id} if workspace_id else {})
+    typer.echo(r["token"])
+
+
+@token_app.command("revoke")
+def token_revoke(token: str):
+    _client().call("GatewayService", "RevokeToken", {"token": token})
+    typer.echo("revoked")
+
+
+@app.command("workspace-create")
+def workspace_create(name: str):
+    r = _client().call("GatewayService", "CreateWorkspace", {"name": name})
+    typer.echo(json.dumps(r))
+
+
 # -- secrets --------------------------------------------------------------------
 
 

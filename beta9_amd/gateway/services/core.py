@@ -55,6 +55,41 @@ class GatewayService:
             return {"ok": False}
         return {"ok": True, "workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
 
+    # -- workspaces / tokens (ops surface) -----------------------------------
+    @rpc()
+    async def CreateWorkspace(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ws = self.db.get_workspace_by_name(req["name"]) or self.db.create_workspace(req["name"])
+        return {"workspace_id": ws["id"], "name": req["name"]}
+
+    @rpc()
+    async def CreateToken(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        ws_id = req.get("workspace_id") or auth["workspace_id"]
+        return {"token": self.db.create_token(ws_id), "workspace_id": ws_id}
+
+    @rpc()
+    async def RevokeToken(self, req: Dict[str, Any], context, auth) -> bool:
+        self.db.revoke_token(req["token"])
+        return True
+
+    @rpc()
+    async def ListApps(self, req: Dict[str, Any], context, auth) -> list:
+        return self.db.list_apps(auth["workspace_id"])
+
+    @rpc()
+    async def RolloutDeployment(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Point a deployment name at a specific existing version
+        (reference: gateway rollout RPCs)."""
+        ws = auth["workspace_id"]
+        name, version = req["name"], int(req["version"])
+        dep = self.db.get_deployment(ws, name, version=version)
+        if dep is None:
+            raise not_found(f"deployment {name} v{version}")
+        self.db._exec(
+            "UPDATE deployments SET active=0 WHERE workspace_id=? AND name=?", (ws, name)
+        )
+        self.db._exec("UPDATE deployments SET active=1 WHERE id=?", (dep["id"],))
+        return {"deployment_id": dep["id"], "version": version, "active": True}
+
     # -- objects (workspace code sync) ---------------------------------------
     @rpc(kind="stream_unary")
     async def PutObjectStream(self, request_iter, context, auth) -> Dict[str, Any]:

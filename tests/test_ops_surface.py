@@ -1,0 +1,65 @@
+"""Ops surface: workspaces/tokens, apps, rollout, worker cordon."""
+
+import asyncio
+
+import pytest
+
+import beta9_amd
+from tests.helpers import cluster
+
+
+@pytest.mark.asyncio
+async def test_workspace_token_lifecycle():
+    async with cluster(n_workers=0) as c:
+        client = beta9_amd.Client()
+        ws = await asyncio.to_thread(client.call, "GatewayService", "CreateWorkspace", {"name": "team-b"})
+        tok = await asyncio.to_thread(
+            client.call, "GatewayService", "CreateToken", {"workspace_id": ws["workspace_id"]}
+        )
+        # new token authorizes
+        c2 = beta9_amd.Client(beta9_amd.ClientConfig(c.grpc_addr, tok["token"]))
+        r = await asyncio.to_thread(c2.call, "GatewayService", "Authorize", {"token": tok["token"]})
+        assert r["ok"] and r["workspace_id"] == ws["workspace_id"]
+        # revoke -> rejected
+        await asyncio.to_thread(client.call, "GatewayService", "RevokeToken", {"token": tok["token"]})
+        import grpc
+
+        with pytest.raises(grpc.RpcError):
+            await asyncio.to_thread(c2.call, "GatewayService", "ListContainers", {})
+        c2.close()
+
+
+@pytest.mark.asyncio
+async def test_rollout_switches_active_version():
+    async with cluster(n_workers=1) as c:
+        import handlers
+
+        fn = beta9_amd.Function(handlers.hello, cpu=100, memory=128)
+        d1 = await asyncio.to_thread(fn.deploy, "app-r")
+        d2 = await asyncio.to_thread(fn.deploy, "app-r")
+        client = beta9_amd.Client()
+        r = await asyncio.to_thread(
+            client.call, "GatewayService", "RolloutDeployment", {"name": "app-r", "version": d1["version"]}
+        )
+        assert r["version"] == d1["version"]
+        dep = c.gateway.db.get_deployment(c.creds["workspace_id"], "app-r")
+        assert dep["version"] == d1["version"]
+
+        apps = await asyncio.to_thread(client.call, "GatewayService", "ListApps", {})
+        assert any(a["name"] == "hello" for a in apps) or len(apps) >= 1
+
+
+@pytest.mark.asyncio
+async def test_cordoned_worker_excluded():
+    async with cluster(n_workers=1) as c:
+        client = beta9_amd.Client()
+        workers = await asyncio.to_thread(client.call, "GatewayService", "ListWorkers", {})
+        wid = workers[0]["id"]
+        await asyncio.to_thread(client.call, "GatewayService", "CordonWorker", {"worker_id": wid})
+        live = await c.gateway.scheduler._live_workers()
+        assert all(w.id != wid for w in live)
+        await asyncio.to_thread(
+            client.call, "GatewayService", "CordonWorker", {"worker_id": wid, "cordoned": False}
+        )
+        live = await c.gateway.scheduler._live_workers()
+        assert any(w.id == wid for w in live)
