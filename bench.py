@@ -58,7 +58,7 @@ async def build_node(tmpdir: str, local_rank: int, args):
     devices = discover_devices(cfg.gpu.kfd_topology_path) or _torch_fallback_devices()
     mine = [d for d in devices if d.ordinal == local_rank]
     gpu_manager = GPUManager(devices=mine) if mine else None
-    repo = DirectRepoClient(gw.worker_repo, gw.container_repo)
+    repo = DirectRepoClient(gw.worker_repo, gw.container_repo, events=gw.events)
     worker = Worker(
         worker_id=f"bench-w{local_rank}",
         repo=repo,
@@ -230,7 +230,7 @@ def main() -> int:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--concurrency", type=int, default=8, help="requests per rank per step")
+    p.add_argument("--concurrency", type=int, default=16, help="requests per rank per step")
     p.add_argument("--tokens", type=int, default=256)
     p.add_argument("--hidden", type=int, default=4096)
     p.add_argument("--inter", type=int, default=14336)

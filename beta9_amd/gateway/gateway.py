@@ -222,7 +222,7 @@ class Gateway:
             return True
 
         self.scheduler.concurrency_check = concurrency_check
-        repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo)
+        repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo, events=self.events)
 
         fn_exec = FunctionExecutor(self.db, self.scheduler, self.store, self.runner_env)
         self.dispatcher.register(FN_EXEC, fn_exec.execute)

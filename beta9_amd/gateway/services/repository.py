@@ -16,9 +16,10 @@ from ...types import ContainerRequest, ContainerState, ContainerStatus, Worker
 class WorkerRepositoryService:
     service_name = "WorkerRepositoryService"
 
-    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo):
+    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo, events=None):
         self._w = worker_repo
         self._c = container_repo
+        self._events = events
 
     @rpc()
     async def RegisterWorker(self, req: Dict[str, Any], context, auth) -> bool:
@@ -102,4 +103,12 @@ class WorkerRepositoryService:
     @rpc()
     async def PushSpan(self, req: Dict[str, Any], context, auth) -> bool:
         await self._c.push_span(req["container_id"], req["span"])
+        return True
+
+    @rpc()
+    async def PushEvent(self, req: Dict[str, Any], context, auth) -> bool:
+        """Worker-originated lifecycle events relayed through the gateway
+        (reference: pkg/repository/events_worker.go)."""
+        if self._events is not None:
+            await self._events.emit(req["name"], **(req.get("payload") or {}))
         return True

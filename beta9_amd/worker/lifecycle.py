@@ -176,6 +176,13 @@ class Lifecycle:
                 inst.started_at = time.time()
 
             await self.repo.update_container_status(cid, ContainerStatus.running.value)
+            try:
+                await self.repo.push_event(
+                    "container.started", container_id=cid, stub_id=request.stub_id,
+                    worker_id=self.worker_id, gpu=request.gpu, gpu_count=request.gpu_count,
+                )
+            except Exception:
+                pass
             self.usage.container_started(
                 cid, gpu=request.gpu, gpu_count=request.gpu_count,
                 cpu=request.cpu, memory_mb=request.memory_mb,
@@ -206,6 +213,15 @@ class Lifecycle:
             usage_rec = self.usage.container_stopped(cid, inst.exit_code)
             if usage_rec is not None:
                 rec.record("worker.usage", start_total, time.time(), **usage_rec)
+            try:
+                name = "container.oom" if inst.exit_code == 137 else "container.stopped"
+                await self.repo.push_event(
+                    name, container_id=cid, stub_id=request.stub_id,
+                    worker_id=self.worker_id, exit_code=inst.exit_code,
+                    duration_s=(usage_rec or {}).get("duration_s"),
+                )
+            except Exception:
+                pass
             with rec.span("worker.cleanup"):
                 if gpu_assignment is not None and self.gpu is not None:
                     self.gpu.unassign(cid)

@@ -20,9 +20,10 @@ from ..types import ContainerRequest, ContainerState, ContainerStatus, Worker
 
 
 class DirectRepoClient:
-    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo):
+    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo, events=None):
         self._w = worker_repo
         self._c = container_repo
+        self._events = events
 
     # worker state
     async def register_worker(self, worker: Worker) -> None:
@@ -76,6 +77,10 @@ class DirectRepoClient:
 
     async def push_span(self, container_id: str, span: Dict[str, Any]) -> None:
         await self._c.push_span(container_id, span)
+
+    async def push_event(self, name: str, **payload: Any) -> None:
+        if self._events is not None:
+            await self._events.emit(name, **payload)
 
 
 SVC = "WorkerRepositoryService"
@@ -151,3 +156,6 @@ class RemoteRepoClient:
 
     async def push_span(self, container_id: str, span: Dict[str, Any]) -> None:
         await self._rpc.call(SVC, "PushSpan", {"container_id": container_id, "span": span})
+
+    async def push_event(self, name: str, **payload: Any) -> None:
+        await self._rpc.call(SVC, "PushEvent", {"name": name, "payload": payload})

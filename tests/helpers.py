@@ -40,7 +40,7 @@ class Cluster:
         await self.gateway.start()
         self.gateway.scheduler.register_pool(FakePoolController())
         self.creds = self.gateway.bootstrap_workspace("test")
-        repo = DirectRepoClient(self.gateway.worker_repo, self.gateway.container_repo)
+        repo = DirectRepoClient(self.gateway.worker_repo, self.gateway.container_repo, events=self.gateway.events)
         for i in range(n_workers):
             gpu_manager: Optional[GPUManager] = None
             gpu_type = ""
