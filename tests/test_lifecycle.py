@@ -17,12 +17,16 @@ class FakeRepo:
     def __init__(self):
         self.spans = []
         self.logs = []
+        self.events = []
         self.addresses = {}
         self.exit_codes = {}
         self.stops = set()
 
     async def push_span(self, cid, span):
         self.spans.append(span)
+
+    async def push_event(self, name, **payload):
+        self.events.append({"name": name, **payload})
 
     async def append_log(self, cid, line):
         self.logs.append(line)
@@ -144,7 +148,13 @@ async def test_checkpoint_env_attach(tmp_path):
         entry_point=["sh", "-c", "echo ck=$B9_CHECKPOINT_KEY"],
     )
     await lc.run_container(req)
+    for _ in range(50):
+        if any("ck=ws-stub" in l for l in lc.repo.logs):
+            break
+        await asyncio.sleep(0.02)
     assert any("ck=ws-stub" in l for l in lc.repo.logs)
+    assert any(e["name"] == "container.started" for e in lc.repo.events)
+    assert any(e["name"] == "container.stopped" for e in lc.repo.events)
 
 
 @pytest.mark.asyncio
