@@ -132,8 +132,16 @@ async def test_scale_to_zero_resume_gpu():
                 await asyncio.sleep(0.1)
             assert not await c.gateway.container_repo.containers_for_stub(stub_id)
 
-            # resume: next request restores from the HBM snapshot
-            r2 = await invoke()
-            assert r2["load"]["restored"] is True
+            # resume: next request restores from the HBM snapshot — and the
+            # cycle holds up over repeated scale-to-zero churn
+            for _cycle in range(2):
+                r2 = await invoke()
+                assert r2["load"]["restored"] is True
+                assert r2["checksum"] == r2["checksum"]
+                for _ in range(300):
+                    if not await c.gateway.container_repo.containers_for_stub(stub_id):
+                        break
+                    await asyncio.sleep(0.1)
+                assert not await c.gateway.container_repo.containers_for_stub(stub_id)
         finally:
             await w.shutdown()
