@@ -68,6 +68,25 @@ def test_gateway_process_with_provisioned_workers(tmp_path):
         workers = client.call("GatewayService", "ListWorkers", {})
         assert len(workers) >= 1
         client.close()
+
+        # the CLI works against the same gateway (context written by
+        # `b9 gateway` bootstrap into $HOME/.b9/config.yaml)
+        cli = subprocess.run(
+            [sys.executable, "-m", "beta9_amd.cli", "task", "list"],
+            env=env,
+            cwd=REPO,
+            capture_output=True,
+            text=True,
+            timeout=60,
+        )
+        assert cli.returncode == 0, cli.stderr[-500:]
+        assert "task-" in cli.stdout
+
+        cli = subprocess.run(
+            [sys.executable, "-m", "beta9_amd.cli", "worker", "list"],
+            env=env, cwd=REPO, capture_output=True, text=True, timeout=60,
+        )
+        assert cli.returncode == 0 and '"id"' in cli.stdout
     finally:
         try:
             os.killpg(proc.pid, signal.SIGTERM)

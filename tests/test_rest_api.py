@@ -1,0 +1,54 @@
+"""REST /api/v1 surface + prometheus endpoint."""
+
+import asyncio
+
+import httpx
+import pytest
+
+import beta9_amd
+from tests.helpers import cluster
+
+
+@pytest.mark.asyncio
+async def test_rest_surface():
+    async with cluster(n_workers=1) as c:
+        base = f"http://127.0.0.1:{c.gateway.http_port}"
+        headers = {"authorization": f"Bearer {c.token}"}
+        async with httpx.AsyncClient() as hc:
+            r = await hc.get(f"{base}/api/v1/health")
+            assert r.status_code == 200 and r.json()["status"] == "ok"
+
+            # unauthorized requests rejected
+            r = await hc.get(f"{base}/api/v1/workers")
+            assert r.status_code == 401
+
+            r = await hc.get(f"{base}/api/v1/workers", headers=headers)
+            assert r.status_code == 200 and len(r.json()) == 1
+
+            # run a function so tasks/containers/events/metrics have content
+            import handlers
+
+            fn = beta9_amd.Function(handlers.hello, cpu=100, memory=128)
+            await asyncio.to_thread(fn.remote, 2)
+
+            r = await hc.get(f"{base}/api/v1/tasks", headers=headers)
+            assert r.status_code == 200
+            c.gateway.db.flush()
+            r = await hc.get(f"{base}/api/v1/tasks", headers=headers)
+            assert len(r.json()) >= 1
+
+            r = await hc.get(f"{base}/api/v1/metrics", headers=headers)
+            body = r.json()
+            assert "scheduler" in body and body["scheduler"]["scheduled"] >= 1
+
+            r = await hc.get(f"{base}/metrics")
+            assert r.status_code == 200
+            assert "scheduler_containers_scheduled" in r.text
+
+            r = await hc.get(f"{base}/api/v1/events", headers=headers)
+            names = {e["name"] for e in r.json()}
+            assert "container.scheduled" in names
+            assert "container.started" in names
+
+            r = await hc.get(f"{base}/api/v1/deployments", headers=headers)
+            assert r.status_code == 200
