@@ -96,6 +96,7 @@ async def deploy_endpoint(gw, creds, args, has_gpu: bool):
             "B9_BENCH_HIDDEN": str(args.hidden),
             "B9_BENCH_INTER": str(args.inter),
             "B9_BENCH_LAYERS": str(args.layers),
+            "B9_BENCH_MODEL": args.model,
         },
         autoscaler=__import__("beta9_amd.types", fromlist=["AutoscalerConfig"]).AutoscalerConfig(
             max_containers=1, tasks_per_container=max(32, args.concurrency * 2)
@@ -235,6 +236,7 @@ def main() -> int:
     p.add_argument("--hidden", type=int, default=4096)
     p.add_argument("--inter", type=int, default=14336)
     p.add_argument("--layers", type=int, default=8)
+    p.add_argument("--model", choices=["mlp", "transformer"], default="mlp")
     args = p.parse_args()
 
     rank, world, local_rank = get_dist()
@@ -271,7 +273,7 @@ def main() -> int:
             "container_boot_phases_ms": r["boot_phases"],
             "native_gpu_path": r["native_gpu"],
             "config": {
-                "model": f"synthetic-mlp-{args.layers}L-{args.hidden}h-{args.inter}i",
+                "model": f"synthetic-{args.model}-{args.layers}L-{args.hidden}h-{args.inter}i",
                 "global_batch": args.concurrency * (world if world > 1 else 1),
                 "seq_len": args.tokens,
                 "parallelism": f"serving dp{world if world > 1 else 1} (1 GPU-container per rank)",

@@ -52,13 +52,21 @@ def load() -> Dict[str, Any]:
             "backend": rdzv.backend,
         }
         return _load_info
+    model_kind = os.environ.get("B9_BENCH_MODEL", "mlp")
     if torch.cuda.is_available():
         from ..runner.common import get_gpu_boot
         from .mlp import SyntheticServingModel
 
         gb = get_gpu_boot()
         restored = False
-        if gb is not None and gb.restored and gb.arena is not None:
+        if model_kind == "transformer":
+            from .transformer import TransformerServingModel
+
+            _model = TransformerServingModel(
+                hidden=hidden, inter=inter, layers=layers, device="cuda:0",
+                arena=(gb.arena if gb is not None else None),
+            )
+        elif gb is not None and gb.restored and gb.arena is not None:
             # scale-to-zero resume: weights already streamed into HBM from
             # the hipsnap snapshot during GPU boot — just remap views.
             _model = SyntheticServingModel.from_arena(gb.arena, hidden, inter, layers)
