@@ -22,6 +22,7 @@ else:
         Function,
         Image,
         MapClient,
+        Output,
         Pod,
         PodInstance,
         QueueClient,

@@ -379,11 +379,7 @@ def shell(container_id: str):
 @app.command()
 def metrics():
     """Control-plane metrics snapshot."""
-    import httpx
-
-    ctx = _load_context()
-    # metrics live on the REST surface; fall back to gRPC-side worker list
-    typer.echo(json.dumps(_client().call("GatewayService", "ListWorkers", {}), indent=2, default=str))
+    typer.echo(json.dumps(_client().call("GatewayService", "GetMetrics", {}), indent=2, default=str))
 
 
 def main():

@@ -288,6 +288,15 @@ class GatewayService:
 
     # -- workers (ops surface) -------------------------------------------------
     @rpc()
+    async def GetMetrics(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        gw = self._gateway()
+        return {
+            "scheduler": gw.scheduler.metrics.snapshot(),
+            "backlog": await gw.scheduler.backlog_len(),
+            "registry": gw.metrics.snapshot() if getattr(gw, "metrics", None) else {},
+        }
+
+    @rpc()
     async def ListWorkers(self, req: Dict[str, Any], context, auth) -> list:
         return [w.model_dump() for w in await self.workers.list()]
 

@@ -773,6 +773,32 @@ class QueueClient:
         return self.client.call("SimpleQueueService", "SimpleQueueSize", {"name": self.name})
 
 
+class Output:
+    """Task output files: save -> retrieval URL (reference:
+    pkg/abstractions/output — SDK Output surface)."""
+
+    def __init__(self, task_id: str = "", client: Optional[Client] = None):
+        self.task_id = task_id or os.environ.get("TASK_ID", "")
+        self.client = client or Client()
+
+    def save(self, data: bytes, filename: str) -> Dict[str, Any]:
+        chunks = [
+            {"task_id": self.task_id, "filename": filename, "data": data[i : i + (1 << 20)]}
+            for i in range(0, max(len(data), 1), 1 << 20)
+        ] or [{"task_id": self.task_id, "filename": filename, "data": b""}]
+        return self.client.client_stream("OutputService", "OutputSaveStream", chunks)
+
+    def save_file(self, path: str, filename: str = "") -> Dict[str, Any]:
+        with open(path, "rb") as f:
+            return self.save(f.read(), filename or os.path.basename(path))
+
+    def public_url(self, path: str, ttl: float = 3600.0) -> str:
+        return self.client.call("OutputService", "OutputPublicURL", {"path": path, "ttl": ttl})["url"]
+
+    def stat(self, path: str) -> Dict[str, Any]:
+        return self.client.call("OutputService", "OutputStat", {"path": path})
+
+
 class SignalClient:
     def __init__(self, name: str, client: Optional[Client] = None):
         self.name = name

@@ -12,7 +12,7 @@ MI355X specifics vs the reference: GPU assignment is xGMI-topology-aware
 (gpu.py), the runner pre-initializes HIP + pre-allocates its HBM arena before
 handler import (runner/common.py + ops hipsnap), and scale-to-zero restore
 streams an HBM snapshot through pinned staging instead of CRIU
-(worker/snapshot.py).
+(ops/hipsnap.py + the grant/commit flow in gateway/services/core.py).
 """
 
 from __future__ import annotations

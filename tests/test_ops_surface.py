@@ -63,3 +63,23 @@ async def test_cordoned_worker_excluded():
         )
         live = await c.gateway.scheduler._live_workers()
         assert any(w.id == wid for w in live)
+
+
+@pytest.mark.asyncio
+async def test_output_sdk_roundtrip():
+    async with cluster(n_workers=0) as c:
+        out = beta9_amd.Output(task_id="task-x")
+        r = await asyncio.to_thread(out.save, b"result-bytes", "result.bin")
+        assert r["path"].endswith("result.bin")
+        st = await asyncio.to_thread(out.stat, r["path"])
+        assert st["exists"] and st["size"] == len(b"result-bytes")
+        url = await asyncio.to_thread(out.public_url, r["path"])
+        assert url.startswith("/api/v1/outputs/")
+        import httpx
+
+        async with httpx.AsyncClient() as hc:
+            resp = await hc.get(
+                f"http://127.0.0.1:{c.gateway.http_port}{url}",
+                headers={"authorization": f"Bearer {c.token}"},
+            )
+            assert resp.status_code == 200 and resp.content == b"result-bytes"
