@@ -5,7 +5,7 @@ python setup.py build_ext --inplace   (PYTORCH_ROCM_ARCH=gfx950)
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -16,7 +16,10 @@ CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "beta9_amd", "op
 ext_flags = ["-O3", "-std=c++17"]
 
 setup(
-    name="beta9_amd_ops",
+    name="beta9-amd",
+    version="0.1.0",
+    packages=find_packages(include=["beta9_amd", "beta9_amd.*"]),
+    entry_points={"console_scripts": ["b9 = beta9_amd.cli.main:main"]},
     ext_modules=[
         CUDAExtension(
             name="beta9_amd.ops._hipsnap_ext",
