@@ -92,8 +92,9 @@ class PrewarmPool:
             return
         self._n += 1
         sock_path = os.path.join(self._dir, f"shell-{self._n}.sock")
-        env = dict(os.environ)
-        env.update(self.base_env)
+        from .runtime.base import container_base_env
+
+        env = container_base_env(self.base_env)
         env["B9_PREWARM_SOCK"] = sock_path
         if key:
             vis = ",".join(str(o) for o in key)

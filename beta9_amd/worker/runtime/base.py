@@ -90,3 +90,34 @@ class Runtime(abc.ABC):
 # Exit code the OOM watcher / killers normalize to (reference:
 # pkg/worker/lifecycle.go:1927-1973 exit-code normalization).
 EXIT_OOM = 137
+
+# Distributed-launcher env that must NEVER leak from the worker process into
+# containers (a torchrun-launched worker would otherwise make every endpoint
+# runner think it is rank N of a world): the platform sets these itself for
+# gpu_count>1 containers (worker/gpu.py provision_rendezvous).
+_DIST_ENV = (
+    "RANK",
+    "LOCAL_RANK",
+    "WORLD_SIZE",
+    "LOCAL_WORLD_SIZE",
+    "GROUP_RANK",
+    "GROUP_WORLD_SIZE",
+    "ROLE_RANK",
+    "ROLE_WORLD_SIZE",
+    "ROLE_NAME",
+    "MASTER_ADDR",
+    "MASTER_PORT",
+    "OMP_NUM_THREADS",
+)
+
+
+def container_base_env(spec_env: Dict[str, str]) -> Dict[str, str]:
+    """os.environ minus launcher-distributed vars, overlaid with the spec."""
+    import os as _os
+
+    env = dict(_os.environ)
+    for k in list(env):
+        if k in _DIST_ENV or k.startswith("TORCHELASTIC_") or k.startswith("PET_"):
+            del env[k]
+    env.update(spec_env)
+    return env

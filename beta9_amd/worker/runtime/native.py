@@ -92,8 +92,9 @@ class NativeRuntime(ProcRuntime):
             argv += ["--cwd", spec.cwd]
         argv += ["--"] + list(spec.entry_point)
 
-        env = dict(os.environ)
-        env.update(spec.env)
+        from .base import container_base_env
+
+        env = container_base_env(spec.env)
         proc = await asyncio.create_subprocess_exec(
             *argv,
             env=env,

@@ -77,8 +77,9 @@ class ProcRuntime(Runtime):
             return ""
 
     async def run(self, spec: ContainerSpec, on_log: Optional[LogCallback] = None) -> ContainerHandle:
-        env = dict(os.environ)
-        env.update(spec.env)
+        from .base import container_base_env
+
+        env = container_base_env(spec.env)
         cg = self._setup_cgroup(spec)
         proc = await asyncio.create_subprocess_exec(
             *spec.entry_point,
