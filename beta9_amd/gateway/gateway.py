@@ -54,6 +54,7 @@ class Gateway:
         self._admin_token: str = ""
         self._runner_tokens: Dict[str, str] = {}  # workspace_id -> token
         self._started = False
+        self.draining = False  # reference: drainMiddleware + 10 s propagation
 
     # -- auth ----------------------------------------------------------------
     async def _auth(self, token: str, full_method: str):
@@ -301,9 +302,24 @@ class Gateway:
         except Exception:
             log.exception("http server failed to start; gRPC-only mode")
 
+    async def drain(self, timeout: float = 10.0) -> None:
+        """Stop accepting new invokes (503), let inflight requests finish
+        (reference: gateway.go:99-101 draining flag + propagation window)."""
+        self.draining = True
+        deadline = asyncio.get_event_loop().time() + timeout
+        while asyncio.get_event_loop().time() < deadline:
+            inflight = sum(
+                getattr(getattr(inst, "buffer", None), "inflight", 0)
+                for inst in self._instances.values()
+            )
+            if inflight == 0:
+                return
+            await asyncio.sleep(0.1)
+
     async def shutdown(self) -> None:
         if not self._started:
             return
+        await self.drain(timeout=self.cfg.gateway.shutdown_timeout_s)
         for loop in getattr(self, "_loops", []):
             await loop.stop()
         for inst in list(self._instances.values()):

@@ -37,6 +37,8 @@ def build_http_app(gateway) -> FastAPI:
 
     # -- invoke routes ------------------------------------------------------
     async def _invoke(stub_id: str, request: Request, auth: dict, path: str = "/"):
+        if gateway.draining:
+            return JSONResponse({"error": "gateway draining"}, status_code=503)
         from ..abstractions.endpoint import get_instance
 
         try:

@@ -83,3 +83,21 @@ async def test_output_sdk_roundtrip():
                 headers={"authorization": f"Bearer {c.token}"},
             )
             assert resp.status_code == 200 and resp.content == b"result-bytes"
+
+
+@pytest.mark.asyncio
+async def test_gateway_drain():
+    async with cluster(n_workers=1) as c:
+        import httpx
+
+        c.gateway.draining = True
+        async with httpx.AsyncClient() as hc:
+            r = await hc.post(
+                f"http://127.0.0.1:{c.gateway.http_port}/endpoint/id/whatever",
+                json={},
+                headers={"authorization": f"Bearer {c.token}"},
+            )
+            assert r.status_code == 503
+        c.gateway.draining = False
+        await c.gateway.drain(timeout=1.0)
+        assert c.gateway.draining is True
