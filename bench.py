@@ -231,7 +231,7 @@ def main() -> int:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--concurrency", type=int, default=16, help="requests per rank per step")
+    p.add_argument("--concurrency", type=int, default=32, help="requests per rank per step")
     p.add_argument("--tokens", type=int, default=256)
     p.add_argument("--hidden", type=int, default=4096)
     p.add_argument("--inter", type=int, default=14336)
