@@ -16,6 +16,15 @@ from .runtime.proc import ProcRuntime
 from .worker import Worker
 
 
+def _select_runtime(name: str):
+    if name == "native":
+        from .runtime.native import NativeRuntime, namespaces_available
+
+        if namespaces_available():
+            return NativeRuntime()
+    return ProcRuntime()
+
+
 async def amain() -> int:
     logging.basicConfig(level=logging.INFO)
     e = os.environ
@@ -36,7 +45,7 @@ async def amain() -> int:
         gpu_type=gpu_type,
         gpu_count=int(e.get("GPU_COUNT", "0") or 0),
         pool_name=e.get("WORKER_POOL_NAME", "default"),
-        runtime=ProcRuntime(),
+        runtime=_select_runtime(e.get("WORKER_RUNTIME", "proc")),
         gpu_manager=gpu_manager,
         gateway_host=host,
         gateway_grpc_port=int(port or 1993),
