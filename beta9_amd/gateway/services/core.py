@@ -301,6 +301,28 @@ class GatewayService:
         return [w.model_dump() for w in await self.workers.list()]
 
     @rpc()
+    async def DrainWorker(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Cordon + wait for the worker's containers to finish, then remove
+        (reference: workers cordon-drain surface)."""
+        import asyncio as _asyncio
+
+        w = await self.workers.get(req["worker_id"])
+        if w is None:
+            raise not_found(req["worker_id"])
+        w.cordoned = True
+        await self.workers.update(w)
+        deadline = time.time() + float(req.get("timeout", 60.0))
+        while time.time() < deadline:
+            w = await self.workers.get(req["worker_id"])
+            if w is None or w.active_containers == 0:
+                break
+            await _asyncio.sleep(0.25)
+        remaining = w.active_containers if w else 0
+        if req.get("remove", False) and remaining == 0:
+            await self.workers.remove(req["worker_id"])
+        return {"drained": remaining == 0, "active_containers": remaining}
+
+    @rpc()
     async def CordonWorker(self, req: Dict[str, Any], context, auth) -> bool:
         w = await self.workers.get(req["worker_id"])
         if w is None:

@@ -101,3 +101,17 @@ async def test_gateway_drain():
         c.gateway.draining = False
         await c.gateway.drain(timeout=1.0)
         assert c.gateway.draining is True
+
+
+@pytest.mark.asyncio
+async def test_drain_worker():
+    async with cluster(n_workers=1) as c:
+        client = beta9_amd.Client()
+        workers = await asyncio.to_thread(client.call, "GatewayService", "ListWorkers", {})
+        wid = workers[0]["id"]
+        r = await asyncio.to_thread(
+            client.call, "GatewayService", "DrainWorker", {"worker_id": wid, "timeout": 5.0}
+        )
+        assert r["drained"] is True
+        w = await c.gateway.worker_repo.get(wid)
+        assert w.cordoned is True
