@@ -62,7 +62,7 @@ class RequestBuffer:
         self._httpx: Optional[httpx.AsyncClient] = None
         # unhealthy->healthy detection granularity; probes are a local gRPC
         # call, so keep this tight — it bounds cold-start detection latency
-        self._probe_interval = 0.02
+        self._probe_interval = 0.005
 
     @property
     def _client(self) -> httpx.AsyncClient:

@@ -89,6 +89,10 @@ class WorkerConfig(BaseModel):
 
 
 class SchedulerConfig(BaseModel):
+    # GPU failover chains: requested GPU -> ranked alternatives appended to
+    # the preference list when the primary has no capacity (reference:
+    # scheduling.failover config + pkg/scheduler/failover.go)
+    gpu_failover: Dict[str, List[str]] = Field(default_factory=dict)
     backlog_poll_interval_s: float = 0.01
     batch_size: int = 10
     max_retries: int = 120

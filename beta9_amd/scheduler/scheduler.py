@@ -158,6 +158,11 @@ class Scheduler:
             request.gpu_request = [request.gpu]
         if GPU_TYPE_ANY in request.gpu_request:
             request.gpu_request = [g for g in GPU_PRIORITY_ORDER if g != GPU_TYPE_ANY]
+        # config-driven failover chains: append ranked alternatives
+        for gpu in list(request.gpu_request):
+            for alt in self.cfg.scheduler.gpu_failover.get(gpu, []):
+                if alt not in request.gpu_request:
+                    request.gpu_request.append(alt)
         if request.gpu_request and request.gpu_count == 0:
             request.gpu_count = 1
         if request.gpu_request:

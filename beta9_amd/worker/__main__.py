@@ -51,6 +51,9 @@ async def amain() -> int:
         gateway_grpc_port=int(port or 1993),
         idle_shutdown_s=float(e.get("IDLE_SHUTDOWN_S", "0") or 0),
         machine_id=e.get("MACHINE_ID", ""),
+        # multi-node: the address other nodes reach this worker's containers
+        # at (container addresses are published as WORKER_HOST:port)
+        worker_host=e.get("WORKER_HOST", "127.0.0.1"),
     )
     stop = asyncio.Event()
     loop = asyncio.get_event_loop()

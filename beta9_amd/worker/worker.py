@@ -44,6 +44,7 @@ class Worker:
         prewarm: bool = True,
         prewarm_cpu_shells: int = 2,
         image_store_path: str = "",
+        worker_host: str = "127.0.0.1",
     ):
         self.id = worker_id
         self.repo = repo
@@ -87,6 +88,7 @@ class Worker:
             gateway_grpc_port=gateway_grpc_port,
             prewarm=self.prewarm_pool,
             image_store_path=image_store_path,
+            worker_host=worker_host,
         )
         self._sema = asyncio.Semaphore(container_start_limit)
         self._keepalive_interval_s = keepalive_interval_s

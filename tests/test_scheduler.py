@@ -173,3 +173,15 @@ async def test_gpu_any_normalization():
     s._normalize_gpu(req)
     assert "MI355X" in req.gpu_request and "any" not in req.gpu_request
     assert req.gpu_count == 1
+
+
+@pytest.mark.asyncio
+async def test_gpu_failover_chain():
+    s, wr, cr, _ = await make_sched()
+    s.cfg.scheduler.gpu_failover = {"MI355X": ["MI300X"]}
+    req = ContainerRequest(container_id="f1", cpu=1, memory_mb=1, gpu="MI355X", gpu_count=1)
+    s._normalize_gpu(req)
+    assert req.gpu_request == ["MI355X", "MI300X"]
+    # an MI300X worker is now eligible, ranked behind MI355X
+    mi300 = make_worker("b", gpus=8, gpu="MI300X")
+    assert s._filter(req, [mi300]) == [mi300]
