@@ -1,5 +1,5 @@
 #!/usr/bin/env python3
-"""Race screen + A/B for the experimental 8-phase GEMM (guide §5.4 two-lane
+"""Race screen + A/B for the phase-split GEMM variants (p8, p2) (guide §5.4 two-lane
 discipline: sync-structure edits get multi-run race screens at
 256/512/4096 plus within-probe A/B vs the unmodified kernel)."""
 
@@ -24,12 +24,12 @@ def main():
         for n in (256, 512, 1024, 4096):
             a = (torch.randn(n, n, device="cuda") * 0.5).to(torch.bfloat16)
             b = (torch.randn(n, n, device="cuda") * 0.5).to(torch.bfloat16)
-            out = ext.gemm_bf16_nt_p8(a, b, False)
             ref = (a.float() @ b.float().t())
-            e = relerr(out, ref)
-            if e > 0.02:
-                print(f"RACE/NUMERICS FAIL run={run} n={n} relerr={e}")
-                ok = False
+            for vname, vfn in (("p8", ext.gemm_bf16_nt_p8), ("p2", ext.gemm_bf16_nt_p2)):
+                e = relerr(vfn(a, b, False), ref)
+                if e > 0.02:
+                    print(f"RACE/NUMERICS FAIL {vname} run={run} n={n} relerr={e}")
+                    ok = False
     print("race screen:", "PASS" if ok else "FAIL")
     if not ok:
         return 1
@@ -39,8 +39,10 @@ def main():
         a = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
         b = (torch.randn(N, K, device="cuda") * 0.5).to(torch.bfloat16)
         for name, fn in [("base", lambda: ext.gemm_bf16_nt(a, b, False, 1)),
+                         ("p2  ", lambda: ext.gemm_bf16_nt_p2(a, b, False)),
                          ("p8  ", lambda: ext.gemm_bf16_nt_p8(a, b, False)),
                          ("base", lambda: ext.gemm_bf16_nt(a, b, False, 1)),
+                         ("p2  ", lambda: ext.gemm_bf16_nt_p2(a, b, False)),
                          ("p8  ", lambda: ext.gemm_bf16_nt_p8(a, b, False))]:
             for _ in range(3):
                 fn()

@@ -412,8 +412,13 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
       if (h < total_halves) stage_half(h);
       if (q == 3) {
         // next K-tile's 4 halves must land before its first phase reads:
-        // allow only the newest stage in flight (2 glds per wave)
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        // allow only the newest stage in flight (2 glds per wave). At the
+        // tail (kt == ntiles-2) the h==total stage above was guarded off, so
+        // the newest in-flight glds ARE the last tile's B1 — drain fully.
+        if (kt == ntiles - 2)
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
@@ -427,6 +432,157 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
           for (int j = 0; j < 2; j++)
             acc[mh * 4 + i][nh * 2 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a_frag[ks][i], b_frag[ks][j], acc[mh * 4 + i][nh * 2 + j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      int col = tile_n + wc * 64 + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 128 + i * 16 + c_row0 + r;
+        C[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 256x256 2-phase ("half-phase") kernel: the middle point between the base
+// 256^2 kernel (full-tile staging, 64 MFMAs between barriers) and the 8-phase
+// variant (16 MFMAs/phase, which measured slower — too many barriers). Each
+// K-tile runs as TWO phases, one per m-half: a phase reads the wave's A
+// half-rows (4 frags x 2 ks) plus full-width B (4 frags x 2 ks) and runs
+// 32 MFMAs. Staging is 2 half-tiles per phase in the order [A0, B0, B1, A1]
+// (B is read full-width in BOTH phases, so both B halves must land before
+// phase 0; A1 only before phase 1), ring lead 6 halves, counted waits
+// vmcnt(8) after phase 0 / vmcnt(6) after phase 1, exact vmcnt(0)/vmcnt(2)
+// tail fallbacks once the stage guard turns off.
+// ---------------------------------------------------------------------------
+
+template <bool XCD_SWIZZLE>
+__global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p2_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ Bt,
+    bf16_t* __restrict__ C, int M, int N, int K) {
+  __shared__ bf16_t lds[2 * 2 * BM2 * BK];  // 2 buf x (A|B)[256][64]
+
+  int nwg_n = N / BN2;
+  int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = (wg / nwg_n) * BM2;
+  int tile_n = (wg % nwg_n) * BN2;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 2, wc = wave & 3;  // wave tile: 128m x 64n
+
+  int ld_row = lane >> 3;
+  int ld_chunk = lane & 7;
+
+  // stage order o within a tile -> half kind: {A rows 0-127, B rows 0-127,
+  // B rows 128-255, A rows 128-255}. kind&1 selects the B image.
+  auto stage_ord = [&](int g) {
+    int kt = g >> 2;
+    int o = g & 3;
+    int kind = o ^ (o >> 1);  // order {0,1,2,3} -> kind {A0,B0,B1,A1}
+    int buf = kt & 1;
+    bf16_t* img = lds + buf * 2 * BM2 * BK + ((kind & 1) ? BM2 * BK : 0);
+    const bf16_t* src = (kind & 1) ? Bt + (int64_t)(tile_n)*K : A + (int64_t)(tile_m)*K;
+    int row0 = (kind >> 1) * 128;
+#pragma unroll
+    for (int call = 0; call < 2; call++) {  // 16 pieces over 8 waves
+      int p = wave + WAVES2 * call;
+      int row = row0 + p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(src + (int64_t)row * K + (int64_t)kt * BK + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(img + row * BK),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+  int ntiles = K / BK;
+  int total = ntiles * 4;
+  const int LEAD = 6;
+
+  // prologue: 6 halves staged; phase (0,0) needs orders 0..2 landed ->
+  // allow the newest 3 stages (6 glds) in flight
+  for (int g = 0; g < LEAD && g < total; g++) stage_ord(g);
+  if (total > LEAD)
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int fr_row = lane & 15;
+  int fr_q = lane >> 4;
+
+  for (int kt = 0; kt < ntiles; kt++) {
+    int buf = kt & 1;
+    bf16_t* Ab = lds + buf * 2 * BM2 * BK;
+    bf16_t* Bb = Ab + BM2 * BK;
+#pragma unroll
+    for (int mh = 0; mh < 2; mh++) {  // phase = m-half
+      // ---- phase reads (current tile, landed) --------------------------
+      bf16x8 a_frag[2][4], b_frag[2][4];
+#pragma unroll
+      for (int ks = 0; ks < 2; ks++) {
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          int row = wr * 128 + mh * 64 + i * 16 + fr_row;
+          int sc = (ks * 4 + fr_q) ^ (row & 7);
+          a_frag[ks][i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
+        }
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          int col = wc * 64 + j * 16 + fr_row;
+          int scb = (ks * 4 + fr_q) ^ (col & 7);
+          b_frag[ks][j] = *(const bf16x8*)(Bb + col * BK + scb * 8);
+        }
+      }
+      // ---- stage 2 future halves + counted landing wait ----------------
+      int g0 = (kt * 2 + mh) * 2 + LEAD;
+      if (g0 < total) stage_ord(g0);
+      if (g0 + 1 < total) stage_ord(g0 + 1);
+      if (mh == 0) {
+        // phase (kt,1) needs order kt*4+3 (A1) landed; steady state leaves
+        // 4 stages (8 glds) in flight
+        if (kt == ntiles - 1)
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      } else if (kt < ntiles - 1) {
+        // phase (kt+1,0) needs orders <= kt*4+6 landed; steady state leaves
+        // 3 stages (6 glds) in flight
+        if (kt == ntiles - 2)
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      // ---- 32 MFMA at raised priority ----------------------------------
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ks++)
+#pragma unroll
+        for (int i = 0; i < 4; i++)
+#pragma unroll
+          for (int j = 0; j < 4; j++)
+            acc[mh * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[ks][i], b_frag[ks][j], acc[mh * 4 + i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
       __builtin_amdgcn_s_barrier();
     }
@@ -819,6 +975,31 @@ torch::Tensor gemm_bf16_nt_p8(torch::Tensor A, torch::Tensor Bt,
   return C;
 }
 
+torch::Tensor gemm_bf16_nt_p2(torch::Tensor A, torch::Tensor Bt,
+                              bool xcd_swizzle) {
+  check_bf16_2d(A, "A");
+  check_bf16_2d(Bt, "Bt");
+  int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % BM2 == 0 && N % BN2 == 0 && K % BK == 0,
+              "p2 kernel needs M,N %% 256 == 0 and K %% 64 == 0");
+  auto C = torch::empty({M, N}, A.options());
+  dim3 grid((M / BM2) * (N / BN2));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (xcd_swizzle) {
+    hipLaunchKernelGGL((gemm_bf16_nt_256p2_kernel<true>), grid, dim3(THREADS2),
+                       0, stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                       (int)M, (int)N, (int)K);
+  } else {
+    hipLaunchKernelGGL((gemm_bf16_nt_256p2_kernel<false>), grid,
+                       dim3(THREADS2), 0, stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
+                       (int)M, (int)N, (int)K);
+  }
+  return C;
+}
+
 torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
                                 torch::Tensor Wu, bool xcd_swizzle) {
   check_bf16_2d(A, "A");
@@ -883,6 +1064,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("xcd_swizzle") = true, py::arg("splitk") = -1);
   m.def("gemm_gate_up_silu", &gemm_gate_up_silu, py::arg("A"), py::arg("Wg"),
         py::arg("Wu"), py::arg("xcd_swizzle") = true);
+  m.def("gemm_bf16_nt_p2", &gemm_bf16_nt_p2, py::arg("A"), py::arg("Bt"),
+        py::arg("xcd_swizzle") = false);
   m.def("gemm_bf16_nt_p8", &gemm_bf16_nt_p8, py::arg("A"), py::arg("Bt"),
         py::arg("xcd_swizzle") = false);
   m.def("silu_mul", &silu_mul);
