@@ -160,3 +160,9 @@ def metrics_series(name: str) -> str:
 
 def events_stream() -> str:
     return "events:stream"
+
+
+def workspace_usage(workspace_id: str) -> str:
+    """Accumulated usage counters hash: task_count, task_duration_ms,
+    cost_cents (reference: usage metrics repo counters)."""
+    return f"usage:{workspace_id}"

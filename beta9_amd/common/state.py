@@ -189,6 +189,11 @@ class StateStore:
         h[f] = int(h.get(f, 0)) + by
         return h[f]
 
+    async def hincrbyfloat(self, key: str, f: str, by: float) -> float:
+        h = self._hashes.setdefault(key, {})
+        h[f] = float(h.get(f, 0.0)) + by
+        return h[f]
+
     # -- sorted sets (scheduler backlog) -------------------------------------
     async def zadd(self, key: str, member: str, score: float) -> None:
         z = self._zsets.setdefault(key, _ZSet())

@@ -151,6 +151,7 @@ class Gateway:
         self.metrics = REGISTRY
         self.scheduler.events = self.events
         self.scheduler.metrics_registry = self.metrics
+        self.dispatcher.metrics = self.metrics
         await self.dispatcher.start()
         await self.scheduler.start()
         self._register_services()

@@ -151,6 +151,18 @@ def build_http_app(gateway) -> FastAPI:
             "registry": gateway.metrics.snapshot() if getattr(gateway, "metrics", None) else {},
         }
 
+    @app.get("/api/v1/usage")
+    async def workspace_usage(auth: dict = Depends(auth_ctx)):
+        from ..common import keys as _keys
+
+        u = await gateway.store.hgetall(_keys.workspace_usage(auth["workspace_id"]))
+        return {
+            "workspace_id": auth["workspace_id"],
+            "task_count": int(u.get("task_count", 0)),
+            "task_duration_ms": float(u.get("task_duration_ms", 0.0)),
+            "cost_cents": float(u.get("cost_cents", 0.0)),
+        }
+
     @app.get("/metrics")
     async def metrics_prometheus():
         if getattr(gateway, "metrics", None) is None:

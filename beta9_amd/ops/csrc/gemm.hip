@@ -459,11 +459,14 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
 // variant (16 MFMAs/phase, which measured slower — too many barriers). Each
 // K-tile runs as TWO phases, one per m-half: a phase reads the wave's A
 // half-rows (4 frags x 2 ks) plus full-width B (4 frags x 2 ks) and runs
-// 32 MFMAs. Staging is 2 half-tiles per phase in the order [A0, B0, B1, A1]
-// (B is read full-width in BOTH phases, so both B halves must land before
-// phase 0; A1 only before phase 1), ring lead 6 halves, counted waits
-// vmcnt(8) after phase 0 / vmcnt(6) after phase 1, exact vmcnt(0)/vmcnt(2)
-// tail fallbacks once the stage guard turns off.
+// 32 MFMAs. The A halves match the phase read pattern (phase mh reads
+// global rows mh*64..+63 and 128+mh*64..+63 across the two wave spans —
+// NOT rows 0-127/128-255). Staging is strict-previous-phase: phase (kt,0)
+// stages B0,B1,Ah1 of tile kt+1, phase (kt,1) stages Ah0 of tile kt+2, so
+// every LDS slot is overwritten a full barrier-delimited phase after its
+// last read (no same-phase read/overwrite windows). Counted waits:
+// vmcnt(8) after phase 0 / vmcnt(4) after phase 1, exact vmcnt(0)/vmcnt(2)
+// tail fallbacks once the stage guards turn off.
 // ---------------------------------------------------------------------------
 
 template <bool XCD_SWIZZLE>
@@ -491,20 +494,29 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p2_kernel(
   int ld_row = lane >> 3;
   int ld_chunk = lane & 7;
 
-  // stage order o within a tile -> half kind: {A rows 0-127, B rows 0-127,
-  // B rows 128-255, A rows 128-255}. kind&1 selects the B image.
+  // stage order o within a tile -> half kind {0:Ah0, 1:B0, 2:B1, 3:Ah1}.
+  // The A halves MATCH THE PHASE READ PATTERN (phase mh reads wave rows
+  // wr*128 + mh*64 + i*16, i.e. global rows mh*64..mh*64+63 and
+  // 128+mh*64..128+mh*64+63): Ah0 = rows 0-63 u 128-191, Ah1 = the rest.
+  // B halves are by image rows (any split works: B is read full-width in
+  // both phases, so both must land before phase 0).
   auto stage_ord = [&](int g) {
     int kt = g >> 2;
     int o = g & 3;
-    int kind = o ^ (o >> 1);  // order {0,1,2,3} -> kind {A0,B0,B1,A1}
     int buf = kt & 1;
-    bf16_t* img = lds + buf * 2 * BM2 * BK + ((kind & 1) ? BM2 * BK : 0);
-    const bf16_t* src = (kind & 1) ? Bt + (int64_t)(tile_n)*K : A + (int64_t)(tile_m)*K;
-    int row0 = (kind >> 1) * 128;
+    bool is_b = (o == 1) | (o == 2);
+    bf16_t* img = lds + buf * 2 * BM2 * BK + (is_b ? BM2 * BK : 0);
+    const bf16_t* src = is_b ? Bt + (int64_t)(tile_n)*K : A + (int64_t)(tile_m)*K;
 #pragma unroll
     for (int call = 0; call < 2; call++) {  // 16 pieces over 8 waves
       int p = wave + WAVES2 * call;
-      int row = row0 + p * 8 + ld_row;
+      int row;
+      if (is_b) {
+        row = (o - 1) * 128 + p * 8 + ld_row;
+      } else {
+        int mh = o >> 1;  // o==0 -> Ah0, o==3 -> Ah1
+        row = (p >> 3) * 128 + mh * 64 + (p & 7) * 8 + ld_row;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(src + (int64_t)row * K + (int64_t)kt * BK + ((ld_chunk ^ (row & 7)) * 8)),
           (__attribute__((address_space(3))) uint32_t*)(img + row * BK),
@@ -514,16 +526,20 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p2_kernel(
 
   f32x4 acc[8][4] = {};
   int ntiles = K / BK;
-  int total = ntiles * 4;
-  const int LEAD = 6;
 
-  // prologue: 6 halves staged; phase (0,0) needs orders 0..2 landed ->
-  // allow the newest 3 stages (6 glds) in flight
-  for (int g = 0; g < LEAD && g < total; g++) stage_ord(g);
-  if (total > LEAD)
-    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-  else
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  // Strict previous-phase staging: every LDS slot is overwritten at least
+  // one full barrier-delimited phase after its last read (no same-phase
+  // read/overwrite windows at all). Schedule: phase (kt,0) stages
+  // B0,B1,Ah1 of tile kt+1; phase (kt,1) stages Ah0 of tile kt+2.
+  // Prologue: tile 0 fully + Ah0 of tile 1; phase (0,0) needs Ah0,B0,B1
+  // of tile 0 -> allow the newest 2 stages (Ah1(0), Ah0(1)) in flight.
+  stage_ord(0); stage_ord(1); stage_ord(2); stage_ord(3);
+  if (ntiles > 1) {
+    stage_ord(4);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  }
   __builtin_amdgcn_s_barrier();
 
   int fr_row = lane & 15;
@@ -552,24 +568,27 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p2_kernel(
           b_frag[ks][j] = *(const bf16x8*)(Bb + col * BK + scb * 8);
         }
       }
-      // ---- stage 2 future halves + counted landing wait ----------------
-      int g0 = (kt * 2 + mh) * 2 + LEAD;
-      if (g0 < total) stage_ord(g0);
-      if (g0 + 1 < total) stage_ord(g0 + 1);
+      // ---- stage future halves + counted landing wait ------------------
       if (mh == 0) {
-        // phase (kt,1) needs order kt*4+3 (A1) landed; steady state leaves
-        // 4 stages (8 glds) in flight
+        if (kt + 1 < ntiles) {
+          stage_ord((kt + 1) * 4 + 1);  // B0(kt+1)
+          stage_ord((kt + 1) * 4 + 2);  // B1(kt+1)
+          stage_ord((kt + 1) * 4 + 3);  // Ah1(kt+1)
+        }
+        // phase (kt,1) needs Ah1(kt) landed (staged at (kt-1,0)); steady
+        // state leaves Ah0(kt+1) + the 3 stages above = 8 glds in flight
         if (kt == ntiles - 1)
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         else
           asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       } else if (kt < ntiles - 1) {
-        // phase (kt+1,0) needs orders <= kt*4+6 landed; steady state leaves
-        // 3 stages (6 glds) in flight
+        if (kt + 2 < ntiles) stage_ord((kt + 2) * 4);  // Ah0(kt+2)
+        // phase (kt+1,0) needs Ah0,B0,B1(kt+1) landed; steady state leaves
+        // Ah1(kt+1) + Ah0(kt+2) = 4 glds in flight
         if (kt == ntiles - 2)
           asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
         else
-          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();

@@ -228,6 +228,27 @@ class Volume:
 
 
 @dataclass
+class PricingPolicy:
+    """Per-task billing on public stubs (reference: sdk type.py:437)."""
+
+    def __init__(self, max_in_flight: int = 10, cost_model: str = "task",
+                 cost_per_task: float = 0.0, cost_per_task_duration_ms: float = 0.0):
+        self.max_in_flight = max_in_flight
+        self.cost_model = cost_model
+        self.cost_per_task = cost_per_task
+        self.cost_per_task_duration_ms = cost_per_task_duration_ms
+
+    def to_config(self):
+        from .types import PricingPolicy as _PP
+
+        return _PP(
+            max_in_flight=self.max_in_flight,
+            cost_model=self.cost_model,
+            cost_per_task=self.cost_per_task,
+            cost_per_task_duration_ms=self.cost_per_task_duration_ms,
+        )
+
+
 class QueueDepthAutoscaler:
     max_containers: int = 1
     tasks_per_container: int = 1
@@ -322,6 +343,7 @@ class RunnerAbstraction:
         client: Optional[Client] = None,
         inputs=None,  # Schema
         outputs=None,  # Schema
+        pricing: Optional["PricingPolicy"] = None,
     ):
         self.func = func
         self.cpu = cpu
@@ -338,6 +360,7 @@ class RunnerAbstraction:
         self.on_start = on_start
         self.volumes = volumes or []
         self.secrets = secrets or []
+        self.pricing = pricing
         self.env = env or {}
         self.autoscaler = autoscaler
         self.checkpoint_enabled = checkpoint_enabled
@@ -394,6 +417,7 @@ class RunnerAbstraction:
             timeout=self.timeout,
             inputs=self.inputs.to_spec() if self.inputs is not None else None,
             outputs=self.outputs.to_spec() if self.outputs is not None else None,
+            pricing=self.pricing.to_config() if self.pricing is not None else None,
         )
         return cfg
 

@@ -36,6 +36,7 @@ class Dispatcher:
         # event-driven completion (hot path: no polling between a runner's
         # EndTask and the caller's wait())
         self._done_events: Dict[str, asyncio.Event] = {}
+        self.metrics = None  # MetricsRegistry, wired by the gateway
 
     def register(self, executor_name: str, fn: Executor) -> None:
         self._executors[executor_name] = fn
@@ -131,7 +132,34 @@ class Dispatcher:
         task = await self.get_task(task_id)
         await self.update_status(task_id, status)
         if task is not None:
+            await self._track_usage(task, status)
             await self._fire_callback(task, status, result)
+
+    async def _track_usage(self, task: Task, status: TaskStatus) -> None:
+        """Meter the finished task into workspace usage counters; cost is
+        derived from the stub's PricingPolicy (reference:
+        pkg/abstractions/common/usage.go TrackTaskCost/TrackTaskCount)."""
+        from ..common import keys as _keys
+
+        ukey = _keys.workspace_usage(task.workspace_id or "default")
+        await self.store.hincr(ukey, "task_count", 1)
+        dur_ms = max(0.0, (task.ended_at or time.time()) - (task.started_at or task.created_at)) * 1000.0
+        await self.store.hincrbyfloat(ukey, "task_duration_ms", dur_ms)
+        cfg = self._stub_config_cached(task.stub_id)
+        pricing = getattr(cfg, "pricing", None) if cfg else None
+        if pricing is None or status != TaskStatus.complete:
+            return
+        if pricing.cost_model == "duration":
+            cost_cents = pricing.cost_per_task_duration_ms * dur_ms * 100.0
+        else:
+            cost_cents = pricing.cost_per_task * 100.0
+        if cost_cents:
+            await self.store.hincrbyfloat(ukey, "cost_cents", cost_cents)
+            if self.metrics is not None:
+                self.metrics.inc(
+                    "task_cost_cents", cost_cents,
+                    workspace_id=task.workspace_id, stub_id=task.stub_id,
+                )
 
     async def _fire_callback(self, task: Task, status: TaskStatus, result: Any) -> None:
         """POST the terminal result to the stub's callback_url (reference:

@@ -181,6 +181,18 @@ class CheckpointState(BaseModel):
     created_at: float = 0.0
 
 
+class PricingPolicy(BaseModel):
+    """Per-task billing policy on public stubs (reference:
+    sdk type.py:431-441 PricingPolicy, pkg/abstractions/common/usage.go
+    TrackTaskCost): cost per task or per task-duration-ms, metered into
+    workspace usage counters at task end."""
+
+    max_in_flight: int = 10
+    cost_model: str = "task"  # task | duration
+    cost_per_task: float = 0.0
+    cost_per_task_duration_ms: float = 0.0
+
+
 class StubConfig(BaseModel):
     """SDK <-> gateway stub contract, stored as JSON on the stub row
     (reference: pkg/types/backend.go:526-567 StubConfigV1)."""
@@ -211,6 +223,7 @@ class StubConfig(BaseModel):
     block_network: bool = False
     pool: str = ""
     timeout: float = 3600.0
+    pricing: Optional[PricingPolicy] = None
 
 
 class ContainerRequest(BaseModel):

@@ -115,3 +115,24 @@ async def test_drain_worker():
         assert r["drained"] is True
         w = await c.gateway.worker_repo.get(wid)
         assert w.cordoned is True
+
+
+@pytest.mark.asyncio
+async def test_pricing_policy_meters_usage():
+    """PricingPolicy on a stub meters task count + cost into workspace usage
+    (reference: pkg/abstractions/common/usage.go TrackTaskCost)."""
+    async with cluster(n_workers=1) as c:
+        import handlers
+
+        fn = beta9_amd.Function(
+            handlers.hello, cpu=100, memory=128,
+            pricing=beta9_amd.PricingPolicy(cost_model="task", cost_per_task=0.02),
+        )
+        for _ in range(3):
+            assert await asyncio.to_thread(fn.remote, 2) == {"hello": 4}
+        usage_hashes = {k: v for k, v in c.gateway.store._hashes.items() if k.startswith("usage:")}
+        assert usage_hashes, "no usage counters recorded"
+        ws = next(iter(usage_hashes.values()))
+        assert int(ws.get("task_count", 0)) >= 3
+        assert abs(float(ws.get("cost_cents", 0)) - 6.0) < 1e-6  # 3 * $0.02 = 6 cents
+        assert float(ws.get("task_duration_ms", 0)) > 0
