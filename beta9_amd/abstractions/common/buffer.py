@@ -117,10 +117,31 @@ class RequestBuffer:
         except Exception:
             return False
 
-    async def _acquire_slot(self, deadline: float):
-        """Pick a healthy container with a free token; wait for one."""
+    async def _acquire_slot(self, deadline: float, affinity_key: Optional[str] = None):
+        """Pick a healthy container with a free token; wait for one.
+
+        affinity_key: consistent-hash routing so repeat requests with the
+        same key land on the same container while it has free tokens —
+        KV-cache/prefix reuse for LLM serving (reference:
+        pkg/abstractions/pod/llm.go prefix-affinity hashing). Falls back to
+        most-free when the preferred container is saturated.
+        """
         while time.monotonic() < deadline:
             slots = await self._discover()
+            if affinity_key is not None:
+                healthy = sorted(
+                    cid for cid, s in slots.items() if s.healthy and not s.sema.locked()
+                )
+                if healthy:
+                    import zlib
+
+                    cid = healthy[zlib.crc32(affinity_key.encode()) % len(healthy)]
+                    slot = slots[cid]
+                    try:
+                        await asyncio.wait_for(slot.sema.acquire(), timeout=0.05)
+                        return cid, slot
+                    except asyncio.TimeoutError:
+                        pass  # saturated: fall through to most-free
             best: Optional[tuple] = None
             for cid, slot in slots.items():
                 if not slot.healthy:
@@ -146,13 +167,14 @@ class RequestBuffer:
         content: bytes = b"",
         json_body: Any = None,
         wait_timeout: float = 120.0,
+        affinity_key: Optional[str] = None,
     ) -> Dict[str, Any]:
         """Dispatch one request; returns {"status_code", "body"}."""
         self.inflight += 1
         if self.on_inflight_change:
             self.on_inflight_change(self.inflight)
         try:
-            cid, slot = await self._acquire_slot(time.monotonic() + wait_timeout)
+            cid, slot = await self._acquire_slot(time.monotonic() + wait_timeout, affinity_key)
             try:
                 if self.protocol == "grpc":
                     args = json_body

@@ -110,11 +110,11 @@ class EndpointInstance:
             timestamp=time.time(),
         )
 
-    async def forward(self, method: str, path: str, headers=None, content: bytes = b"", json_body=None):
+    async def forward(self, method: str, path: str, headers=None, content: bytes = b"", json_body=None, affinity_key=None):
         self.instance.autoscaler.trigger()
         return await self.buffer.dispatch(
             method, path, headers=headers, content=content, json_body=json_body,
-            wait_timeout=min(self.cfg.timeout, 120.0),
+            wait_timeout=min(self.cfg.timeout, 120.0), affinity_key=affinity_key,
         )
 
     async def shutdown(self) -> None:
@@ -133,12 +133,16 @@ class EndpointExecutor:
         instance = await get_instance(self.gw, task.stub_id)
         payload = payload or {}
         try:
+            # affinity: explicit payload key or X-B9-Affinity header ->
+            # consistent-hash container routing (KV-cache/prefix reuse)
+            hdrs = payload.get("headers") or {}
             resp = await instance.forward(
                 payload.get("method", "POST"),
                 payload.get("path", "/"),
                 headers=payload.get("headers"),
                 json_body=payload.get("json"),
                 content=payload.get("content", b""),
+                affinity_key=payload.get("affinity") or hdrs.get("x-b9-affinity"),
             )
             await self.gw.dispatcher.complete(
                 task.task_id,

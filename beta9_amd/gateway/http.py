@@ -53,7 +53,14 @@ def build_http_app(gateway) -> FastAPI:
             {
                 "method": request.method,
                 "path": path,
-                "headers": {"content-type": request.headers.get("content-type", "application/json")},
+                "headers": {
+                    "content-type": request.headers.get("content-type", "application/json"),
+                    **(
+                        {"x-b9-affinity": request.headers["x-b9-affinity"]}
+                        if "x-b9-affinity" in request.headers
+                        else {}
+                    ),
+                },
                 "content": body,
             },
         )

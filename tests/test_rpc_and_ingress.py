@@ -172,3 +172,46 @@ async def test_function_invoke_streams_logs():
         assert "result" in kinds
         result = [m for m in msgs if m["type"] == "result"][0]
         assert result["result"] == {"hello": 4}
+
+
+@pytest.mark.asyncio
+async def test_affinity_routing_consistent_hash():
+    """Same affinity key -> same healthy container; saturation falls back
+    (reference analog: pod/llm.go prefix-affinity hashing)."""
+    from beta9_amd.abstractions.common.buffer import RequestBuffer, _ContainerSlot
+
+    buf = RequestBuffer.__new__(RequestBuffer)
+    buf.tokens_per_container = 2
+    slots = {f"c{i}": _ContainerSlot(f"127.0.0.1:{9000+i}", 2) for i in range(4)}
+    for s in slots.values():
+        s.healthy = True
+
+    async def fake_discover():
+        return slots
+
+    buf._discover = fake_discover
+    import time as _t
+
+    picks = set()
+    for _ in range(5):
+        cid, slot = await buf._acquire_slot(_t.monotonic() + 1.0, affinity_key="session-42")
+        picks.add(cid)
+        slot.sema.release()
+    assert len(picks) == 1  # deterministic routing
+
+    # saturate the preferred container -> falls back to another
+    pref = picks.pop()
+    await slots[pref].sema.acquire()
+    await slots[pref].sema.acquire()
+    cid2, slot2 = await buf._acquire_slot(_t.monotonic() + 1.0, affinity_key="session-42")
+    assert cid2 != pref
+    slot2.sema.release()
+    slots[pref].sema.release(); slots[pref].sema.release()
+
+    # different keys spread across containers
+    spread = set()
+    for k in range(20):
+        cid, slot = await buf._acquire_slot(_t.monotonic() + 1.0, affinity_key=f"k{k}")
+        spread.add(cid)
+        slot.sema.release()
+    assert len(spread) >= 2
