@@ -110,6 +110,14 @@ class RequestBuffer:
                 r = await slot.rpc.call("Runner", "Health", {}, timeout=2.0)
                 return bool(r.get("ok"))
             except Exception:
+                # a probe that raced the runner's bind leaves the channel in
+                # reconnect backoff; every retry through it then fails until
+                # the backoff expires (~100 ms — measured, benchmarks/
+                # cold_trace.py). A fresh channel per failed probe makes the
+                # next 5 ms probe do an immediate TCP connect instead.
+                if slot.rpc is not None:
+                    rpc, slot.rpc = slot.rpc, None
+                    asyncio.ensure_future(rpc.close())
                 return False
         try:
             r = await self._client.get(f"http://{slot.addr}/health", timeout=2.0)

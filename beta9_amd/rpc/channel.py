@@ -224,6 +224,13 @@ class RpcClient:
                     ("grpc.initial_reconnect_backoff_ms", 20),
                     ("grpc.min_reconnect_backoff_ms", 20),
                     ("grpc.max_reconnect_backoff_ms", 500),
+                    # ...and backoff state lives on the SUBCHANNEL, which the
+                    # GLOBAL pool shares across channels to the same target:
+                    # recreating the channel would still reuse the backed-off
+                    # subchannel (measured: ~90 ms of refused probes after the
+                    # runner was live, benchmarks/cold_trace.py). A private
+                    # pool makes every fresh channel connect immediately.
+                    ("grpc.use_local_subchannel_pool", 1),
                 ],
             )
 

@@ -203,13 +203,19 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
                 return {"status_code": 500, "body": {"error": str(e)}}
 
     boot = {}
+    if os.environ.get("B9_BOOT_TRACE"):
+        print(f"[boottrace] run_rpc_server enter {time.time():.6f}", flush=True)
     t0 = time.perf_counter()
     server = RpcServer(host="0.0.0.0", port=env.bind_port)
     server.register_service(RunnerSvc())
     await server.start()
     boot["rpc_bind_ms"] = round((time.perf_counter() - t0) * 1000, 2)
 
+    if os.environ.get("B9_BOOT_TRACE"):
+        print(f"[boottrace] pre RunnerClient {time.time():.6f}", flush=True)
     client = RunnerClient(env)
+    if os.environ.get("B9_BOOT_TRACE"):
+        print(f"[boottrace] post RunnerClient {time.time():.6f}", flush=True)
     t1 = time.perf_counter()
     await run_on_start(env)
     boot["on_start_ms"] = round((time.perf_counter() - t1) * 1000, 2)
@@ -217,6 +223,8 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
     boot["total_ms"] = round((time.perf_counter() - t0) * 1000, 2)
     gpu.phases.update(boot)
     state["ready"] = True
+    if os.environ.get("B9_BOOT_TRACE"):
+        print(f"[boottrace] runner ready {time.time():.6f}", flush=True)
     if env.checkpoint_enabled and not gpu.restored:
         try:
             await client.take_checkpoint_if_granted(gpu)
@@ -283,6 +291,9 @@ def _with_health(user_app):
 
 
 def main() -> int:
+    if os.environ.get("B9_BOOT_TRACE"):
+        import time as _t
+        print(f"[boottrace] endpoint main start {_t.time():.6f}", flush=True)
     logging.basicConfig(level=logging.INFO)
     world = int(os.environ.get("WORLD_SIZE", "1") or 1)
     if world > 1:

@@ -91,6 +91,9 @@ def main() -> int:
             return 0
         data += chunk
     spec = json.loads(data.decode())
+    if os.environ.get("B9_BOOT_TRACE"):
+        import time as _t
+        print(f"[boottrace] shell spec received {_t.time():.6f}", flush=True)
     os.environ.update({str(k): str(v) for k, v in spec.get("env", {}).items()})
     conn.sendall(b"ok\n")
     conn.close()
