@@ -247,6 +247,40 @@ class ProcessManager:
         return {"files_changed": changed}
 
     @rpc()
+    async def SnapshotFilesystem(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Archive a subtree of the sandbox filesystem (tar.gz bytes) so the
+        gateway can register it as a new image (reference:
+        sandbox.py:527-549 create_image_from_filesystem -> worker archives
+        the container fs into a CLIP image)."""
+        import io
+        import tarfile
+
+        # default: the sandbox workspace (cwd). "/" only makes sense under
+        # the native runtime where it is the container overlay root.
+        sub = self._path(req.get("path") or ".")
+        max_bytes = int(req.get("max_bytes", 512 * 1024 * 1024))
+        buf = io.BytesIO()
+        total = 0
+        with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+            for dirpath, dirnames, filenames in os.walk(sub):
+                # never descend into kernel/virtual trees
+                dirnames[:] = [d for d in dirnames if os.path.join(dirpath, d) not in ("/proc", "/sys", "/dev")]
+                for fn in filenames:
+                    full = os.path.join(dirpath, fn)
+                    try:
+                        st = os.lstat(full)
+                    except OSError:
+                        continue
+                    if not (os.path.isfile(full) and not os.path.islink(full)):
+                        continue
+                    total += st.st_size
+                    if total > max_bytes:
+                        return {"error": f"subtree exceeds max_bytes={max_bytes}", "size": total}
+                    tf.add(full, arcname=os.path.relpath(full, sub))
+        data = buf.getvalue()
+        return {"archive": data, "size": len(data), "files_root": sub}
+
+    @rpc()
     async def FindInFiles(self, req: Dict[str, Any], context, auth) -> List[Dict]:
         root = self._path(req["path"])
         regex = re.compile(req["pattern"])

@@ -184,6 +184,8 @@ class Image:
 
     @property
     def is_empty(self) -> bool:
+        if getattr(self, "_snapshot_id", ""):
+            return False
         return not (self.python_packages or self.commands or self.base_image or self.env_vars)
 
     def build_payload(self) -> Dict[str, Any]:
@@ -195,7 +197,17 @@ class Image:
             "env_vars": self.env_vars,
         }
 
+    @classmethod
+    def from_snapshot(cls, image_id: str) -> "Image":
+        """An Image pinned to an existing id (e.g. from
+        Sandbox.create_image_from_filesystem())."""
+        img = cls()
+        img._snapshot_id = image_id
+        return img
+
     def build(self, client: Optional[Client] = None, verbose: bool = False) -> str:
+        if getattr(self, "_snapshot_id", ""):
+            return self._snapshot_id
         client = client or Client()
         v = client.call("ImageService", "VerifyImageBuild", self.build_payload())
         if v["exists"]:
@@ -701,6 +713,18 @@ class SandboxInstance:
 
     def expose_port(self, port: int) -> str:
         return self._call("SandboxExposePort", {"port": port})["url"]
+
+    def create_image_from_filesystem(self, path: str = ".", timeout: float = 300.0) -> str:
+        """Snapshot the sandbox filesystem into a new image id usable by
+        future stubs via Image.from_snapshot() (reference:
+        sandbox.py:527-549)."""
+        r = self._call("SandboxCreateImageFromFilesystem", {"path": path, "timeout": timeout})
+        if r.get("error"):
+            raise RuntimeError(r["error"])
+        return r["image_id"]
+
+    def snapshot_memory(self) -> Dict:
+        return self._call("SandboxSnapshotMemory", {})
 
     def terminate(self) -> None:
         self.client.call("PodService", "StopPod", {"container_id": self.container_id})

@@ -44,3 +44,9 @@ def on_start():
 
 def check_started(**kwargs):
     return _started["on_start_ran"]
+
+
+def image_files_env():
+    import os
+
+    return os.environ.get("B9_IMAGE_FILES", "")

@@ -137,3 +137,45 @@ async def test_log_streaming():
         await asyncio.to_thread(consume)
         assert "line-A" in lines and "line-B" in lines
         assert exit_code == 0
+
+
+@pytest.mark.asyncio
+async def test_sandbox_create_image_from_filesystem():
+    """Live sandbox fs -> new image -> new container sees the files tree
+    (reference: sandbox.py:527-549 create_image_from_filesystem)."""
+    async with cluster(n_workers=1) as c:
+        sb = await asyncio.to_thread(beta9_amd.Sandbox(cpu=100, memory=128).create)
+        try:
+            await asyncio.to_thread(sb.fs.upload, "data/hello.txt", b"snapshot-payload")
+            image_id = await asyncio.to_thread(sb.create_image_from_filesystem)
+            assert image_id.startswith("snap-")
+            # image registered with a files tree containing the sandbox file
+            import os
+
+            files = os.path.join(c.gateway.cfg.image_service.store_path, image_id, "files")
+            found = []
+            for dirpath, _, fns in os.walk(files):
+                found += [os.path.join(dirpath, f) for f in fns]
+            assert any(f.endswith("data/hello.txt") for f in found), found
+            # a function using the snapshot image gets B9_IMAGE_FILES
+            import handlers
+
+            fn = beta9_amd.Function(
+                handlers.image_files_env, cpu=100, memory=128,
+                image=beta9_amd.Image.from_snapshot(image_id),
+            )
+            out = await asyncio.to_thread(fn.remote)
+            assert out and image_id in out
+        finally:
+            await asyncio.to_thread(sb.terminate)
+
+
+@pytest.mark.asyncio
+async def test_sandbox_snapshot_memory_reports_unsupported():
+    async with cluster(n_workers=1) as c:
+        sb = await asyncio.to_thread(beta9_amd.Sandbox(cpu=100, memory=128).create)
+        try:
+            r = await asyncio.to_thread(sb.snapshot_memory)
+            assert r["supported"] is False and "hipsnap" in r["reason"]
+        finally:
+            await asyncio.to_thread(sb.terminate)
