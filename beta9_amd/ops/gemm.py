@@ -87,3 +87,31 @@ def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-6) -> torch.Tensor
     xf = x.float()
     out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
     return (out * w.float()).to(x.dtype)
+
+
+def warm_all(device: str = "cuda:0") -> bool:
+    """Launch every hot kernel once on tiny aligned shapes: HIP loads each
+    kernel's code object lazily at FIRST launch (~10-30 ms per kernel on a
+    cold page cache), which otherwise lands in the first serving request
+    (measured via benchmarks/cold_trace.py on a fresh box: probe healthy at
+    ~10 ms but first request ~150 ms). Called from the prewarm shell."""
+    ext = _load()
+    if ext is None or not torch.cuda.is_available():
+        return False
+    a = torch.randn(256, 64, dtype=torch.bfloat16, device=device)
+    w = torch.randn(256, 64, dtype=torch.bfloat16, device=device)
+    big = torch.randn(256, 256, dtype=torch.bfloat16, device=device)
+    bigw = torch.randn(256, 256, dtype=torch.bfloat16, device=device)
+    ext.gemm_bf16_nt(a, w, False)            # 128-tile (+ split-K reduce path)
+    ext.gemm_bf16_nt(a, w, True)             # XCD-swizzle template
+    ext.gemm_bf16_nt(big, bigw, False)       # 256-tile variant
+    ext.gemm_bf16_nt_p2(a, w, False)
+    ext.gemm_bf16_nt_p8(a, w, False)
+    ext.gemm_gate_up_silu(a, w, w, False)    # fused MLP gate (both BNJ tiles)
+    ext.gemm_gate_up_silu(torch.randn(1024, 64, dtype=torch.bfloat16, device=device),
+                          torch.randn(1024, 64, dtype=torch.bfloat16, device=device),
+                          torch.randn(1024, 64, dtype=torch.bfloat16, device=device), False)
+    ext.silu_mul(a, a)
+    ext.rmsnorm(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6)
+    torch.cuda.synchronize()
+    return True

@@ -43,7 +43,7 @@ def main() -> int:
                 from beta9_amd.ops import gemm as _g
 
                 if _g.has_native():
-                    _g.gemm_nt(x.repeat(1, 1), x)
+                    _g.warm_all()  # every kernel's first-launch cost
             except Exception:
                 pass
             # pre-create the HBM arena + pinned staging (hipMalloc slab and
