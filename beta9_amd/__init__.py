@@ -42,7 +42,7 @@ else:
         task_queue,
     )
     from .schema import Schema  # noqa: F401
-    from .sdk import LLMTokenPressureAutoscaler, PricingPolicy  # noqa: F401
+    from .sdk import Bot, LLMTokenPressureAutoscaler, PricingPolicy  # noqa: F401
     from .types import TaskPolicy  # noqa: F401
 
 __version__ = "0.1.0"

@@ -184,6 +184,7 @@ class Gateway:
         from ..abstractions.taskqueue import TaskQueueExecutor, TaskQueueService
         from ..abstractions.volume import VolumeService
         from ..abstractions.output import OutputService
+        from ..abstractions.bot import BotService
         from ..abstractions.pod import PodService
         from ..abstractions.image import ImageService
 
@@ -273,6 +274,7 @@ class Gateway:
             OutputService(self.db, self.cfg.storage.root),
             PodService(self),
             ImageService(self),
+            BotService(self),
         ]
         for svc in self._services:
             self.rpc_server.register_service(svc)

@@ -775,6 +775,114 @@ class SandboxProcessAPI:
 # ---------------------------------------------------------------------------
 
 
+class Bot:
+    """Petri-net agent orchestration (reference: experimental/bot — SDK Bot
+    with locations + transitions; each transition firing runs as a real
+    function container).
+
+    Usage::
+
+        bot = Bot(name="research")
+        bot.location("queries", marker={"q": "str"})
+        bot.location("answers", marker={"text": "str"})
+
+        @bot.transition(inputs={"queries": 1}, outputs=["answers"])
+        def answer(markers):
+            q = markers["queries"][0]["q"]
+            return {"markers": {"answers": [{"text": f"answer to {q}"}]}}
+
+        session = bot.create_session()
+        session.push("queries", {"q": "why is the sky blue"})
+        for ev in session.events(timeout=30):
+            ...
+    """
+
+    def __init__(self, name: str = "bot", model: str = "", welcome_message: str = "",
+                 client: Optional[Client] = None):
+        self.name = name
+        self.model = model
+        self.welcome_message = welcome_message
+        self.client = client
+        self._locations: Dict[str, Dict[str, Any]] = {}
+        self._transitions: Dict[str, Dict[str, Any]] = {}
+
+    def location(self, name: str, marker: Optional[Dict[str, str]] = None, expose: bool = True):
+        self._locations[name] = {"name": name, "marker": marker or {}, "expose": expose}
+        return self
+
+    def transition(self, inputs: Dict[str, int], outputs: Optional[List[str]] = None,
+                   cpu: int = 1000, memory: int = 1024, gpu: str = "",
+                   timeout: float = 600.0, description: str = "", confirm: bool = False):
+        def deco(fn):
+            self._transitions[fn.__name__] = {
+                "name": fn.__name__,
+                "handler": _handler_spec(fn),
+                "inputs": inputs,
+                "outputs": outputs or [],
+                "cpu": cpu, "memory": memory, "gpu": gpu,
+                "timeout": timeout, "description": description, "confirm": confirm,
+            }
+            return fn
+        return deco
+
+    def create_session(self, keep_warm_seconds: float = 10.0) -> "BotSessionClient":
+        client = self.client or Client()
+        r = client.call("BotService", "CreateBotSession", {
+            "name": self.name,
+            "keep_warm_seconds": keep_warm_seconds,
+            "config": {
+                "model": self.model,
+                "welcome_message": self.welcome_message,
+                "locations": self._locations,
+                "transitions": self._transitions,
+            },
+        })
+        return BotSessionClient(client, r["session_id"])
+
+
+class BotSessionClient:
+    def __init__(self, client: Client, session_id: str):
+        self.client = client
+        self.session_id = session_id
+        self._offset = 0
+
+    def push(self, location: str, *markers: Dict[str, Any]) -> None:
+        r = self.client.call("BotService", "PushBotMarkers",
+                             {"session_id": self.session_id, "markers": {location: list(markers)}})
+        if not r.get("ok"):
+            raise RuntimeError(r.get("error", "push failed"))
+
+    def send_event(self, event: Dict[str, Any]) -> None:
+        self.client.call("BotService", "PushBotEvent",
+                         {"session_id": self.session_id, "event": event})
+
+    def confirm(self, transition: str) -> None:
+        self.send_event({"type": "confirm_transition", "transition": transition})
+
+    def state(self) -> Dict[str, Any]:
+        return self.client.call("BotService", "GetBotState", {"session_id": self.session_id})
+
+    def events(self, timeout: float = 10.0):
+        """Yield events from the current offset; returns when a poll window
+        passes with no new events."""
+        import time as _time
+
+        deadline = _time.monotonic() + timeout
+        while _time.monotonic() < deadline:
+            r = self.client.call(
+                "BotService", "PopBotEvents",
+                {"session_id": self.session_id, "offset": self._offset,
+                 "timeout": min(5.0, max(0.1, deadline - _time.monotonic()))},
+                timeout=40.0,
+            )
+            evs = r["events"]
+            self._offset = r["next_offset"]
+            for ev in evs:
+                yield ev
+            if evs:
+                deadline = _time.monotonic() + timeout
+
+
 class MapClient:
     def __init__(self, name: str, client: Optional[Client] = None):
         self.name = name

@@ -50,3 +50,16 @@ def image_files_env():
     import os
 
     return os.environ.get("B9_IMAGE_FILES", "")
+
+
+def bot_answer(markers=None, **kw):
+    loc = next(iter(markers))
+    q = markers[loc][0].get("q", "?")
+    # answers go to whatever single output the net wires up
+    return {"markers": {"answers": [{"text": f"answer:{q}"}],
+                        "outbox": [{"text": f"answer:{q}"}]}}
+
+
+def bot_summarize(markers=None, **kw):
+    texts = [m["text"] for m in markers.get("answers", [])]
+    return {"markers": {"summaries": [{"text": " | ".join(sorted(texts))}]}}
