@@ -1,7 +1,11 @@
 #!/usr/bin/env python3
 """Fresh-box decomposition of the first serving request: what pays the
 ~145 ms after the runner is healthy (benchmarks/cold_trace.py)."""
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 def t(name, fn):
@@ -33,4 +37,5 @@ t("hipBLASLt second same shape", lambda: xa @ wd.t())
 h = torch.randn(64, 4096, dtype=torch.bfloat16, device="cuda:0")
 t("our mlp_gate_up FIRST 64->pad128", lambda: g.mlp_gate_up(h, model.blocks[0].w_gate, model.blocks[0].w_up))
 t("first full infer", lambda: model.infer(64))
+t("first infer 2048 tokens", lambda: model.infer(2048))
 t("second infer", lambda: model.infer(64))

@@ -113,5 +113,23 @@ def warm_all(device: str = "cuda:0") -> bool:
                           torch.randn(1024, 64, dtype=torch.bfloat16, device=device), False)
     ext.silu_mul(a, a)
     ext.rmsnorm(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6)
+    # torch EAGER kernels the serving path touches (residual add, checksum
+    # cast/abs/reduce, softmax, cat): each costs 10-40 ms at first launch on
+    # a cold box (measured 122 ms total, benchmarks/first_req_micro.py)
+    t = a + a
+    t = t * a
+    t = t / 2
+    f = t.float()
+    f.abs().mean().item()
+    f.sum().item()
+    torch.softmax(f, dim=-1)
+    torch.exp(f)
+    torch.rsqrt(f.pow(2).mean(-1, keepdim=True) + 1e-6)
+    torch.cat([a, a], dim=0)
+    a.t().contiguous()
+    torch.nn.functional.silu(f)
+    f.to(torch.bfloat16)
+    torch.randn(64, 64, device=device, dtype=torch.bfloat16)
+    torch.zeros(64, device=device)
     torch.cuda.synchronize()
     return True
