@@ -29,6 +29,9 @@ class GatewayConfig(BaseModel):
     http_port: int = 1994
     shutdown_timeout_s: float = 10.0
     invoke_url_type: str = "path"  # path | host
+    # host-style invoke URLs resolve {name}[-v{N}|-latest].app.{base_domain}
+    # (reference: pkg/gateway/middleware/subdomain.go, config invokeURLType)
+    base_domain: str = "b9.local"
     stub_limits_cpu: int = 128000
     stub_limits_memory_mb: int = 32768
     stub_limits_max_replicas: int = 64

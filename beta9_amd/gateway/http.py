@@ -35,6 +35,29 @@ def build_http_app(gateway) -> FastAPI:
     async def health():
         return {"status": "ok"}
 
+    # -- subdomain (host-style) routing -------------------------------------
+    # {name}.app.{base}, {name}-latest.app.{base}, {name}-v{N}.app.{base} or
+    # {stub_id}.app.{base} rewrite onto the matching invoke route before the
+    # router runs (reference: pkg/gateway/middleware/subdomain.go:19-36)
+    @app.middleware("http")
+    async def subdomain_routing(request, call_next):
+        host = request.headers.get("host", "").split(":")[0]
+        suffix = ".app." + gateway.cfg.gateway.base_domain
+        if host.endswith(suffix):
+            sub = host[: -len(suffix)]
+            path = request.scope.get("path", "/")
+            if gateway.db.get_stub(sub) is not None:
+                request.scope["path"] = f"/endpoint/id/{sub}" + (path if path != "/" else "/")
+            else:
+                name, version = sub, ""
+                if "-v" in sub and sub.rsplit("-v", 1)[1].isdigit():
+                    name, version = sub.rsplit("-v", 1)
+                elif sub.endswith("-latest"):
+                    name = sub[: -len("-latest")]
+                target = f"/deployment/{name}" + (f"/v{version}" if version else "")
+                request.scope["path"] = target + (path if path != "/" else "/")
+        return await call_next(request)
+
     # -- invoke routes ------------------------------------------------------
     async def _invoke(stub_id: str, request: Request, auth: dict, path: str = "/"):
         if gateway.draining:

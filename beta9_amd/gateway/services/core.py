@@ -172,7 +172,13 @@ class GatewayService:
                 inst.instance.autoscaler.trigger()
         except Exception:
             pass
-        return {"deployment_id": dep["id"], "version": dep["version"], "invoke_url": f"/{stub['type'].split('/')[0]}/id/{stub_id}"}
+        base_kind = stub["type"].split("/")[0]
+        if self.scheduler.cfg.gateway.invoke_url_type == "host":
+            base = self.scheduler.cfg.gateway.base_domain
+            invoke_url = f"http://{req['name']}-v{dep['version']}.app.{base}"
+        else:
+            invoke_url = f"/{base_kind}/id/{stub_id}"
+        return {"deployment_id": dep["id"], "version": dep["version"], "invoke_url": invoke_url}
 
     def _gateway(self):
         # the service only holds sub-objects; recover the gateway through

@@ -52,3 +52,26 @@ async def test_rest_surface():
 
             r = await hc.get(f"{base}/api/v1/deployments", headers=headers)
             assert r.status_code == 200
+
+
+@pytest.mark.asyncio
+async def test_host_style_subdomain_routing():
+    """{name}-v{N}.app.{base} and {name}-latest.app.{base} hosts resolve to
+    deployment invoke routes (reference: subdomain.go:19-36)."""
+    async with cluster(n_workers=1) as c:
+        import handlers
+        import httpx
+
+        fn = beta9_amd.Endpoint(handlers.hello, cpu=100, memory=128)
+        d = await asyncio.to_thread(fn.deploy, "subdom")
+        base = f"http://127.0.0.1:{c.gateway.http_port}"
+        tok = c.token
+        async with httpx.AsyncClient() as client:
+            for host in ("subdom-latest.app.b9.local", f"subdom-v{d['version']}.app.b9.local"):
+                r = await client.post(
+                    base + "/",
+                    headers={"host": host, "authorization": f"Bearer {tok}"},
+                    json={"x": 3},
+                )
+                assert r.status_code == 200, (host, r.text)
+                assert r.json().get("hello") == 6 or r.json().get("result", {}).get("hello") == 6, r.text
