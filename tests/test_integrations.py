@@ -11,11 +11,13 @@ from beta9_amd.integrations.llm import LLMEngine
 
 
 def tiny_args(**kw):
-    return LLMArgs(
+    base = dict(
         model="synthetic-llama", hidden_size=64, intermediate_size=128,
         num_hidden_layers=2, num_attention_heads=4, max_model_len=256,
-        max_num_seqs=4, **kw,
+        max_num_seqs=4,
     )
+    base.update(kw)
+    return LLMArgs(**base)
 
 
 @pytest.mark.asyncio
