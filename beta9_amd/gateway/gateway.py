@@ -165,6 +165,19 @@ class Gateway:
                 inflight_bytes_cap=self.cfg.cache.inflight_bytes_cap,
             )
             self.cache_service.transport_port = await self.cache_transport.start()
+            # this node's reconciler: host-map lease + peer pulls for missing
+            # content (multi-node worker caches run the same class against
+            # the gateway coordinator)
+            from ..cache.reconciler import CacheReconciler
+
+            self.cache_reconciler = CacheReconciler(
+                self.content_store,
+                self.cache_coordinator,
+                host_id=f"gateway-{os.getpid()}",
+                addr=self.cfg.gateway.host,
+                transport_port=self.cache_service.transport_port,
+            )
+            await self.cache_reconciler.start()
         await self._start_http()
         from .loops import CronLoop, InstanceController, PoolCleaner, PoolSizer
 
@@ -320,6 +333,8 @@ class Gateway:
             await asyncio.sleep(0.1)
 
     async def shutdown(self) -> None:
+        if getattr(self, "cache_reconciler", None) is not None:
+            await self.cache_reconciler.stop()
         if not self._started:
             return
         await self.drain(timeout=self.cfg.gateway.shutdown_timeout_s)

@@ -99,3 +99,50 @@ async def test_cache_service_via_gateway():
         t = RawTransportClient("127.0.0.1", h["transport_port"])
         assert await t.fetch(digest) == data
         await t.close()
+
+
+@pytest.mark.asyncio
+async def test_reconciler_pulls_from_peer(tmp_path):
+    """Node B's reconciler resolves a digest only node A has: host-map
+    lease + RTT-ranked peer + raw-transport pull + hash verification
+    (reference analog: pkg/cache hostmap + raw transport)."""
+    from beta9_amd.cache.reconciler import CacheReconciler
+    from beta9_amd.cache.service import CacheCoordinator
+    from beta9_amd.cache.store import ContentStore
+    from beta9_amd.cache.transport import RawTransportServer
+    from beta9_amd.common.state import StateStore
+
+    store = StateStore()
+    coord = CacheCoordinator(store)
+
+    a = ContentStore(str(tmp_path / "a"))
+    b = ContentStore(str(tmp_path / "b"))
+    payload = os.urandom(9 << 20)  # multi-chunk
+    digest = a.add_bytes(payload)
+
+    srv_a = RawTransportServer(a, host="127.0.0.1", port=0)
+    await srv_a.start()
+    srv_b = RawTransportServer(b, host="127.0.0.1", port=0)
+    await srv_b.start()
+    try:
+        ra = CacheReconciler(a, coord, "node-a", "127.0.0.1", srv_a.port)
+        rb = CacheReconciler(b, coord, "node-b", "127.0.0.1", srv_b.port)
+        await ra.start()
+        await rb.start()
+        try:
+            assert b.get_path(digest) is None
+            path = await rb.ensure(digest)
+            assert path is not None and b.has(digest)
+            with open(path, "rb") as f:
+                assert f.read() == payload
+            # peers ranked, self excluded
+            peers = await rb.ranked_peers()
+            assert [p["host_id"] for p in peers] == ["node-a"]
+            # unknown digest -> None (no peer has it)
+            assert await rb.ensure("0" * 64) is None
+        finally:
+            await ra.stop()
+            await rb.stop()
+    finally:
+        await srv_a.stop()
+        await srv_b.stop()
