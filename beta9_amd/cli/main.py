@@ -377,6 +377,21 @@ def shell(container_id: str):
 
 
 @app.command()
+def usage():
+    """Workspace usage counters (task count, duration, billed cost)."""
+    import httpx
+
+    ctx = _load_context()
+    host = (ctx.get("gateway_addr") or "127.0.0.1:1993").split(":")[0]
+    r = httpx.get(
+        f"http://{host}:{ctx.get('http_port', 1994)}/api/v1/usage",
+        headers={"authorization": f"Bearer {ctx.get('token', '')}"},
+        timeout=10.0,
+    )
+    typer.echo(json.dumps(r.json(), indent=2))
+
+
+@app.command()
 def metrics():
     """Control-plane metrics snapshot."""
     typer.echo(json.dumps(_client().call("GatewayService", "GetMetrics", {}), indent=2, default=str))
