@@ -53,7 +53,7 @@ template <bool XCD_SWIZZLE, bool PARTIAL>
 __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ Bt,
     bf16_t* __restrict__ C, float* __restrict__ Cpart, int M, int N, int K,
-    int splitk) {
+    int splitk, const bf16_t* __restrict__ R) {
   // One __shared__ object only (a second one forces vmcnt(0) before every
   // ds_read of a glds pipeline — §5 trap 4a).
   __shared__ bf16_t lds[2 * 2 * BM * BK];  // [buf][A|B][128][64]
@@ -189,7 +189,9 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_nt_kernel(
         if (PARTIAL) {
           Cpart[(int64_t)slice * M * N + (int64_t)row * N + col] = acc[i][j][r];
         } else {
-          C[(int64_t)row * N + col] = (bf16_t)acc[i][j][r];
+          float v = acc[i][j][r];
+          if (R != nullptr) v += (float)R[(int64_t)row * N + col];
+          C[(int64_t)row * N + col] = (bf16_t)v;
         }
       }
     }
@@ -771,7 +773,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_kernel(
 // combine split-K fp32 partials -> bf16 (8-wide)
 __global__ __launch_bounds__(256) void splitk_reduce_kernel(
     const float* __restrict__ Cpart, bf16_t* __restrict__ C, int64_t mn,
-    int splitk) {
+    int splitk, const bf16_t* __restrict__ R) {
   int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   typedef float f32x4v __attribute__((ext_vector_type(4)));
@@ -788,6 +790,11 @@ __global__ __launch_bounds__(256) void splitk_reduce_kernel(
       }
     }
     bf16x8 o;
+    if (R != nullptr) {
+      bf16x8 rv = *(const bf16x8*)(R + idx);
+#pragma unroll
+      for (int k = 0; k < 8; k++) sum[k] += (float)((bf16_t*)&rv)[k];
+    }
 #pragma unroll
     for (int k = 0; k < 8; k++) ((bf16_t*)&o)[k] = (bf16_t)sum[k];
     *(bf16x8*)(C + idx) = o;
@@ -797,6 +804,7 @@ __global__ __launch_bounds__(256) void splitk_reduce_kernel(
     if (t < mn) {
       float sum = 0.f;
       for (int s = 0; s < splitk; s++) sum += Cpart[(int64_t)s * mn + t];
+      if (R != nullptr) sum += (float)R[t];
       C[t] = (bf16_t)sum;
     }
   }
@@ -895,8 +903,9 @@ static void check_bf16_2d(const torch::Tensor& t, const char* name) {
               " must be 2D contiguous");
 }
 
-torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
-                           bool xcd_swizzle, int64_t splitk) {
+static torch::Tensor gemm_bf16_nt_impl(torch::Tensor A, torch::Tensor Bt,
+                                       bool xcd_swizzle, int64_t splitk,
+                                       const bf16_t* Rp) {
   check_bf16_2d(A, "A");
   check_bf16_2d(Bt, "Bt");
   int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
@@ -907,10 +916,12 @@ torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
   auto C = torch::empty({M, N}, A.options());
   int64_t tiles = (M / BM) * (N / BN);
   if (splitk < 0) {
-    // auto: fill the chip (~256 CUs, 2 blocks/CU residency) for small-M
-    // serving shapes; big shapes run plain
+    // auto: fill the chip for small-M serving shapes; measured sweep
+    // (benchmarks/downproj_sweep.py): the sweet spot is ~1.5-2 blocks/CU,
+    // e.g. M=256,N=4096,K=14336 best at splitk=8 (512 blocks, 692 TF,
+    // beats hipBLASLt's 586); big shapes run plain
     splitk = 1;
-    while (splitk < 8 && tiles * splitk * 2 <= 512 && (K / BK) % (splitk * 2) == 0)
+    while (splitk < 16 && tiles * splitk < 384 && (K / BK) % (splitk * 2) == 0)
       splitk *= 2;
   }
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -922,24 +933,24 @@ torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
                          dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),
                          (const bf16_t*)Bt.data_ptr(), nullptr,
                          (float*)Cpart.data_ptr(), (int)M, (int)N, (int)K,
-                         (int)splitk);
+                         (int)splitk, nullptr);
     } else {
       hipLaunchKernelGGL((gemm_bf16_nt_kernel<false, true>), grid,
                          dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),
                          (const bf16_t*)Bt.data_ptr(), nullptr,
                          (float*)Cpart.data_ptr(), (int)M, (int)N, (int)K,
-                         (int)splitk);
+                         (int)splitk, nullptr);
     }
     int64_t mn = M * N;
     int64_t blocks = std::min<int64_t>((mn / 8 + 255) / 256, 4096);
     hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
                        stream, (const float*)Cpart.data_ptr(),
-                       (bf16_t*)C.data_ptr(), mn, (int)splitk);
+                       (bf16_t*)C.data_ptr(), mn, (int)splitk, Rp);
     return C;
   }
   // 256^2 tile when the shape divides and yields enough blocks to fill the
   // chip at 1 block/CU (double the MFMA work per staged byte)
-  if (M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 128) {
+  if (Rp == nullptr && M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 128) {
     dim3 grid2((M / BM2) * (N / BN2));
     if (xcd_swizzle) {
       hipLaunchKernelGGL((gemm_bf16_nt_256_kernel<true>), grid2,
@@ -959,14 +970,32 @@ torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
     hipLaunchKernelGGL((gemm_bf16_nt_kernel<true, false>), grid, dim3(THREADS),
                        0, stream, (const bf16_t*)A.data_ptr(),
                        (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
-                       nullptr, (int)M, (int)N, (int)K, 1);
+                       nullptr, (int)M, (int)N, (int)K, 1, Rp);
   } else {
     hipLaunchKernelGGL((gemm_bf16_nt_kernel<false, false>), grid,
                        dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),
                        (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
-                       nullptr, (int)M, (int)N, (int)K, 1);
+                       nullptr, (int)M, (int)N, (int)K, 1, Rp);
   }
   return C;
+}
+
+torch::Tensor gemm_bf16_nt(torch::Tensor A, torch::Tensor Bt,
+                           bool xcd_swizzle, int64_t splitk) {
+  return gemm_bf16_nt_impl(A, Bt, xcd_swizzle, splitk, nullptr);
+}
+
+// out = R + A @ Bt^T — the serving down-projection with its residual add
+// fused into the epilogue/reduce (saves one full [M,N] HBM round trip and
+// an elementwise kernel vs gemm-then-add).
+torch::Tensor gemm_bf16_nt_residual(torch::Tensor A, torch::Tensor Bt,
+                                    torch::Tensor R, bool xcd_swizzle,
+                                    int64_t splitk) {
+  check_bf16_2d(R, "R");
+  TORCH_CHECK(R.size(0) == A.size(0) && R.size(1) == Bt.size(0),
+              "residual must be [M, N]");
+  return gemm_bf16_nt_impl(A, Bt, xcd_swizzle, splitk,
+                           (const bf16_t*)R.data_ptr());
 }
 
 torch::Tensor gemm_bf16_nt_p8(torch::Tensor A, torch::Tensor Bt,
@@ -1079,6 +1108,9 @@ torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "hand-written gfx950 MFMA kernels: bf16 GEMM, silu_mul, rmsnorm";
+  m.def("gemm_bf16_nt_residual", &gemm_bf16_nt_residual, py::arg("A"),
+        py::arg("Bt"), py::arg("R"), py::arg("xcd_swizzle") = false,
+        py::arg("splitk") = -1);
   m.def("gemm_bf16_nt", &gemm_bf16_nt, py::arg("A"), py::arg("Bt"),
         py::arg("xcd_swizzle") = true, py::arg("splitk") = -1);
   m.def("gemm_gate_up_silu", &gemm_gate_up_silu, py::arg("A"), py::arg("Wg"),

@@ -56,6 +56,26 @@ def gemm_nt(a: torch.Tensor, w: torch.Tensor, xcd_swizzle: bool = False) -> torc
     return a @ w.t()
 
 
+def gemm_nt_residual(a: torch.Tensor, w: torch.Tensor, r: torch.Tensor,
+                     xcd_swizzle: bool = False) -> torch.Tensor:
+    """out = r + a @ w.T with the residual fused into the GEMM epilogue /
+    split-K reduce (one less [M,N] HBM round trip; measured faster than
+    hipBLASLt + separate add on the M=256 serving down-projection —
+    benchmarks/downproj_sweep.py: 43 us vs 51+5 us)."""
+    ext = _load()
+    if ext is not None and a.is_cuda:
+        M, K = a.shape
+        N = w.shape[0]
+        if N % 128 == 0 and K % 64 == 0:
+            if M % 128 != 0:
+                pad = (128 - M % 128) % 128
+                a2 = torch.nn.functional.pad(a, (0, 0, 0, pad))
+                r2 = torch.nn.functional.pad(r, (0, 0, 0, pad))
+                return ext.gemm_bf16_nt_residual(a2.contiguous(), w.contiguous(), r2.contiguous(), xcd_swizzle, -1)[:M]
+            return ext.gemm_bf16_nt_residual(a.contiguous(), w.contiguous(), r.contiguous(), xcd_swizzle, -1)
+    return r + a @ w.t()
+
+
 def mlp_gate_up(a: torch.Tensor, w_gate: torch.Tensor, w_up: torch.Tensor) -> torch.Tensor:
     """Fused act = silu(a @ w_gate.T) * (a @ w_up.T) — one hand-written
     kernel; A staged once, no gate/up intermediates in HBM."""
@@ -111,6 +131,7 @@ def warm_all(device: str = "cuda:0") -> bool:
     ext.gemm_gate_up_silu(torch.randn(1024, 64, dtype=torch.bfloat16, device=device),
                           torch.randn(1024, 64, dtype=torch.bfloat16, device=device),
                           torch.randn(1024, 64, dtype=torch.bfloat16, device=device), False)
+    ext.gemm_bf16_nt_residual(a, w, torch.randn(256, 256, dtype=torch.bfloat16, device=device), False, -1)
     ext.silu_mul(a, a)
     ext.rmsnorm(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6)
     # torch EAGER kernels the serving path touches (residual add, checksum
