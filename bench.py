@@ -87,7 +87,9 @@ async def deploy_endpoint(gw, creds, args, has_gpu: bool):
             gpus=["MI355X"] if has_gpu else [],
             gpu_count=1 if has_gpu else 0,
         ),
-        handler="beta9_amd.models.bench_handler:infer",
+        handler=("beta9_amd.models.bench_handler:infer_batched"
+                 if getattr(args, "dyn_batch", False)
+                 else "beta9_amd.models.bench_handler:infer"),
         on_start="beta9_amd.models.bench_handler:load",
         keep_warm_seconds=600.0,
         workers=1,
@@ -237,6 +239,10 @@ def main() -> int:
     p.add_argument("--inter", type=int, default=14336)
     p.add_argument("--layers", type=int, default=8)
     p.add_argument("--model", choices=["mlp", "transformer"], default="mlp")
+    p.add_argument("--dyn-batch", action="store_true",
+                   help="serve through the dynamic-batching handler "
+                        "(one coalesced forward per group of concurrent "
+                        "requests; same total tokens)")
     args = p.parse_args()
 
     rank, world, local_rank = get_dist()
@@ -276,7 +282,7 @@ def main() -> int:
                 "model": f"synthetic-{args.model}-{args.layers}L-{args.hidden}h-{args.inter}i",
                 "global_batch": args.concurrency * (world if world > 1 else 1),
                 "seq_len": args.tokens,
-                "parallelism": f"serving dp{world if world > 1 else 1} (1 GPU-container per rank)",
+                "dyn_batch": bool(getattr(args, "dyn_batch", False)), "parallelism": f"serving dp{world if world > 1 else 1} (1 GPU-container per rank)",
                 "requests_per_step_per_gpu": args.concurrency,
                 "baseline_metric": "p50 GPU-container cold-start (ms) + tasks/sec at 1/2/4/8 concurrent MI355X",
             },

@@ -43,6 +43,7 @@ else:
     )
     from .schema import Schema  # noqa: F401
     from .sdk import Bot, LLMTokenPressureAutoscaler, PricingPolicy  # noqa: F401
+    from .runner.batching import batchable  # noqa: F401
     from .types import TaskPolicy  # noqa: F401
 
 __version__ = "0.1.0"

@@ -120,3 +120,23 @@ def infer(tokens: int = 256, **kwargs) -> Dict[str, Any]:
     r = _model.infer(batch_tokens=int(tokens))
     r["load"] = _load_info
     return r
+
+
+def infer_batched(batch=None, **kw):
+    """Dynamic-batching variant: one model forward serves every coalesced
+    request (total tokens = sum of per-request tokens; no work skipped).
+    Used by `bench.py --dyn-batch`."""
+    load()
+    import torch
+
+    counts = [int((a or {}).get("tokens", 256) or 256) for a in (batch or [{}])]
+    total = sum(counts)
+    r = _model.infer(total)
+    out = []
+    for c in counts:
+        out.append({"tokens": c, "ms": r["ms"], "tflops": r["tflops"],
+                    "checksum": r["checksum"], "batched": len(counts)})
+    return out
+
+
+infer_batched._b9_batch = {"max_size": 64, "wait_ms": 2.0}

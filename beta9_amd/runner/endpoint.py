@@ -20,6 +20,7 @@ from typing import Any
 
 from .common import (
     GPUBoot,
+    unwrap_handler,
     RunnerClient,
     RunnerEnv,
     call_handler,
@@ -113,7 +114,8 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
     (build_app) for raw-protocol user code."""
     from ..rpc.channel import RpcServer, rpc as rpc_deco
 
-    state: dict = {"ready": False, "handler": None, "dispatcher": None}
+    state: dict = {"ready": False, "handler": None, "dispatcher": None,
+                   "batcher": None}
     _inputs_raw = os.environ.get("B9_INPUTS", "")
     if _inputs_raw:
         import json as _json
@@ -191,6 +193,10 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
                         state["dispatcher"].dispatch(state["handler"], args),
                         timeout=env.timeout,
                     )
+                elif state["batcher"] is not None:
+                    result = await asyncio.wait_for(
+                        state["batcher"].submit(args), timeout=env.timeout
+                    )
                 else:
                     result = await asyncio.wait_for(
                         call_handler(state["handler"], args), timeout=env.timeout
@@ -220,6 +226,14 @@ async def run_rpc_server(env: RunnerEnv, gpu: GPUBoot, rdzv=None) -> int:
     await run_on_start(env)
     boot["on_start_ms"] = round((time.perf_counter() - t1) * 1000, 2)
     state["handler"] = load_callable(env.handler)
+    bm = getattr(unwrap_handler(state["handler"]), "_b9_batch", None)
+    if bm:
+        from .batching import DynamicBatcher
+
+        state["batcher"] = DynamicBatcher(
+            state["handler"], max_size=int(bm.get("max_size", 32)),
+            wait_ms=float(bm.get("wait_ms", 2.0)),
+        )
     boot["total_ms"] = round((time.perf_counter() - t0) * 1000, 2)
     gpu.phases.update(boot)
     state["ready"] = True

@@ -63,3 +63,16 @@ def bot_answer(markers=None, **kw):
 def bot_summarize(markers=None, **kw):
     texts = [m["text"] for m in markers.get("answers", [])]
     return {"markers": {"summaries": [{"text": " | ".join(sorted(texts))}]}}
+
+
+def batched_echo(batch=None, **kw):
+    out = []
+    for args in batch:
+        pos = args.get("args") or []
+        kws = args.get("kwargs") or {}
+        x = pos[0] if pos else kws.get("x", args.get("x"))
+        out.append({"x": x, "batch_size": len(batch)})
+    return out
+
+
+batched_echo._b9_batch = {"max_size": 16, "wait_ms": 20.0}

@@ -155,3 +155,21 @@ async def test_deployment_versioning():
         deps = c.gateway.db.list_deployments(c.creds["workspace_id"])
         active = [d for d in deps if d["active"]]
         assert len(active) == 1 and active[0]["version"] == d2["version"]
+
+
+@pytest.mark.asyncio
+async def test_dynamic_batching_coalesces_requests():
+    """@batchable handlers get concurrent invokes coalesced into one
+    handler(batch=[...]) call (runner/batching.py)."""
+    async with cluster(n_workers=1) as c:
+        import handlers
+
+        ep = beta9_amd.Endpoint(handlers.batched_echo, cpu=100, memory=128,
+                                concurrent_requests=16)
+        stub_id = await asyncio.to_thread(ep.prepare_runtime)
+        outs = await asyncio.gather(*[
+            asyncio.to_thread(lambda i=i: ep.remote(x=i)) for i in range(8)
+        ])
+        assert sorted(o["x"] for o in outs) == list(range(8))
+        # at least one response saw a batch of >1 (coalescing happened)
+        assert max(o["batch_size"] for o in outs) > 1
