@@ -92,6 +92,11 @@ class Scheduler:
         self.metrics_registry = None  # MetricsRegistry
         self._loop_task: Optional[asyncio.Task] = None
         self._wake = asyncio.Event()
+        # provisioning tracker (reference: reserve.go:343-501
+        # provisioningTracker): request -> (pending worker id, ts) so retries
+        # of an already-provisioned request don't spawn duplicate workers
+        self._reservations: Dict[str, tuple] = {}
+        self.reservation_ttl_s = 60.0
 
     def register_pool(self, pool: PoolController) -> None:
         self.pools[pool.name] = pool
@@ -231,6 +236,7 @@ class Scheduler:
         await self._no_fit(request, workers)
 
     async def _committed(self, request: ContainerRequest, worker: Worker) -> None:
+        self._reservations.pop(request.container_id, None)
         await self.store.delete(keys.scheduler_request(request.container_id))
         self.metrics.scheduled += 1
         wait_ms = (time.time() - request.timestamp) * 1000.0
@@ -295,8 +301,11 @@ class Scheduler:
             and (not request.requires_gpu() or (w.gpu in request.gpu_request and w.free_gpu_count >= request.gpu_count))
             for w in workers
         )
-        provisioned = False
-        if not pending_ok:
+        res = self._reservations.get(request.container_id)
+        if res is not None and time.time() - res[1] > self.reservation_ttl_s:
+            self._reservations.pop(request.container_id, None)
+            res = None
+        if not pending_ok and res is None:
             pool = self._pool_for(request)
             if pool is not None:
                 wid = await pool.add_worker(
@@ -304,9 +313,9 @@ class Scheduler:
                     memory_mb=max(request.memory_mb, self.cfg.worker.memory_limit_mb // 4),
                     gpu_count=request.gpu_count,
                 )
-                provisioned = wid is not None
-                if provisioned:
+                if wid is not None:
                     self.metrics.provisioned += 1
+                    self._reservations[request.container_id] = (wid, time.time())
         await self._retry(request)
 
     def _pool_for(self, request: ContainerRequest) -> Optional[PoolController]:

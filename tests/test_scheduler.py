@@ -185,3 +185,26 @@ async def test_gpu_failover_chain():
     # an MI300X worker is now eligible, ranked behind MI355X
     mi300 = make_worker("b", gpus=8, gpu="MI300X")
     assert s._filter(req, [mi300]) == [mi300]
+
+
+@pytest.mark.asyncio
+async def test_reservation_prevents_duplicate_provisioning():
+    """Retries of an already-provisioned request must not spawn duplicate
+    workers (reference: reserve.go provisioningTracker)."""
+    s, wr, cr, store = await make_sched()
+    pool = FakePoolController()
+    s.pools = {"default": pool}
+    s.cfg.scheduler.retry_backoff_base_s = 0.01
+    await s.start()
+    try:
+        req = ContainerRequest(container_id="c-res", cpu=1000, memory_mb=256)
+        await s.run(req)
+        await asyncio.sleep(0.5)  # several retry rounds
+        assert s.metrics.retries >= 2
+        assert len(pool.added) == 1  # one reservation, no duplicates
+        await wr.register(make_worker("late-res"))
+        d = await store.blpop(keys.worker_request_queue("late-res"), 3.0)
+        assert d is not None
+        assert "c-res" not in s._reservations  # cleared on commit
+    finally:
+        await s.shutdown()
