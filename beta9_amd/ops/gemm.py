@@ -152,5 +152,10 @@ def warm_all(device: str = "cuda:0") -> bool:
     f.to(torch.bfloat16)
     torch.randn(64, 64, device=device, dtype=torch.bfloat16)
     torch.zeros(64, device=device)
+    # attention path (transformer models): SDPA backend selection + kernels
+    q = torch.randn(4, 64, 64, device=device, dtype=torch.bfloat16)
+    torch.nn.functional.scaled_dot_product_attention(q, q, q, is_causal=True)
+    torch.nn.functional.softmax(q.float(), dim=-1)
+    q.transpose(1, 2).contiguous()
     torch.cuda.synchronize()
     return True
