@@ -24,11 +24,12 @@ from beta9_amd.worker.worker import Worker
 class Cluster:
     """Gateway + N embedded workers in one event loop."""
 
-    def __init__(self, tmpdir: str, gpu_workers: int = 0, fake_gpus: int = 0):
+    def __init__(self, tmpdir: str, gpu_workers: int = 0, fake_gpus: int = 0,
+                 db_path: str = ":memory:"):
         cfg = AppConfig().with_root(tmpdir)
         cfg.gateway.grpc_port = 0
         cfg.gateway.http_port = 0
-        cfg.database.path = ":memory:"
+        cfg.database.path = db_path
         self.cfg = cfg
         self.gateway = Gateway(cfg)
         self.workers: list[Worker] = []
@@ -85,9 +86,9 @@ class Cluster:
 
 
 @contextlib.asynccontextmanager
-async def cluster(n_workers: int = 1, fake_gpus: int = 0, **kw):
+async def cluster(n_workers: int = 1, fake_gpus: int = 0, db_path: str = ":memory:", **kw):
     with tempfile.TemporaryDirectory() as tmp:
-        c = Cluster(tmp, fake_gpus=fake_gpus)
+        c = Cluster(tmp, fake_gpus=fake_gpus, db_path=db_path)
         await c.start(n_workers=n_workers, **kw)
         try:
             yield c

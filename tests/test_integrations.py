@@ -112,3 +112,46 @@ async def test_llm_engine_gpu_bf16_decode():
         assert len(outs) == 8
     finally:
         await eng.stop()
+
+
+@pytest.mark.asyncio
+async def test_llm_served_through_platform():
+    """The LLM integration deployed as an asgi stub: OpenAI-surface requests
+    route gateway HTTP -> endpoint buffer -> uvicorn in the container."""
+    import json as _json
+
+    from tests.helpers import cluster
+
+    async with cluster(n_workers=1) as c:
+        import llm_app  # fixture module: builds a tiny synthetic LLM app
+
+        import beta9_amd
+
+        ep = beta9_amd.Endpoint(llm_app.make_app, cpu=1000, memory=2048)
+        ep.stub_kind = "asgi"
+        stub_id = await asyncio.to_thread(ep.prepare_runtime)
+        # wait for the uvicorn container to come up and serve
+        base = f"http://127.0.0.1:{c.gateway.http_port}"
+        headers = {"authorization": f"Bearer {c.token}"}
+        deadline = 60.0
+        async with httpx.AsyncClient() as hc:
+            r = None
+            import time as _time
+
+            t0 = _time.monotonic()
+            while _time.monotonic() - t0 < deadline:
+                r = await hc.post(
+                    f"{base}/asgi/id/{stub_id}/v1/completions",
+                    headers=headers,
+                    json={"prompt": "hi", "max_tokens": 3},
+                    timeout=30.0,
+                )
+                if r.status_code == 200:
+                    break
+                await asyncio.sleep(1.0)
+            assert r is not None and r.status_code == 200, r.text if r else "no response"
+            body = r.json()
+            payload = body.get("result", body)
+            if isinstance(payload, str):
+                payload = _json.loads(payload)
+            assert payload["object"] == "text_completion", payload
