@@ -77,12 +77,21 @@ def gemm_nt_residual(a: torch.Tensor, w: torch.Tensor, r: torch.Tensor,
 
 
 def mlp_gate_up(a: torch.Tensor, w_gate: torch.Tensor, w_up: torch.Tensor) -> torch.Tensor:
-    """Fused act = silu(a @ w_gate.T) * (a @ w_up.T) — one hand-written
-    kernel; A staged once, no gate/up intermediates in HBM."""
+    """act = silu(a @ w_gate.T) * (a @ w_up.T).
+
+    Serving M (<2048): one fused kernel — A staged once, no gate/up
+    intermediates in HBM (925 TF at M=256, 1.5x the library path).
+    Large M: two 256-tile GEMMs + silu_mul win instead (the fused 128-row
+    tile runs out of compute-per-staged-byte; measured 919 vs 878 TF at
+    M=2048 — docs/BENCHMARKS.md phase table)."""
     ext = _load()
     if ext is not None and a.is_cuda:
         M, K = a.shape
         N = w_gate.shape[0]
+        if M >= 2048 and M % 256 == 0 and N % 256 == 0 and K % 64 == 0:
+            gate = ext.gemm_bf16_nt(a, w_gate, False, 1)
+            up = ext.gemm_bf16_nt(a, w_up, False, 1)
+            return ext.silu_mul(gate, up)
         if N % 128 == 0 and K % 64 == 0:
             if M % 128 != 0:
                 pad = (128 - M % 128) % 128
