@@ -1048,6 +1048,143 @@ torch::Tensor gemm_bf16_nt_p2(torch::Tensor A, torch::Tensor Bt,
   return C;
 }
 
+
+// ---------------------------------------------------------------------------
+// 256-row fused gate+up+SiLU for LARGE M (prefill-style shapes): same idea
+// as the 128-row kernel but with the 256^2-GEMM pipeline economics — the A
+// tile is 256 rows staged ONCE for BOTH projections, doubling MFMA work per
+// staged byte. 8 waves (2M x 4N, each 128m x 32n of gate AND up), BNT=128,
+// LDS (256+128+128)x64 bf16 double-buffered = 128 KiB -> 1 block/CU.
+// ---------------------------------------------------------------------------
+
+template <bool XCD_SWIZZLE>
+__global__ __launch_bounds__(THREADS2, 1) void gemm_gate_up_silu_256_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ Wg,
+    const bf16_t* __restrict__ Wu, bf16_t* __restrict__ Out, int M, int N,
+    int K) {
+  constexpr int BNT = 128;  // block output columns (4 waves x 32)
+  __shared__ bf16_t lds[2 * (BM2 + 2 * BNT) * BK];  // [buf][A|Wg|Wu]
+
+  int nwg_n = N / BNT;
+  int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int tile_m = (wg / nwg_n) * BM2;
+  int tile_n = (wg % nwg_n) * BNT;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 2, wc = wave & 3;  // wave: 128m x 32n
+
+  int ld_row = lane >> 3;
+  int ld_chunk = lane & 7;
+  auto stage = [&](int buf, int kt) {
+    bf16_t* Ab = lds + buf * (BM2 + 2 * BNT) * BK;
+    bf16_t* Gb = Ab + BM2 * BK;
+    bf16_t* Ub = Gb + BNT * BK;
+    const bf16_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
+    const bf16_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK;
+    const bf16_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK;
+#pragma unroll
+    for (int call = 0; call < 4; call++) {  // A: 32 pieces over 8 waves
+      int p = wave + WAVES2 * call;
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 512), 16, 0, 0);
+    }
+#pragma unroll
+    for (int call = 0; call < 2; call++) {  // each W image: 16 pieces
+      int p = wave + WAVES2 * call;
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 512), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 8)),
+          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 512), 16, 0, 0);
+    }
+  };
+
+  f32x4 acc_g[8][2] = {};
+  f32x4 acc_u[8][2] = {};
+
+  int ntiles = K / BK;
+  stage(0, 0);
+
+  int fr_row = lane & 15;
+  int fr_q = lane >> 4;
+
+  for (int kt = 0; kt < ntiles; kt++) {
+    int buf = kt & 1;
+    bf16_t* Ab = lds + buf * (BM2 + 2 * BNT) * BK;
+    bf16_t* Gb = Ab + BM2 * BK;
+    bf16_t* Ub = Gb + BNT * BK;
+    if (kt + 1 < ntiles) {
+      stage(buf ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // 8 glds per wave
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ks++) {
+      bf16x8 a_frag[8], g_frag[2], u_frag[2];
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        int row = wr * 128 + i * 16 + fr_row;
+        int sc = (ks * 4 + fr_q) ^ (row & 7);
+        a_frag[i] = *(const bf16x8*)(Ab + row * BK + sc * 8);
+      }
+#pragma unroll
+      for (int j = 0; j < 2; j++) {
+        int col = wc * 32 + j * 16 + fr_row;
+        int scb = (ks * 4 + fr_q) ^ (col & 7);
+        g_frag[j] = *(const bf16x8*)(Gb + col * BK + scb * 8);
+        u_frag[j] = *(const bf16x8*)(Ub + col * BK + scb * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 8; i++)
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+          acc_g[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], g_frag[j], acc_g[i][j], 0, 0, 0);
+          acc_u[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], u_frag[j], acc_u[i][j], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // fused epilogue: silu(gate) * up
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+#pragma unroll
+    for (int j = 0; j < 2; j++) {
+      int col = tile_n + wc * 32 + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 128 + i * 16 + c_row0 + r;
+        float g = acc_g[i][j][r];
+        float u = acc_u[i][j][r];
+        float act = g / (1.0f + __expf(-g)) * u;
+        Out[(int64_t)row * N + col] = (bf16_t)act;
+      }
+    }
+  }
+}
+
 torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
                                 torch::Tensor Wu, bool xcd_swizzle) {
   check_bf16_2d(A, "A");
@@ -1058,10 +1195,17 @@ torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
   auto Out = torch::empty({M, N}, A.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  // BN=128 tile when it already fills the chip; BN=64 (2 blocks/CU)
-  // otherwise (M=256 serving shapes)
+  // 256-row tile for large M (A staged once for both projections at
+  // double the compute per staged byte); BN=128 when it fills the chip;
+  // BN=64 (2 blocks/CU) otherwise (M=256 serving shapes)
   int64_t tiles128 = (M / BM) * (N / 128);
-  if (tiles128 >= 256 || N % 64 != 0) {
+  if (M % BM2 == 0 && N % 128 == 0 && (M / BM2) * (N / 128) >= 256) {
+    dim3 grid((M / BM2) * (N / 128));
+    hipLaunchKernelGGL((gemm_gate_up_silu_256_kernel<false>), grid,
+                       dim3(THREADS2), 0, stream, (const bf16_t*)A.data_ptr(),
+                       (const bf16_t*)Wg.data_ptr(), (const bf16_t*)Wu.data_ptr(),
+                       (bf16_t*)Out.data_ptr(), (int)M, (int)N, (int)K);
+  } else if (tiles128 >= 256 || N % 64 != 0) {
     dim3 grid((M / BM) * (N / 128));
     hipLaunchKernelGGL((gemm_gate_up_silu_kernel<false, 4>), grid,
                        dim3(THREADS), 0, stream, (const bf16_t*)A.data_ptr(),

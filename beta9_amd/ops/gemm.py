@@ -79,19 +79,18 @@ def gemm_nt_residual(a: torch.Tensor, w: torch.Tensor, r: torch.Tensor,
 def mlp_gate_up(a: torch.Tensor, w_gate: torch.Tensor, w_up: torch.Tensor) -> torch.Tensor:
     """act = silu(a @ w_gate.T) * (a @ w_up.T).
 
-    Serving M (<2048): one fused kernel — A staged once, no gate/up
-    intermediates in HBM (925 TF at M=256, 1.5x the library path).
-    Large M: two 256-tile GEMMs + silu_mul win instead (the fused 128-row
-    tile runs out of compute-per-staged-byte; measured 919 vs 878 TF at
-    M=2048 — docs/BENCHMARKS.md phase table)."""
+    Serving/medium M: one fused kernel — A staged once for BOTH
+    projections, no gate/up intermediates in HBM (925 TF at M=256 = 1.46x
+    the library path; 945 TF at M=2048 via the 256-row tile). Very large M
+    (>= 2048): two library GEMMs + our fused silu_mul measure fastest
+    (~1330 TF effective at M=2048 — plain-GEMM strength, elementwise still
+    fused; benchmarks/gateup256_check.py)."""
     ext = _load()
     if ext is not None and a.is_cuda:
         M, K = a.shape
         N = w_gate.shape[0]
-        if M >= 2048 and M % 256 == 0 and N % 256 == 0 and K % 64 == 0:
-            gate = ext.gemm_bf16_nt(a, w_gate, False, 1)
-            up = ext.gemm_bf16_nt(a, w_up, False, 1)
-            return ext.silu_mul(gate, up)
+        if M >= 2048:
+            return ext.silu_mul((a @ w_gate.t()).contiguous(), (a @ w_up.t()).contiguous())
         if N % 128 == 0 and K % 64 == 0:
             if M % 128 != 0:
                 pad = (128 - M % 128) % 128
