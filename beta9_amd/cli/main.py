@@ -338,6 +338,17 @@ def worker_list():
 @worker_app.command("cordon")
 def worker_cordon(worker_id: str, uncordon: bool = typer.Option(False)):
     _client().call("GatewayService", "CordonWorker", {"worker_id": worker_id, "cordoned": not uncordon})
+
+
+@worker_app.command("drain")
+def worker_drain(worker_id: str, timeout: float = typer.Option(120.0), remove: bool = typer.Option(False)):
+    """Cordon, wait for active containers to finish, optionally remove."""
+    r = _client().call(
+        "GatewayService", "DrainWorker",
+        {"worker_id": worker_id, "timeout": timeout, "remove": remove},
+        timeout=timeout + 30.0,
+    )
+    typer.echo(json.dumps(r, indent=2, default=str))
     typer.echo("ok")
 
 
