@@ -146,3 +146,31 @@ async def test_reconciler_pulls_from_peer(tmp_path):
     finally:
         await srv_a.stop()
         await srv_b.stop()
+
+
+@pytest.mark.asyncio
+async def test_reconciler_skips_dead_peer(tmp_path):
+    """A registered-but-dead peer is ranked out by the RTT probe; content
+    still resolves from the live one."""
+    from beta9_amd.cache.reconciler import CacheReconciler
+    from beta9_amd.cache.service import CacheCoordinator
+    from beta9_amd.cache.store import ContentStore
+    from beta9_amd.cache.transport import RawTransportServer
+    from beta9_amd.common.state import StateStore
+
+    coord = CacheCoordinator(StateStore())
+    a = ContentStore(str(tmp_path / "a"))
+    b = ContentStore(str(tmp_path / "b"))
+    digest = a.add_bytes(b"live-content" * 1000)
+    srv_a = RawTransportServer(a, host="127.0.0.1", port=0)
+    await srv_a.start()
+    try:
+        # dead peer: registered with a port nobody listens on
+        await coord.register("dead", "127.0.0.1", 1)
+        await coord.register("node-a", "127.0.0.1", srv_a.port)
+        rb = CacheReconciler(b, coord, "node-b", "127.0.0.1", 0)
+        peers = await rb.ranked_peers()
+        assert [p["host_id"] for p in peers] == ["node-a"]
+        assert await rb.ensure(digest) is not None
+    finally:
+        await srv_a.stop()

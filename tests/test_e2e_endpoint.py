@@ -173,3 +173,29 @@ async def test_dynamic_batching_coalesces_requests():
         assert sorted(o["x"] for o in outs) == list(range(8))
         # at least one response saw a batch of >1 (coalescing happened)
         assert max(o["batch_size"] for o in outs) > 1
+
+
+@pytest.mark.asyncio
+async def test_dynamic_batcher_unit_timeout_and_errors():
+    """DynamicBatcher: wait_ms flush with partial batch; handler errors
+    propagate to every waiter."""
+    from beta9_amd.runner.batching import DynamicBatcher
+
+    calls = []
+
+    def handler(batch=None):
+        calls.append(len(batch))
+        if batch and batch[0] == "boom":
+            raise ValueError("nope")
+        return [f"r{a}" for a in batch]
+
+    b = DynamicBatcher(handler, max_size=8, wait_ms=10.0)
+    # partial batch flushes on the timer
+    r = await asyncio.gather(b.submit("a"), b.submit("b"))
+    assert r == ["ra", "rb"] and calls == [2]
+    # errors propagate
+    with pytest.raises(ValueError):
+        await b.submit("boom")
+    # max_size triggers an immediate flush
+    rs = await asyncio.gather(*[b.submit(i) for i in range(8)])
+    assert rs == [f"r{i}" for i in range(8)]
