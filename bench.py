@@ -175,7 +175,7 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
         if use_dist:
             import torch.distributed as dist
 
-            if torch.cuda.is_available():
+            if torch.cuda.is_available() and local_rank < torch.cuda.device_count():
                 torch.cuda.set_device(local_rank)
             dist.barrier()
         if torch.cuda.is_available():
