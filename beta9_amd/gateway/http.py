@@ -31,6 +31,21 @@ def build_http_app(gateway) -> FastAPI:
             raise HTTPException(status_code=401, detail="invalid token")
         return {"workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
 
+    def _public_auth(stub_id: str, request: Request) -> Optional[dict]:
+        """authorized=False stubs are publicly invokable: the request is
+        accounted to the stub's own workspace (reference: StubConfigV1
+        .authorized + public invoke URLs)."""
+        stub = gateway.db.get_stub(stub_id)
+        if stub is None:
+            return None
+        try:
+            cfg = gateway.db.get_stub_config(stub_id)
+        except Exception:
+            return None
+        if cfg is not None and getattr(cfg, "authorized", True) is False:
+            return {"workspace_id": stub["workspace_id"], "workspace_name": "public"}
+        return None
+
     @app.get("/api/v1/health")
     async def health():
         return {"status": "ok"}
@@ -98,13 +113,19 @@ def build_http_app(gateway) -> FastAPI:
             )
         return JSONResponse({"task_id": task.task_id, "result": result})
 
+    async def _auth_or_public(stub_id: str, request: Request) -> dict:
+        pub = _public_auth(stub_id, request)
+        if pub is not None:
+            return pub
+        return await auth_ctx(request)
+
     @app.api_route("/endpoint/id/{stub_id}", methods=["GET", "POST"])
-    async def invoke_endpoint(stub_id: str, request: Request, auth: dict = Depends(auth_ctx)):
-        return await _invoke(stub_id, request, auth)
+    async def invoke_endpoint(stub_id: str, request: Request):
+        return await _invoke(stub_id, request, await _auth_or_public(stub_id, request))
 
     @app.api_route("/endpoint/id/{stub_id}/{path:path}", methods=["GET", "POST"])
-    async def invoke_endpoint_path(stub_id: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):
-        return await _invoke(stub_id, request, auth, path="/" + path)
+    async def invoke_endpoint_path(stub_id: str, path: str, request: Request):
+        return await _invoke(stub_id, request, await _auth_or_public(stub_id, request), path="/" + path)
 
     @app.api_route("/asgi/id/{stub_id}/{path:path}", methods=["GET", "POST"])
     async def invoke_asgi(stub_id: str, path: str, request: Request, auth: dict = Depends(auth_ctx)):

@@ -356,6 +356,7 @@ class RunnerAbstraction:
         inputs=None,  # Schema
         outputs=None,  # Schema
         pricing: Optional["PricingPolicy"] = None,
+        authorized: bool = True,
     ):
         self.func = func
         self.cpu = cpu
@@ -373,6 +374,7 @@ class RunnerAbstraction:
         self.volumes = volumes or []
         self.secrets = secrets or []
         self.pricing = pricing
+        self.authorized = authorized
         self.env = env or {}
         self.autoscaler = autoscaler
         self.checkpoint_enabled = checkpoint_enabled
@@ -430,6 +432,7 @@ class RunnerAbstraction:
             inputs=self.inputs.to_spec() if self.inputs is not None else None,
             outputs=self.outputs.to_spec() if self.outputs is not None else None,
             pricing=self.pricing.to_config() if self.pricing is not None else None,
+            authorized=self.authorized,
         )
         return cfg
 

@@ -75,3 +75,24 @@ async def test_host_style_subdomain_routing():
                 )
                 assert r.status_code == 200, (host, r.text)
                 assert r.json().get("hello") == 6 or r.json().get("result", {}).get("hello") == 6, r.text
+
+
+@pytest.mark.asyncio
+async def test_public_stub_invokable_without_token():
+    """authorized=False endpoints accept anonymous invokes, accounted to
+    the stub's workspace; authorized stubs still 401."""
+    async with cluster(n_workers=1) as c:
+        import handlers
+
+        pub = beta9_amd.Endpoint(handlers.hello, cpu=100, memory=128, authorized=False)
+        pub_id = await asyncio.to_thread(pub.prepare_runtime)
+        priv = beta9_amd.Endpoint(handlers.hello, cpu=100, memory=128, name="priv")
+        priv_id = await asyncio.to_thread(priv.prepare_runtime)
+        base = f"http://127.0.0.1:{c.gateway.http_port}"
+        async with httpx.AsyncClient() as hc:
+            r = await hc.post(f"{base}/endpoint/id/{pub_id}", json={"x": 4}, timeout=60.0)
+            assert r.status_code == 200, r.text
+            body = r.json()
+            assert body.get("hello") == 8 or body.get("result", {}).get("hello") == 8
+            r = await hc.post(f"{base}/endpoint/id/{priv_id}", json={"x": 4})
+            assert r.status_code == 401
