@@ -207,6 +207,30 @@ class BackendDB:
         rows = self._query("SELECT * FROM workspaces WHERE name=?", (name,))
         return dict(rows[0]) if rows else None
 
+    def get_or_create_signing_key(self, workspace_id: str) -> str:
+        """Per-workspace HMAC key for signed public-endpoint payloads
+        (reference: auth SignPayload surface)."""
+        self._exec(
+            "CREATE TABLE IF NOT EXISTS workspace_keys "
+            "(workspace_id TEXT PRIMARY KEY, key TEXT)", ()
+        )
+        rows = self._query(
+            "SELECT key FROM workspace_keys WHERE workspace_id=?", (workspace_id,)
+        )
+        if rows:
+            return rows[0]["key"]
+        import secrets as _secrets
+
+        key = _secrets.token_hex(32)
+        self._exec(
+            "INSERT OR IGNORE INTO workspace_keys (workspace_id, key) VALUES (?,?)",
+            (workspace_id, key),
+        )
+        rows = self._query(
+            "SELECT key FROM workspace_keys WHERE workspace_id=?", (workspace_id,)
+        )
+        return rows[0]["key"]
+
     def create_token(self, workspace_id: str) -> str:
         key = uuid.uuid4().hex + uuid.uuid4().hex[:8]
         self._exec(

@@ -31,7 +31,7 @@ def build_http_app(gateway) -> FastAPI:
             raise HTTPException(status_code=401, detail="invalid token")
         return {"workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
 
-    def _public_auth(stub_id: str, request: Request) -> Optional[dict]:
+    async def _public_auth(stub_id: str, request: Request) -> Optional[dict]:
         """authorized=False stubs are publicly invokable: the request is
         accounted to the stub's own workspace (reference: StubConfigV1
         .authorized + public invoke URLs)."""
@@ -43,8 +43,30 @@ def build_http_app(gateway) -> FastAPI:
         except Exception:
             return None
         if cfg is not None and getattr(cfg, "authorized", True) is False:
+            if (cfg.extra or {}).get("require_signature"):
+                sig = request.headers.get("x-b9-signature", "")
+                if not await _verify_signature(stub["workspace_id"], sig, request):
+                    raise HTTPException(status_code=401, detail="bad signature")
             return {"workspace_id": stub["workspace_id"], "workspace_name": "public"}
         return None
+
+    async def _verify_signature(workspace_id: str, sig: str, request: Request, max_age_s: float = 300.0) -> bool:
+        import hashlib
+        import hmac as _hmac
+        import time as _time
+
+        if not sig.startswith("v1="):
+            return False
+        try:
+            ts, mac = sig[3:].split(":", 1)
+        except ValueError:
+            return False
+        if abs(_time.time() - float(ts)) > max_age_s:
+            return False
+        key = gateway.db.get_or_create_signing_key(workspace_id).encode()
+        body = await request.body()  # cached by starlette for the route
+        want = _hmac.new(key, ts.encode() + b"." + body, hashlib.sha256).hexdigest()
+        return _hmac.compare_digest(want, mac)
 
     @app.get("/api/v1/health")
     async def health():
@@ -114,7 +136,7 @@ def build_http_app(gateway) -> FastAPI:
         return JSONResponse({"task_id": task.task_id, "result": result})
 
     async def _auth_or_public(stub_id: str, request: Request) -> dict:
-        pub = _public_auth(stub_id, request)
+        pub = await _public_auth(stub_id, request)
         if pub is not None:
             return pub
         return await auth_ctx(request)

@@ -55,6 +55,23 @@ class GatewayService:
             return {"ok": False}
         return {"ok": True, "workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"]}
 
+    @rpc()
+    async def SignPayload(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """HMAC-sign a payload with the workspace signing key so public
+        (authorized=False) endpoints can verify callers (reference:
+        auth SignPayload RPC). Signature format: v1=<unix_ts>:<hex hmac of
+        ts.payload>."""
+        import hashlib
+        import hmac as _hmac
+
+        payload = req.get("payload", b"")
+        if isinstance(payload, str):
+            payload = payload.encode()
+        key = self.db.get_or_create_signing_key(auth["workspace_id"]).encode()
+        ts = str(int(time.time()))
+        sig = _hmac.new(key, ts.encode() + b"." + payload, hashlib.sha256).hexdigest()
+        return {"signature": f"v1={ts}:{sig}", "expires_in_s": 300}
+
     # -- workspaces / tokens (ops surface) -----------------------------------
     @rpc()
     async def CreateWorkspace(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:

@@ -96,3 +96,46 @@ async def test_public_stub_invokable_without_token():
             assert body.get("hello") == 8 or body.get("result", {}).get("hello") == 8
             r = await hc.post(f"{base}/endpoint/id/{priv_id}", json={"x": 4})
             assert r.status_code == 401
+
+
+@pytest.mark.asyncio
+async def test_signed_public_payloads():
+    """SignPayload RPC + require_signature public stubs: a signed request
+    passes, unsigned/garbage is 401."""
+    import json as _json
+
+    async with cluster(n_workers=1) as c:
+        import handlers
+
+        ep = beta9_amd.Endpoint(handlers.hello, cpu=100, memory=128, authorized=False)
+        stub_id = await asyncio.to_thread(ep.prepare_runtime)
+        cfg = c.gateway.db.get_stub_config(stub_id)
+        cfg.extra["require_signature"] = True
+        c.gateway.db.update_stub_config(stub_id, cfg) if hasattr(c.gateway.db, "update_stub_config") else None
+        # fall back: mutate the row directly if no update helper
+        if not hasattr(c.gateway.db, "update_stub_config"):
+            c.gateway.db._exec(
+                "UPDATE stubs SET config=? WHERE id=?",
+                (cfg.model_dump_json(), stub_id),
+            )
+        # bust caches
+        c.gateway.dispatcher._cfg_cache = {}
+
+        body = _json.dumps({"x": 7}).encode()
+        client = beta9_amd.Client()
+        signed = await asyncio.to_thread(
+            client.call, "GatewayService", "SignPayload", {"payload": body}
+        )
+        base = f"http://127.0.0.1:{c.gateway.http_port}"
+        async with httpx.AsyncClient() as hc:
+            r = await hc.post(f"{base}/endpoint/id/{stub_id}", content=body,
+                              headers={"x-b9-signature": signed["signature"],
+                                       "content-type": "application/json"}, timeout=60.0)
+            assert r.status_code == 200, r.text
+            r = await hc.post(f"{base}/endpoint/id/{stub_id}", content=body,
+                              headers={"content-type": "application/json"})
+            assert r.status_code == 401
+            r = await hc.post(f"{base}/endpoint/id/{stub_id}", content=body,
+                              headers={"x-b9-signature": "v1=1:deadbeef",
+                                       "content-type": "application/json"})
+            assert r.status_code == 401
