@@ -42,7 +42,7 @@ else:
         task_queue,
     )
     from .schema import Schema  # noqa: F401
-    from .sdk import Bot, LLMTokenPressureAutoscaler, PricingPolicy  # noqa: F401
+    from .sdk import Bot, LLMConfig, LLMTokenPressureAutoscaler, PricingPolicy, Service  # noqa: F401
     from .runner.batching import batchable  # noqa: F401
     from .types import TaskPolicy  # noqa: F401
 

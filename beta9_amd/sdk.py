@@ -668,6 +668,63 @@ class Pod(RunnerAbstraction):
 
 
 @dataclass
+class LLMConfig:
+    """LLM serving metadata for Service pods (reference: type.py:314
+    LLMConfig): drives OpenAI-protocol routing (prefix affinity) and the
+    token-pressure autoscaler target."""
+
+    model_id: str = ""
+    engine: str = ""
+    served_model_name: str = ""
+    context_length: int = 0
+    metrics_path: str = "/metrics"
+    slo_tier: str = ""
+
+
+class Service(Pod):
+    """Long-running serving pod (reference: abstractions/service.py:161
+    Service(Pod)): a pod with a serving port, readiness semantics, and
+    optional LLM config. The platform applies prefix-affinity routing and
+    the LLM token-pressure autoscaler to llm-kind services.
+
+    Usage::
+
+        svc = Service(entrypoint=["python", "-m", "my_server"],
+                      port=8000, llm=LLMConfig(model_id="my-model"),
+                      gpu="MI355X")
+        inst = svc.create()
+    """
+
+    stub_kind = "pod"
+
+    def __init__(self, entrypoint: Optional[List[str]] = None, port: int = 8000,
+                 llm: Optional[LLMConfig] = None, app_kind: str = "",
+                 keep_warm_seconds: float = 600.0, **kwargs):
+        super().__init__(entrypoint=entrypoint, ports=[port],
+                         keep_warm_seconds=keep_warm_seconds, **kwargs)
+        self.port = port
+        self.llm = llm
+        self.app_kind = app_kind or ("llm" if llm else "")
+        self.name = kwargs.get("name") or "service"
+
+    def _stub_config(self) -> StubConfig:
+        cfg = super()._stub_config()
+        cfg.keep_warm_seconds = self.keep_warm_seconds
+        serving = {"app_kind": self.app_kind, "port": self.port}
+        if self.llm is not None:
+            serving["llm"] = {
+                "model_id": self.llm.model_id,
+                "engine": self.llm.engine,
+                "served_model_name": self.llm.served_model_name,
+                "context_length": self.llm.context_length,
+                "metrics_path": self.llm.metrics_path,
+                "slo_tier": self.llm.slo_tier,
+            }
+        cfg.extra["serving"] = serving
+        return cfg
+
+
+@dataclass
 class PodInstance:
     client: Client
     container_id: str

@@ -199,3 +199,23 @@ async def test_dynamic_batcher_unit_timeout_and_errors():
     # max_size triggers an immediate flush
     rs = await asyncio.gather(*[b.submit(i) for i in range(8)])
     assert rs == [f"r{i}" for i in range(8)]
+
+
+@pytest.mark.asyncio
+async def test_service_pod_with_llm_config():
+    """Service(Pod): serving config recorded on the stub; pod boots and
+    exposes its port (reference: abstractions/service.py Service(Pod))."""
+    async with cluster(n_workers=1) as c:
+        svc = beta9_amd.Service(
+            port=8123, cpu=100, memory=128,
+            llm=beta9_amd.LLMConfig(model_id="synthetic-llama", context_length=2048),
+        )
+        inst = await asyncio.to_thread(svc.create)
+        try:
+            stub_id = svc._stub_id
+            cfg = c.gateway.db.get_stub_config(stub_id)
+            assert cfg.extra["serving"]["app_kind"] == "llm"
+            assert cfg.extra["serving"]["llm"]["model_id"] == "synthetic-llama"
+            assert cfg.ports == [8123]
+        finally:
+            await asyncio.to_thread(inst.terminate) if hasattr(inst, "terminate") else None
