@@ -106,7 +106,7 @@ class RequestBuffer:
                 if slot.rpc is None:
                     from ...rpc.channel import RpcClient
 
-                    slot.rpc = RpcClient(slot.addr)
+                    slot.rpc = RpcClient(slot.addr, private_pool=True)
                 r = await slot.rpc.call("Runner", "Health", {}, timeout=2.0)
                 return bool(r.get("ok"))
             except Exception:

@@ -201,9 +201,10 @@ def rpc(kind: str = "unary_unary", name: Optional[str] = None, open: bool = Fals
 class RpcClient:
     """Client channel for msgpack generic services."""
 
-    def __init__(self, target: str, token: str = ""):
+    def __init__(self, target: str, token: str = "", private_pool: bool = False):
         self.target = target
         self._token = token
+        self._private_pool = private_pool
         self._channel: Optional[grpc.aio.Channel] = None
 
     def _md(self):
@@ -224,14 +225,21 @@ class RpcClient:
                     ("grpc.initial_reconnect_backoff_ms", 20),
                     ("grpc.min_reconnect_backoff_ms", 20),
                     ("grpc.max_reconnect_backoff_ms", 500),
-                    # ...and backoff state lives on the SUBCHANNEL, which the
-                    # GLOBAL pool shares across channels to the same target:
-                    # recreating the channel would still reuse the backed-off
+                ]
+                + (
+                    # backoff state lives on the SUBCHANNEL, which the GLOBAL
+                    # pool shares across channels to the same target:
+                    # recreating a channel would still reuse the backed-off
                     # subchannel (measured: ~90 ms of refused probes after the
-                    # runner was live, benchmarks/cold_trace.py). A private
-                    # pool makes every fresh channel connect immediately.
-                    ("grpc.use_local_subchannel_pool", 1),
-                ],
+                    # runner was live, benchmarks/cold_trace.py). Probe
+                    # channels opt into a private pool so a fresh channel
+                    # connects immediately; everything else shares subchannels
+                    # (100-way create bursts exhausted accept queues with
+                    # all-private pools).
+                    [("grpc.use_local_subchannel_pool", 1)]
+                    if self._private_pool
+                    else []
+                ),
             )
 
     async def close(self) -> None:
