@@ -20,7 +20,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
-#include <ATen/cuda/CUDAContext.h>
+#include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
 
 #define HIP_CHECK(expr)                                                        \
@@ -924,7 +924,7 @@ static torch::Tensor gemm_bf16_nt_impl(torch::Tensor A, torch::Tensor Bt,
     while (splitk < 16 && tiles * splitk < 384 && (K / BK) % (splitk * 2) == 0)
       splitk *= 2;
   }
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (splitk > 1) {
     auto Cpart = torch::empty({splitk, M, N}, A.options().dtype(torch::kFloat32));
     dim3 grid(tiles * splitk);
@@ -1008,7 +1008,7 @@ torch::Tensor gemm_bf16_nt_p8(torch::Tensor A, torch::Tensor Bt,
               "p8 kernel needs M,N %% 256 == 0 and K %% 64 == 0");
   auto C = torch::empty({M, N}, A.options());
   dim3 grid((M / BM2) * (N / BN2));
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (xcd_swizzle) {
     hipLaunchKernelGGL((gemm_bf16_nt_256p8_kernel<true>), grid, dim3(THREADS2),
                        0, stream, (const bf16_t*)A.data_ptr(),
@@ -1033,7 +1033,7 @@ torch::Tensor gemm_bf16_nt_p2(torch::Tensor A, torch::Tensor Bt,
               "p2 kernel needs M,N %% 256 == 0 and K %% 64 == 0");
   auto C = torch::empty({M, N}, A.options());
   dim3 grid((M / BM2) * (N / BN2));
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (xcd_swizzle) {
     hipLaunchKernelGGL((gemm_bf16_nt_256p2_kernel<true>), grid, dim3(THREADS2),
                        0, stream, (const bf16_t*)A.data_ptr(),
@@ -1340,7 +1340,7 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(M % BM == 0 && K % BK == 0 && (N % 64 == 0),
               "fp8 gate-up needs M%128, K%64, N%64 == 0");
   auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t tiles128 = (M / BM) * (N / 128);
   if (N % 128 == 0 && tiles128 >= 256) {
     dim3 grid(tiles128);
@@ -1371,7 +1371,7 @@ torch::Tensor gemm_gate_up_silu(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(Wu.size(0) == N && Wu.size(1) == K && Wg.size(1) == K);
   TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
   auto Out = torch::empty({M, N}, A.options());
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   // 256-row tile for large M (A staged once for both projections at
   // double the compute per staged byte); BN=128 when it fills the chip;
   // BN=64 (2 blocks/CU) otherwise (M=256 serving shapes)
@@ -1406,7 +1406,7 @@ torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
   int64_t n = gate.numel();
   int64_t blocks = std::min<int64_t>((n / 8 + 255) / 256, 4096);
   if (blocks < 1) blocks = 1;
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(silu_mul_kernel, dim3(blocks), dim3(256), 0, stream,
                      (const bf16_t*)gate.data_ptr(),
                      (const bf16_t*)up.data_ptr(), (bf16_t*)out.data_ptr(), n);
@@ -1420,7 +1420,7 @@ torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
   TORCH_CHECK(dim % 8 == 0, "dim must be multiple of 8");
   TORCH_CHECK(w.numel() == dim);
   auto out = torch::empty_like(x);
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(rmsnorm_kernel, dim3(rows), dim3(256), 0, stream,
                      (const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
                      (bf16_t*)out.data_ptr(), rows, dim, (float)eps);
