@@ -1194,9 +1194,12 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_gate_up_silu_256_kernel(
 // (guide: fp8 table) — the win is pure bandwidth. Per-output-channel
 // weight scales + one activation scale are applied in the fp32 epilogue
 // before silu*mul (standard w8a8 serving quantization).
-// LDS: fp8 rows are 64 B; 16-byte staging chunks are source-swizzled by
-// (row & 3) — residual 4-way read conflicts are acceptable because the op
-// is bandwidth-bound, not LDS-bound.
+// LDS: fp8 rows are 64 B = 16 dwords; staging uses WIDTH-4 glds with a
+// full 16-row XOR swizzle (dword position p holds global dword p^(row&15))
+// so the b32 fragment reads are conflict-free: bank(r, c4) =
+// (16r + (c4^r)) mod 64 is injective over the 16 rows of a quarter-wave
+// (16-byte-chunk staging left 4-way conflicts that made fp8 SLOWER than
+// bf16 — measured 68 vs 64 us before this layout).
 // ---------------------------------------------------------------------------
 
 typedef unsigned char fp8_t;
@@ -1221,9 +1224,10 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
   int lane = tid & 63;
   int wr = wave >> 1, wc = wave & 1;
 
-  // staging: rows are 64 B -> piece = 16 rows x 4 chunks of 16 B
-  int ld_row = lane >> 2;   // 0..15 within piece
-  int ld_chunk = lane & 3;  // 16B chunk in the 64B row
+  // staging: width-4 glds, piece = 4 rows x 16 dwords (256 B per instr);
+  // LDS dword position p holds global dword p ^ (row & 15)
+  int ld_row = lane >> 4;   // 0..3 within piece
+  int ld_dw = lane & 15;    // dword within the 64B row
   auto stage = [&](int buf, int kt) {
     fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
     fp8_t* Gb = Ab + BM * BK;
@@ -1232,23 +1236,23 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
     const fp8_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK;
     const fp8_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK;
 #pragma unroll
-    for (int call = 0; call < BM / 16 / WAVES; call++) {  // A pieces
+    for (int call = 0; call < BM / 4 / WAVES; call++) {  // A pieces
       int p = wave + WAVES * call;
-      int row = p * 16 + ld_row;
+      int row = p * 4 + ld_row;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 3)) * 16)),
-          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 1024), 16, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 256), 4, 0, 0);
     }
 #pragma unroll
-    for (int call = 0; call < BNT / 16 / WAVES; call++) {
+    for (int call = 0; call < BNT / 4 / WAVES; call++) {
       int p = wave + WAVES * call;
-      int row = p * 16 + ld_row;
+      int row = p * 4 + ld_row;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_chunk ^ (row & 3)) * 16)),
-          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 1024), 16, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
+          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 256), 4, 0, 0);
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_chunk ^ (row & 3)) * 16)),
-          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 1024), 16, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
+          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 256), 4, 0, 0);
     }
   };
 
@@ -1261,14 +1265,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
   int fr_row = lane & 15;
   int fr_q = lane >> 4;  // k-group (8 fp8) within the 32-wide k-step
 
-  // LDS read for (row r, 8B k-chunk c8): the 16B staging chunk holding c8
-  // was source-swizzled by (r & 3)
+  // LDS read for (row r, 8B k-chunk c8) = two swizzled b32 reads
   auto rd = [&](const fp8_t* img, int r, int c8) -> long {
-    int off = r * BK + (((c8 >> 1) ^ (r & 3)) << 4) + ((c8 & 1) << 3);
-    return *(const long*)(img + off);
+    const int base = r * BK;
+    unsigned lo = *(const unsigned*)(img + base + (((c8 * 2) ^ (r & 15)) << 2));
+    unsigned hi = *(const unsigned*)(img + base + (((c8 * 2 + 1) ^ (r & 15)) << 2));
+    return (long)(((unsigned long)hi << 32) | lo);
   };
 
-  constexpr int GLDS = BM / 16 / WAVES + 2 * (BNT / 16 / WAVES);
+  constexpr int GLDS = BM / 4 / WAVES + 2 * (BNT / 4 / WAVES);
   for (int kt = 0; kt < ntiles; kt++) {
     int buf = kt & 1;
     fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
