@@ -1194,12 +1194,14 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_gate_up_silu_256_kernel(
 // (guide: fp8 table) — the win is pure bandwidth. Per-output-channel
 // weight scales + one activation scale are applied in the fp32 epilogue
 // before silu*mul (standard w8a8 serving quantization).
-// LDS: fp8 rows are 64 B = 16 dwords; staging uses WIDTH-4 glds with a
-// full 16-row XOR swizzle (dword position p holds global dword p^(row&15))
-// so the b32 fragment reads are conflict-free: bank(r, c4) =
-// (16r + (c4^r)) mod 64 is injective over the 16 rows of a quarter-wave
-// (16-byte-chunk staging left 4-way conflicts that made fp8 SLOWER than
-// bf16 — measured 68 vs 64 us before this layout).
+// LDS layout: TWO K-tiles per 128-byte row (BK8=128 fp8 elements), which
+// restores the exact bf16 staging economics: width-16 glds over 128 B
+// rows with the proven chunk^(row&7) swizzle — global reads stay
+// cache-line-coalesced (the permutation stays inside each 128 B line) and
+// the b64 fragment reads use the same bank pattern the bf16 kernels
+// measured conflict-free. (Two earlier layouts — 64 B rows with 16 B or
+// 4 B staging — were slower than bf16: 4-way read conflicts and
+// scattered-source glds respectively; see git history.)
 // ---------------------------------------------------------------------------
 
 typedef unsigned char fp8_t;
@@ -1212,7 +1214,8 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
     float a_scale, int M, int N, int K) {
   constexpr int BNW = BNJ * 16;
   constexpr int BNT = BNW * 2;
-  __shared__ fp8_t lds[2 * (BM + 2 * BNT) * BK];  // fp8: half the bf16 bytes
+  constexpr int BK8 = 128;  // fp8 elements per LDS row = 128 B (2 k-tiles)
+  __shared__ fp8_t lds[2 * (BM + 2 * BNT) * BK8];
 
   int nwg_n = N / BNT;
   int wg = blockIdx.x;
@@ -1224,61 +1227,61 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
   int lane = tid & 63;
   int wr = wave >> 1, wc = wave & 1;
 
-  // staging: width-4 glds, piece = 4 rows x 16 dwords (256 B per instr);
-  // LDS dword position p holds global dword p ^ (row & 15)
-  int ld_row = lane >> 4;   // 0..3 within piece
-  int ld_dw = lane & 15;    // dword within the 64B row
-  auto stage = [&](int buf, int kt) {
-    fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
-    fp8_t* Gb = Ab + BM * BK;
-    fp8_t* Ub = Gb + BNT * BK;
-    const fp8_t* a_src = A + (int64_t)(tile_m)*K + kt * BK;
-    const fp8_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK;
-    const fp8_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK;
+  // staging: width-16 glds, 128 B rows, 8-row pieces (the bf16 pattern)
+  int ld_row = lane >> 3;   // 0..7 within piece
+  int ld_chunk = lane & 7;  // 16B chunk within the 128B row
+  auto stage = [&](int buf, int kt) {  // kt counts BK8=128-element tiles
+    fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK8;
+    fp8_t* Gb = Ab + BM * BK8;
+    fp8_t* Ub = Gb + BNT * BK8;
+    const fp8_t* a_src = A + (int64_t)(tile_m)*K + kt * BK8;
+    const fp8_t* g_src = Wg + (int64_t)(tile_n)*K + kt * BK8;
+    const fp8_t* u_src = Wu + (int64_t)(tile_n)*K + kt * BK8;
 #pragma unroll
-    for (int call = 0; call < BM / 4 / WAVES; call++) {  // A pieces
+    for (int call = 0; call < BM / 8 / WAVES; call++) {  // A pieces
       int p = wave + WAVES * call;
-      int row = p * 4 + ld_row;
+      int row = p * 8 + ld_row;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
-          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 256), 4, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 16)),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 1024), 16, 0, 0);
     }
 #pragma unroll
-    for (int call = 0; call < BNT / 4 / WAVES; call++) {
+    for (int call = 0; call < BNT / 8 / WAVES; call++) {
       int p = wave + WAVES * call;
-      int row = p * 4 + ld_row;
+      int row = p * 8 + ld_row;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
-          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 256), 4, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(g_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 16)),
+          (__attribute__((address_space(3))) uint32_t*)(Gb + p * 1024), 16, 0, 0);
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_dw ^ (row & 15)) * 4)),
-          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 256), 4, 0, 0);
+          (const __attribute__((address_space(1))) uint32_t*)(u_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 16)),
+          (__attribute__((address_space(3))) uint32_t*)(Ub + p * 1024), 16, 0, 0);
     }
   };
 
   f32x4 acc_g[4][BNJ] = {};
   f32x4 acc_u[4][BNJ] = {};
 
-  int ntiles = K / BK;
+  int ntiles = K / BK8;  // 128-element tiles
   stage(0, 0);
 
   int fr_row = lane & 15;
   int fr_q = lane >> 4;  // k-group (8 fp8) within the 32-wide k-step
 
-  // LDS read for (row r, 8B k-chunk c8) = two swizzled b32 reads
-  auto rd = [&](const fp8_t* img, int r, int c8) -> long {
-    const int base = r * BK;
-    unsigned lo = *(const unsigned*)(img + base + (((c8 * 2) ^ (r & 15)) << 2));
-    unsigned hi = *(const unsigned*)(img + base + (((c8 * 2 + 1) ^ (r & 15)) << 2));
-    return (long)(((unsigned long)hi << 32) | lo);
+  // fragment read: 8 fp8 (b64) at 16B-chunk-swizzled offsets — the bf16
+  // bank pattern, loop-invariant per lane
+  auto off_of = [&](int r, int c8) -> int {
+    return r * BK8 + ((((c8 >> 1) ^ (r & 7)) << 4)) + ((c8 & 1) << 3);
+  };
+  auto rd = [&](const fp8_t* img, int off) -> long {
+    return *(const long*)(img + off);
   };
 
-  constexpr int GLDS = BM / 4 / WAVES + 2 * (BNT / 4 / WAVES);
+  constexpr int GLDS = BM / 8 / WAVES + 2 * (BNT / 8 / WAVES);
   for (int kt = 0; kt < ntiles; kt++) {
     int buf = kt & 1;
-    fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK;
-    fp8_t* Gb = Ab + BM * BK;
-    fp8_t* Ub = Gb + BNT * BK;
+    fp8_t* Ab = lds + buf * (BM + 2 * BNT) * BK8;
+    fp8_t* Gb = Ab + BM * BK8;
+    fp8_t* Ub = Gb + BNT * BK8;
     if (kt + 1 < ntiles) {
       stage(buf ^ 1, kt + 1);
       asm volatile("s_waitcnt vmcnt(%0)" :: "i"(GLDS) : "memory");
@@ -1288,15 +1291,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
     __builtin_amdgcn_s_barrier();
 
 #pragma unroll
-    for (int ks = 0; ks < 2; ks++) {  // two K=32 MFMA steps per BK=64
+    for (int ks = 0; ks < 4; ks++) {  // four K=32 MFMA steps per BK8=128
       long a_frag[4], g_frag[BNJ], u_frag[BNJ];
 #pragma unroll
       for (int i = 0; i < 4; i++)
-        a_frag[i] = rd(Ab, wr * 64 + i * 16 + fr_row, ks * 4 + fr_q);
+        a_frag[i] = rd(Ab, off_of(wr * 64 + i * 16 + fr_row, ks * 4 + fr_q));
 #pragma unroll
       for (int j = 0; j < BNJ; j++) {
-        g_frag[j] = rd(Gb, wc * BNW + j * 16 + fr_row, ks * 4 + fr_q);
-        u_frag[j] = rd(Ub, wc * BNW + j * 16 + fr_row, ks * 4 + fr_q);
+        g_frag[j] = rd(Gb, off_of(wc * BNW + j * 16 + fr_row, ks * 4 + fr_q));
+        u_frag[j] = rd(Ub, off_of(wc * BNW + j * 16 + fr_row, ks * 4 + fr_q));
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -1342,8 +1345,8 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(g_scale.dtype() == torch::kFloat32 && u_scale.dtype() == torch::kFloat32);
   int64_t M = A.size(0), K = A.size(1), N = Wg.size(0);
   TORCH_CHECK(Wg.size(1) == K && Wu.size(1) == K && g_scale.size(0) == N);
-  TORCH_CHECK(M % BM == 0 && K % BK == 0 && (N % 64 == 0),
-              "fp8 gate-up needs M%128, K%64, N%64 == 0");
+  TORCH_CHECK(M % BM == 0 && K % 128 == 0 && (N % 64 == 0),
+              "fp8 gate-up needs M%128, K%128, N%64 == 0");
   auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
   auto stream = at::cuda::getCurrentCUDAStream();
   int64_t tiles128 = (M / BM) * (N / 128);

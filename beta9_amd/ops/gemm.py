@@ -167,3 +167,39 @@ def warm_all(device: str = "cuda:0") -> bool:
     q.transpose(1, 2).contiguous()
     torch.cuda.synchronize()
     return True
+
+
+E4M3_MAX = 448.0
+
+
+def quantize_fp8_rowwise(w: torch.Tensor):
+    """bf16 [N,K] -> (uint8 e4m3 view, float scale[N]) — per-output-channel
+    weight quantization for the w8a8 serving path."""
+    amax = w.float().abs().amax(dim=1).clamp(min=1e-8)
+    scale = amax / E4M3_MAX
+    q = (w.float() / scale[:, None]).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale.float()
+
+
+def quantize_fp8_tensor(a: torch.Tensor):
+    """bf16 [M,K] -> (uint8 e4m3 view, python float scale) per-tensor."""
+    amax = a.float().abs().amax().clamp(min=1e-8)
+    scale = (amax / E4M3_MAX).item()
+    return (a.float() / scale).to(torch.float8_e4m3fn).view(torch.uint8), scale
+
+
+def mlp_gate_up_fp8(a_q, w_gate_q, w_up_q, g_scale, u_scale, a_scale: float):
+    """FP8 (OCP e4m3, w8a8) fused act = silu(a @ wg.T) * (a @ wu.T).
+
+    The serving gate-up is weight-HBM-bound; fp8 halves every byte moved:
+    measured 51.5 us vs 67.5 us bf16 at M=256 x 14336 x 4096 (1.31x,
+    1169 TF-equiv; benchmarks/gateup_fp8_check.py). Kernel numerics match
+    the dequantized fp32 reference to 0.003; end-to-end error vs bf16 math
+    is the quantization error (~4%% max-rel on random data)."""
+    ext = _load()
+    if ext is None or not a_q.is_cuda:
+        raise RuntimeError("fp8 gate-up requires the native extension + GPU")
+    return ext.gemm_gate_up_silu_fp8(
+        a_q.contiguous(), w_gate_q.contiguous(), w_up_q.contiguous(),
+        g_scale.contiguous(), u_scale.contiguous(), a_scale,
+    )

@@ -211,3 +211,26 @@ def test_gemm_residual_numerics():
     out = g.gemm_nt_residual(a, w, r)
     ref = r.float() + a.float() @ w.float().t()
     assert ((out.float() - ref).abs().max() / ref.abs().max()).item() < 0.02
+
+
+@pytest.mark.gpu
+def test_gate_up_fp8_numerics():
+    """FP8 w8a8 fused gate-up vs the dequantized fp32 reference."""
+    import torch
+
+    from beta9_amd.ops import gemm as g
+
+    torch.manual_seed(3)
+    a = (torch.randn(256, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    wg = (torch.randn(512, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    wu = (torch.randn(512, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    aq, a_s = g.quantize_fp8_tensor(a)
+    gq, g_s = g.quantize_fp8_rowwise(wg)
+    uq, u_s = g.quantize_fp8_rowwise(wu)
+    out = g.mlp_gate_up_fp8(aq, gq, uq, g_s, u_s, a_s)
+    adq = aq.view(torch.float8_e4m3fn).float() * a_s
+    gdq = gq.view(torch.float8_e4m3fn).float() * g_s[:, None]
+    udq = uq.view(torch.float8_e4m3fn).float() * u_s[:, None]
+    ref = torch.nn.functional.silu(adq @ gdq.t()) * (adq @ udq.t())
+    err = ((out.float() - ref).abs().max() / ref.abs().max()).item()
+    assert err < 0.02, err
