@@ -142,6 +142,10 @@ def warm_all(device: str = "cuda:0") -> bool:
     ext.gemm_bf16_nt_residual(a, w, torch.randn(256, 256, dtype=torch.bfloat16, device=device), False, -1)
     ext.silu_mul(a, a)
     ext.rmsnorm(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6)
+    # fp8 serving kernel (both BNJ tiles use the same symbol per template)
+    q = torch.ones(128, 128, dtype=torch.uint8, device=device)
+    sc = torch.ones(128, dtype=torch.float32, device=device)
+    ext.gemm_gate_up_silu_fp8(q, q, q, sc, sc, 1.0)
     # torch EAGER kernels the serving path touches (residual add, checksum
     # cast/abs/reduce, softmax, cat): each costs 10-40 ms at first launch on
     # a cold box (measured 122 ms total, benchmarks/first_req_micro.py)
