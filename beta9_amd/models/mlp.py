@@ -42,13 +42,38 @@ class MLPBlock:
         self.w_up = make("w_up", (inter, hidden))
         self.w_down = make("w_down", (hidden, inter))
         self.norm_w = make("norm_w", (hidden,))
+        self._fp8 = None  # lazily-built w8a8 gate/up weights
+
+    def _fp8_weights(self):
+        if self._fp8 is None:
+            gq, gs = ops.quantize_fp8_rowwise(self.w_gate)
+            uq, us = ops.quantize_fp8_rowwise(self.w_up)
+            self._fp8 = (gq, gs, uq, us)
+        return self._fp8
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         import os
 
         h = ops.rmsnorm(x, self.norm_w)
-        # fused hand-written gate+up+silu (one A pass, no intermediates)
-        act = ops.mlp_gate_up(h, self.w_gate, self.w_up)
+        if os.environ.get("B9_GEMM_BACKEND", "native") == "fp8" and h.is_cuda:
+            # w8a8 serving path: per-channel fp8 weights (quantized once),
+            # per-call per-tensor activation quant; the gate-up is
+            # weight-HBM-bound so fp8 halves its floor (ops/gemm.py
+            # mlp_gate_up_fp8). Down-projection + residual stay bf16.
+            gq, gs, uq, us = self._fp8_weights()
+            hq, h_s = ops.quantize_fp8_tensor(h)
+            M = hq.shape[0]
+            if M % 128:
+                import torch.nn.functional as F
+
+                pad = (128 - M % 128) % 128
+                hq = F.pad(hq, (0, 0, 0, pad))
+                act = ops.mlp_gate_up_fp8(hq, gq, uq, gs, us, h_s)[:M]
+            else:
+                act = ops.mlp_gate_up_fp8(hq, gq, uq, gs, us, h_s)
+        else:
+            # fused hand-written gate+up+silu (one A pass, no intermediates)
+            act = ops.mlp_gate_up(h, self.w_gate, self.w_up)
         backend = os.environ.get("B9_GEMM_BACKEND", "native")
         if backend == "hybrid":
             # library GEMM + separate residual add (kept for A/B)

@@ -234,3 +234,30 @@ def test_gate_up_fp8_numerics():
     ref = torch.nn.functional.silu(adq @ gdq.t()) * (adq @ udq.t())
     err = ((out.float() - ref).abs().max() / ref.abs().max()).item()
     assert err < 0.02, err
+
+
+@pytest.mark.gpu
+def test_serving_model_fp8_backend():
+    """B9_GEMM_BACKEND=fp8 runs the w8a8 serving path end to end and stays
+    close to the bf16 output (quantization-level error)."""
+    import os
+
+    import torch
+
+    from beta9_amd.models.mlp import SyntheticServingModel
+
+    m = SyntheticServingModel(512, 1024, 2, "cuda:0", use_arena=False)
+    x = torch.randn(64, 512, dtype=torch.bfloat16, device="cuda:0")
+    old = os.environ.get("B9_GEMM_BACKEND")
+    try:
+        os.environ["B9_GEMM_BACKEND"] = "native"
+        ref = m.forward(x).float()
+        os.environ["B9_GEMM_BACKEND"] = "fp8"
+        out = m.forward(x).float()
+    finally:
+        if old is None:
+            os.environ.pop("B9_GEMM_BACKEND", None)
+        else:
+            os.environ["B9_GEMM_BACKEND"] = old
+    rel = ((out - ref).abs().mean() / ref.abs().mean()).item()
+    assert rel < 0.1, rel  # quantization-level deviation, not garbage
