@@ -146,6 +146,7 @@ def warm_all(device: str = "cuda:0") -> bool:
     q = torch.ones(128, 128, dtype=torch.uint8, device=device)
     sc = torch.ones(128, dtype=torch.float32, device=device)
     ext.gemm_gate_up_silu_fp8(q, q, q, sc, sc, 1.0)
+    ext.rmsnorm_fp8(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6, 1.0)
     # torch EAGER kernels the serving path touches (residual add, checksum
     # cast/abs/reduce, softmax, cat): each costs 10-40 ms at first launch on
     # a cold box (measured 122 ms total, benchmarks/first_req_micro.py)
@@ -207,3 +208,17 @@ def mlp_gate_up_fp8(a_q, w_gate_q, w_up_q, g_scale, u_scale, a_scale: float):
         a_q.contiguous(), w_gate_q.contiguous(), w_up_q.contiguous(),
         g_scale.contiguous(), u_scale.contiguous(), a_scale,
     )
+
+
+def rmsnorm_fp8_out(x: torch.Tensor, w: torch.Tensor, act_scale: float,
+                    eps: float = 1e-6) -> torch.Tensor:
+    """rmsnorm with the w8a8 activation quantization fused into its epilogue:
+    emits uint8 e4m3 of (rmsnorm(x)*w / act_scale) directly — no separate
+    amax/div/cast kernels (those made the fp8 serving path SLOWER than
+    bf16 end-to-end despite the 1.31x gate-up kernel win)."""
+    ext = _load()
+    if ext is None or not x.is_cuda:
+        raise RuntimeError("rmsnorm_fp8_out requires the native extension + GPU")
+    shape = x.shape
+    return ext.rmsnorm_fp8(x.reshape(-1, shape[-1]).contiguous(), w.contiguous(),
+                           eps, act_scale).reshape(shape)

@@ -261,3 +261,25 @@ def test_serving_model_fp8_backend():
             os.environ["B9_GEMM_BACKEND"] = old
     rel = ((out - ref).abs().mean() / ref.abs().mean()).item()
     assert rel < 0.1, rel  # quantization-level deviation, not garbage
+
+
+@pytest.mark.gpu
+def test_rmsnorm_fp8_fused_quant():
+    """rmsnorm with fused e4m3 output vs separate rmsnorm + quantize."""
+    import torch
+
+    from beta9_amd.ops import gemm as g
+
+    torch.manual_seed(5)
+    x = (torch.randn(64, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    w = torch.ones(512, dtype=torch.bfloat16, device="cuda")
+    h = g.rmsnorm(x, w)
+    scale = (h.float().abs().amax() / g.E4M3_MAX).item()
+    ref = (h.float() / scale).to(torch.float8_e4m3fn).float()
+    out = g.rmsnorm_fp8_out(x, w, scale).view(torch.float8_e4m3fn).float()
+    # small elementwise deviation allowed (fp32 vs bf16 intermediate paths)
+    err = (out - ref).abs().max().item()
+    assert err <= 1.0, err  # within one e4m3 quantization step
+    # dequantized output close to bf16 rmsnorm
+    rel = ((out * scale - h.float()).abs().mean() / h.float().abs().mean()).item()
+    assert rel < 0.05, rel
