@@ -32,7 +32,8 @@ CREATE TABLE IF NOT EXISTS tokens (
   key TEXT PRIMARY KEY,
   workspace_id TEXT NOT NULL,
   active INTEGER DEFAULT 1,
-  created_at REAL NOT NULL
+  created_at REAL NOT NULL,
+  token_type TEXT DEFAULT 'workspace'
 );
 CREATE TABLE IF NOT EXISTS apps (
   id TEXT PRIMARY KEY,
@@ -133,6 +134,12 @@ class BackendDB:
             self._conn.execute("PRAGMA journal_mode=WAL")
             self._conn.execute("PRAGMA synchronous=NORMAL")
             self._conn.executescript(_SCHEMA)
+            try:  # migrate pre-token_type DB files in place
+                self._conn.execute(
+                    "ALTER TABLE tokens ADD COLUMN token_type TEXT DEFAULT 'workspace'"
+                )
+            except sqlite3.OperationalError:
+                pass
             self._conn.commit()
         # write-behind queue for hot-path task rows (authoritative task
         # state lives in the state store; rows are the durable audit log)
@@ -231,20 +238,27 @@ class BackendDB:
         )
         return rows[0]["key"]
 
-    def create_token(self, workspace_id: str) -> str:
+    def create_token(self, workspace_id: str, token_type: str = "workspace") -> str:
+        """token_type 'workspace' (tenant-scoped) or 'cluster_admin'
+        (reference: pkg/auth token types gate cross-workspace ops,
+        pkg/auth/http.go:106)."""
         key = uuid.uuid4().hex + uuid.uuid4().hex[:8]
         self._exec(
-            "INSERT INTO tokens (key, workspace_id, created_at) VALUES (?,?,?)",
-            (key, workspace_id, time.time()),
+            "INSERT INTO tokens (key, workspace_id, created_at, token_type) VALUES (?,?,?,?)",
+            (key, workspace_id, time.time(), token_type),
         )
         return key
 
     def authorize_token(self, key: str) -> Optional[Dict[str, Any]]:
         rows = self._query(
-            "SELECT t.key, t.workspace_id, w.name as workspace_name FROM tokens t "
+            "SELECT t.key, t.workspace_id, t.token_type, w.name as workspace_name FROM tokens t "
             "JOIN workspaces w ON w.id = t.workspace_id WHERE t.key=? AND t.active=1",
             (key,),
         )
+        return dict(rows[0]) if rows else None
+
+    def get_token(self, key: str) -> Optional[Dict[str, Any]]:
+        rows = self._query("SELECT * FROM tokens WHERE key=?", (key,))
         return dict(rows[0]) if rows else None
 
     def revoke_token(self, key: str) -> None:

@@ -61,14 +61,21 @@ class Gateway:
         ctx = self.db.authorize_token(token)
         if ctx is None:
             return None
-        return {"workspace_id": ctx["workspace_id"], "workspace_name": ctx["workspace_name"], "token": token}
+        return {
+            "workspace_id": ctx["workspace_id"],
+            "workspace_name": ctx["workspace_name"],
+            "token": token,
+            "token_type": ctx.get("token_type", "workspace"),
+        }
 
     def bootstrap_workspace(self, name: str = "default") -> Dict[str, str]:
-        """Create (or fetch) a workspace + token — first-boot convenience."""
+        """Create (or fetch) a workspace + cluster-admin token — first-boot
+        convenience (reference: initial ClusterAdmin token,
+        pkg/auth/http.go:106)."""
         ws = self.db.get_workspace_by_name(name)
         if ws is None:
             ws = self.db.create_workspace(name)
-        token = self.db.create_token(ws["id"])
+        token = self.db.create_token(ws["id"], token_type="cluster_admin")
         return {"workspace_id": ws["id"], "workspace_name": name, "token": token}
 
     def runner_token(self, workspace_id: str) -> str:

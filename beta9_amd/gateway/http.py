@@ -45,12 +45,14 @@ def build_http_app(gateway) -> FastAPI:
         if cfg is not None and getattr(cfg, "authorized", True) is False:
             if (cfg.extra or {}).get("require_signature"):
                 sig = request.headers.get("x-b9-signature", "")
-                if not await _verify_signature(stub["workspace_id"], sig, request):
+                if not await _verify_signature(stub["workspace_id"], sig, request, stub_id):
                     raise HTTPException(status_code=401, detail="bad signature")
             return {"workspace_id": stub["workspace_id"], "workspace_name": "public"}
         return None
 
-    async def _verify_signature(workspace_id: str, sig: str, request: Request, max_age_s: float = 300.0) -> bool:
+    async def _verify_signature(
+        workspace_id: str, sig: str, request: Request, stub_id: str, max_age_s: float = 300.0
+    ) -> bool:
         import hashlib
         import hmac as _hmac
         import time as _time
@@ -59,13 +61,18 @@ def build_http_app(gateway) -> FastAPI:
             return False
         try:
             ts, mac = sig[3:].split(":", 1)
+            ts_val = float(ts)
         except ValueError:
-            return False
-        if abs(_time.time() - float(ts)) > max_age_s:
+            return False  # malformed timestamp is a 401, never a 500
+        if abs(_time.time() - ts_val) > max_age_s:
             return False
         key = gateway.db.get_or_create_signing_key(workspace_id).encode()
         body = await request.body()  # cached by starlette for the route
-        want = _hmac.new(key, ts.encode() + b"." + body, hashlib.sha256).hexdigest()
+        msg = (
+            ts.encode() + b"." + request.method.upper().encode() + b"."
+            + stub_id.encode() + b"." + body
+        )
+        want = _hmac.new(key, msg, hashlib.sha256).hexdigest()
         return _hmac.compare_digest(want, mac)
 
     @app.get("/api/v1/health")
@@ -181,17 +188,36 @@ def build_http_app(gateway) -> FastAPI:
                 out.append(d)
         return out
 
+    async def _own_container_or_404(container_id: str, auth: dict) -> None:
+        # workspace-scoped access: 404 on other tenants' containers
+        # (reference: workspace auth on container endpoints)
+        if auth.get("token_type") == "cluster_admin":
+            return
+        st = await gateway.container_repo.get_state(container_id)
+        if st is not None:
+            if st.get("workspace_id") != auth["workspace_id"]:
+                raise HTTPException(status_code=404)
+            return
+        rows = gateway.db._query(
+            "SELECT workspace_id FROM tasks WHERE container_id=? LIMIT 1", (container_id,)
+        )
+        if not rows or rows[0]["workspace_id"] != auth["workspace_id"]:
+            raise HTTPException(status_code=404)
+
     @app.post("/api/v1/containers/{container_id}/stop")
     async def stop_container(container_id: str, auth: dict = Depends(auth_ctx)):
+        await _own_container_or_404(container_id, auth)
         await gateway.scheduler.stop(container_id)
         return {"ok": True}
 
     @app.get("/api/v1/logs/{container_id}")
     async def container_logs(container_id: str, tail: int = 200, auth: dict = Depends(auth_ctx)):
+        await _own_container_or_404(container_id, auth)
         return await gateway.container_repo.get_logs(container_id, tail=tail)
 
     @app.get("/api/v1/spans/{container_id}")
     async def container_spans(container_id: str, auth: dict = Depends(auth_ctx)):
+        await _own_container_or_404(container_id, auth)
         return await gateway.container_repo.get_spans(container_id)
 
     @app.get("/api/v1/tasks")
@@ -200,12 +226,15 @@ def build_http_app(gateway) -> FastAPI:
 
     @app.get("/api/v1/tasks/{task_id}")
     async def get_task(task_id: str, auth: dict = Depends(auth_ctx)):
+        admin = auth.get("token_type") == "cluster_admin"
         t = await gateway.dispatcher.get_task(task_id)
         if t is None:
             row = gateway.db.get_task(task_id)
-            if row is None:
+            if row is None or (not admin and row.get("workspace_id") != auth["workspace_id"]):
                 raise HTTPException(status_code=404)
             return row
+        if not admin and t.workspace_id != auth["workspace_id"]:
+            raise HTTPException(status_code=404)
         return t.model_dump()
 
     @app.get("/api/v1/workers")
@@ -250,8 +279,10 @@ def build_http_app(gateway) -> FastAPI:
 
     @app.get("/api/v1/outputs/{rel:path}")
     async def get_output(rel: str, auth: dict = Depends(auth_ctx)):
-        path = os.path.realpath(os.path.join(gateway.cfg.storage.root, rel))
-        if not path.startswith(os.path.realpath(gateway.cfg.storage.root)) or not os.path.isfile(path):
+        root = os.path.realpath(gateway.cfg.storage.root)
+        path = os.path.realpath(os.path.join(root, rel))
+        # commonpath (not startswith) so a sibling like <root>2 is unreachable
+        if os.path.commonpath([root, path]) != root or not os.path.isfile(path):
             raise HTTPException(status_code=404)
         with open(path, "rb") as f:
             return Response(content=f.read(), media_type="application/octet-stream")

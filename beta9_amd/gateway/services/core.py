@@ -19,7 +19,11 @@ from ...backend.db import BackendDB
 from ...backend.repos import ContainerRepo, WorkerRepo
 from ...common import keys
 from ...common.state import StateStore
-from ...rpc.channel import RpcError, invalid, not_found, rpc
+from ...rpc.channel import RpcError, invalid, not_found, permission_denied, rpc
+
+
+def _is_admin(auth: Dict[str, Any]) -> bool:
+    return auth.get("token_type") == "cluster_admin"
 from ...task.dispatch import Dispatcher
 from ...types import StubConfig, StubType, TaskStatus
 
@@ -60,31 +64,47 @@ class GatewayService:
         """HMAC-sign a payload with the workspace signing key so public
         (authorized=False) endpoints can verify callers (reference:
         auth SignPayload RPC). Signature format: v1=<unix_ts>:<hex hmac of
-        ts.payload>."""
+        ts.method.stub_id.payload> — method + stub bound into the MAC so a
+        signature is not replayable against other stubs/methods in the
+        workspace within the timestamp window."""
         import hashlib
         import hmac as _hmac
 
         payload = req.get("payload", b"")
         if isinstance(payload, str):
             payload = payload.encode()
+        method = str(req.get("method", "POST")).upper()
+        stub_id = str(req.get("stub_id", ""))
         key = self.db.get_or_create_signing_key(auth["workspace_id"]).encode()
         ts = str(int(time.time()))
-        sig = _hmac.new(key, ts.encode() + b"." + payload, hashlib.sha256).hexdigest()
+        msg = ts.encode() + b"." + method.encode() + b"." + stub_id.encode() + b"." + payload
+        sig = _hmac.new(key, msg, hashlib.sha256).hexdigest()
         return {"signature": f"v1={ts}:{sig}", "expires_in_s": 300}
 
     # -- workspaces / tokens (ops surface) -----------------------------------
     @rpc()
     async def CreateWorkspace(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Admin-only: creating tenants is a cluster operation (reference
+        gates cross-workspace ops behind ClusterAdmin, pkg/auth/http.go:106)."""
+        if not _is_admin(auth):
+            raise permission_denied("cluster_admin token required to create workspaces")
         ws = self.db.get_workspace_by_name(req["name"]) or self.db.create_workspace(req["name"])
         return {"workspace_id": ws["id"], "name": req["name"]}
 
     @rpc()
     async def CreateToken(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
         ws_id = req.get("workspace_id") or auth["workspace_id"]
+        if ws_id != auth["workspace_id"] and not _is_admin(auth):
+            raise permission_denied("cannot mint tokens for another workspace")
         return {"token": self.db.create_token(ws_id), "workspace_id": ws_id}
 
     @rpc()
     async def RevokeToken(self, req: Dict[str, Any], context, auth) -> bool:
+        row = self.db.get_token(req["token"])
+        if row is None:
+            return True  # already gone / never existed — idempotent
+        if row["workspace_id"] != auth["workspace_id"] and not _is_admin(auth):
+            raise permission_denied("cannot revoke another workspace's token")
         self.db.revoke_token(req["token"])
         return True
 
@@ -228,17 +248,49 @@ class GatewayService:
                 out.append(d)
         return out
 
+    async def _check_container_access(self, container_id: str, auth) -> None:
+        """404 unless the container belongs to the caller's workspace (or
+        admin) — cross-tenant kill/log-disclosure guard (reference:
+        workspace-scoped auth on container RPCs)."""
+        if _is_admin(auth):
+            return
+        st = await self.containers.get_state(container_id)
+        if st is None:
+            # exited containers: fall back to the task audit rows
+            row = self.db._query(
+                "SELECT workspace_id FROM tasks WHERE container_id=? LIMIT 1", (container_id,)
+            )
+            if row and row[0]["workspace_id"] == auth["workspace_id"]:
+                return
+            raise not_found(f"container {container_id}")
+        if st.get("workspace_id") != auth["workspace_id"]:
+            raise not_found(f"container {container_id}")
+
+    async def _check_task_access(self, task_id: str, auth) -> None:
+        if _is_admin(auth):
+            return
+        t = await self.dispatcher.get_task(task_id)
+        ws = t.workspace_id if t is not None else None
+        if ws is None:
+            row = self.db.get_task(task_id)
+            ws = row.get("workspace_id") if row else None
+        if ws != auth["workspace_id"]:
+            raise not_found(f"task {task_id}")
+
     @rpc()
     async def StopContainer(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._check_container_access(req["container_id"], auth)
         await self.scheduler.stop(req["container_id"])
         return True
 
     @rpc()
     async def GetContainerLogs(self, req: Dict[str, Any], context, auth) -> list:
+        await self._check_container_access(req["container_id"], auth)
         return await self.containers.get_logs(req["container_id"], tail=int(req.get("tail", 200)))
 
     @rpc()
     async def GetContainerSpans(self, req: Dict[str, Any], context, auth) -> list:
+        await self._check_container_access(req["container_id"], auth)
         return await self.containers.get_spans(req["container_id"])
 
     @rpc(kind="unary_stream")
@@ -248,6 +300,7 @@ class GatewayService:
         import asyncio as _asyncio
 
         cid = req["container_id"]
+        await self._check_container_access(cid, auth)
         sent = 0
         idle = 0.0
         while True:
@@ -268,6 +321,7 @@ class GatewayService:
     # -- tasks ----------------------------------------------------------------
     @rpc()
     async def GetTask(self, req: Dict[str, Any], context, auth) -> Optional[Dict]:
+        await self._check_task_access(req["task_id"], auth)
         t = await self.dispatcher.get_task(req["task_id"])
         return t.model_dump() if t else None
 
@@ -277,11 +331,13 @@ class GatewayService:
 
     @rpc()
     async def StopTask(self, req: Dict[str, Any], context, auth) -> bool:
+        await self._check_task_access(req["task_id"], auth)
         await self.dispatcher.cancel(req["task_id"])
         return True
 
     @rpc()
     async def GetTaskResult(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        await self._check_task_access(req["task_id"], auth)
         t = await self.dispatcher.get_task(req["task_id"])
         if t is None:
             raise not_found(f"task {req['task_id']}")

@@ -1393,6 +1393,10 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   TORCH_CHECK(g_scale.dtype() == torch::kFloat32 && u_scale.dtype() == torch::kFloat32);
   int64_t M = A.size(0), K = A.size(1), N = Wg.size(0);
   TORCH_CHECK(Wg.size(1) == K && Wu.size(1) == K && g_scale.size(0) == N);
+  TORCH_CHECK(u_scale.size(0) == N, "u_scale length must equal N");
+  TORCH_CHECK(A.is_contiguous() && Wg.is_contiguous() && Wu.is_contiguous() &&
+              g_scale.is_contiguous() && u_scale.is_contiguous());
+  TORCH_CHECK(Wg.is_cuda() && Wu.is_cuda() && g_scale.is_cuda() && u_scale.is_cuda());
   TORCH_CHECK(M % BM == 0 && K % 128 == 0 && (N % 64 == 0),
               "fp8 gate-up needs M%128, K%128, N%64 == 0");
   auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));

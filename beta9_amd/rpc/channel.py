@@ -62,6 +62,10 @@ def unauthenticated(msg: str = "invalid token") -> RpcError:
     return RpcError(grpc.StatusCode.UNAUTHENTICATED, msg)
 
 
+def permission_denied(msg: str = "permission denied") -> RpcError:
+    return RpcError(grpc.StatusCode.PERMISSION_DENIED, msg)
+
+
 Handler = Callable[..., Any]
 AuthFn = Callable[[str, str], Awaitable[Optional[Dict[str, Any]]]]
 # AuthFn(token, full_method) -> auth context dict or None (reject)

@@ -124,7 +124,8 @@ async def test_signed_public_payloads():
         body = _json.dumps({"x": 7}).encode()
         client = beta9_amd.Client()
         signed = await asyncio.to_thread(
-            client.call, "GatewayService", "SignPayload", {"payload": body}
+            client.call, "GatewayService", "SignPayload",
+            {"payload": body, "method": "POST", "stub_id": stub_id},
         )
         base = f"http://127.0.0.1:{c.gateway.http_port}"
         async with httpx.AsyncClient() as hc:
