@@ -39,10 +39,14 @@ class GatewayConfig(BaseModel):
 
 
 class ImageServiceConfig(BaseModel):
-    # Content-addressed image store root (CLIP-equivalent, chunked).
+    # Content-addressed image store root (OCI layers + prefix images).
     store_path: str = "/var/lib/b9/images"
     chunk_size: int = 4 * 1024 * 1024
     build_container_ttl_s: float = 60.0
+    # build containers (steps NEVER run on the gateway host)
+    build_cpu: int = 4000
+    build_memory_mb: int = 8192
+    build_timeout_s: float = 900.0
 
 
 class StorageConfig(BaseModel):

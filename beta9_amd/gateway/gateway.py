@@ -293,9 +293,10 @@ class Gateway:
             VolumeService(self.db, self.cfg.storage.root),
             OutputService(self.db, self.cfg.storage.root),
             PodService(self),
-            ImageService(self),
             BotService(self),
         ]
+        self.image_service = ImageService(self)
+        self._services.append(self.image_service)
         for svc in self._services:
             self.rpc_server.register_service(svc)
 

@@ -53,6 +53,9 @@ class StubType(str, Enum):
     pod_deployment = "pod/deployment"
     pod_run = "pod/run"
     sandbox = "sandbox"
+    image_build = "image/build"
+    bot = "bot"
+    bot_deployment = "bot/deployment"
 
     @property
     def kind(self) -> str:
@@ -226,6 +229,21 @@ class StubConfig(BaseModel):
     pricing: Optional[PricingPolicy] = None
 
 
+class BuildSpec(BaseModel):
+    """Build-container payload. Build steps execute INSIDE a scheduled
+    container — never on the gateway host (reference:
+    pkg/abstractions/image/build.go:62-343 build containers into the build
+    pool). mode "prefix" = host-base image (site/files dirs bound at
+    /b9img); mode "rootfs" = layered OCI-style build (base image layers as
+    overlayfs lowerdirs, host-backed upper captured as the new layer)."""
+
+    image_id: str = ""
+    mode: str = "prefix"  # prefix | rootfs
+    base_image_id: str = ""
+    build_dir: str = ""  # host dir receiving prefix content / upper delta
+    steps: List[Dict[str, Any]] = Field(default_factory=list)
+
+
 class ContainerRequest(BaseModel):
     """Scheduler unit of work (reference: pkg/types/scheduler.go:255-298)."""
 
@@ -256,6 +274,7 @@ class ContainerRequest(BaseModel):
     delivery_token: str = ""
     hostname: str = ""
     rendezvous: Optional[RendezvousSpec] = None
+    build: Optional[BuildSpec] = None
 
     def requires_gpu(self) -> bool:
         return bool(self.gpu) or bool(self.gpu_request) or self.gpu_count > 0

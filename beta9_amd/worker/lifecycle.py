@@ -85,6 +85,7 @@ class Lifecycle:
         self.worker_host = worker_host
         self.prewarm = prewarm
         self.image_store_path = image_store_path
+        self.images = None  # ImageStore, lazy (store may not exist yet)
         self.containers: Dict[str, ContainerInstance] = {}
         from .usage import UsageTracker
 
@@ -103,7 +104,13 @@ class Lifecycle:
         self._container_runtime: Dict[str, object] = {}
 
     def _runtime_for(self, spec) -> "Runtime":
-        if self.native_runtime is not None and (spec.mounts or spec.block_network):
+        needs_native = (
+            spec.mounts
+            or spec.block_network
+            or spec.image_lowerdirs
+            or spec.upper_dir
+        )
+        if self.native_runtime is not None and needs_native:
             self._container_runtime[spec.container_id] = self.native_runtime
             return self.native_runtime
         self._container_runtime[spec.container_id] = self.runtime
@@ -234,7 +241,89 @@ class Lifecycle:
             inst.done.set()
             self.containers.pop(cid, None)
 
+    def _image_store(self):
+        if self.images is None and self.image_store_path:
+            from ..images.store import ImageStore
+
+            self.images = ImageStore(self.image_store_path)
+        return self.images
+
+    def _build_spec(self, request: ContainerRequest, inst: ContainerInstance):
+        """Spec for an image-build container (reference: build containers,
+        pkg/abstractions/image/build.go:89-343). Steps run inside the
+        container runtime; the product leaves via bind mounts (prefix) or
+        the host-backed overlay upper (rootfs)."""
+        import json as _json
+
+        b = request.build
+        env = dict(request.env)
+        env.setdefault("CONTAINER_ID", request.container_id)
+        env.setdefault("STUB_ID", request.stub_id)
+        env["B9_BUILD_STEPS"] = _json.dumps(b.steps)
+        env["B9_BUILD_MODE"] = b.mode
+        mounts = []
+        lowerdirs: list = []
+        upper = work = ""
+        native_ok = self.native_runtime is not None
+        if b.mode == "prefix":
+            site_host = os.path.join(b.build_dir, "site")
+            files_host = os.path.join(b.build_dir, "files")
+            os.makedirs(site_host, exist_ok=True)
+            os.makedirs(files_host, exist_ok=True)
+            if native_ok:
+                mounts = [
+                    {"local_path": site_host, "mount_path": "/b9img/site"},
+                    {"local_path": files_host, "mount_path": "/b9img/files"},
+                ]
+                env["IMAGE_SITE"] = "/b9img/site"
+                env["PREFIX"] = "/b9img/files"
+            else:
+                # no namespace privileges on this host: steps still run in a
+                # worker-side container process, product written directly
+                env["IMAGE_SITE"] = site_host
+                env["PREFIX"] = files_host
+            entry = [sys.executable, "-u", "-m", "beta9_amd.runner.build"]
+        else:  # rootfs
+            if not native_ok:
+                raise RuntimeError("rootfs builds need namespace privileges (b9launch)")
+            store = self._image_store()
+            lowerdirs = store.lowerdirs(b.base_image_id)
+            upper = os.path.join(b.build_dir, "upper")
+            work = os.path.join(b.build_dir, "work")
+            os.makedirs(upper, exist_ok=True)
+            os.makedirs(work, exist_ok=True)
+            repo_root = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+            mounts = [{"local_path": repo_root, "mount_path": "/opt/b9", "read_only": True}]
+            env["PYTHONPATH"] = "/opt/b9"
+            cfg_env = (store.get_manifest(b.base_image_id) or {}).get("config", {}).get("env") or []
+            for kv in cfg_env:
+                k, _, v = kv.partition("=")
+                env.setdefault(k, v)
+            env.setdefault("PATH", "/usr/local/sbin:/usr/local/bin:/usr/sbin:/usr/bin:/sbin:/bin")
+            if b.steps and all(s.get("kind") == "shell" for s in b.steps):
+                # shell-only builds run with the image's own /bin/sh — no
+                # python required inside minimal base images
+                script = " && ".join(s.get("cmd", "true") for s in b.steps)
+                entry = ["/bin/sh", "-ec", script]
+            else:
+                entry = ["python3", "-u", "-m", "beta9_amd.runner.build"]
+        spec = ContainerSpec(
+            container_id=request.container_id,
+            entry_point=entry,
+            env=env,
+            cpu_millicores=request.cpu,
+            memory_mb=request.memory_mb,
+            hostname=request.container_id,
+            mounts=mounts,
+            image_lowerdirs=lowerdirs,
+            upper_dir=upper,
+            work_dir=work,
+        )
+        return spec, False
+
     def _spec_from_request(self, request: ContainerRequest, inst: ContainerInstance):
+        if request.build is not None:
+            return self._build_spec(request, inst)
         kind = StubType(request.stub_type).kind if request.stub_type else "function"
         needs_addr = kind in _HTTP_KINDS
         env = dict(request.env)
@@ -248,27 +337,59 @@ class Lifecycle:
         env.setdefault("B9_LIGHT_IMPORT", "1")  # runners skip the SDK surface
         if request.task_id:
             env.setdefault("TASK_ID", request.task_id)
+        lowerdirs: list = []
+        extra_mounts: list = []
+        rootfs_manifest = None
         if request.image_id and self.image_store_path:
-            # image prefix injection: site dir onto PYTHONPATH + files tree
-            from ..abstractions.image import image_env
+            store = self._image_store()
+            m = store.get_manifest(request.image_id) if store else None
+            if m is not None and m.get("kind") == "rootfs" and self.native_runtime is not None:
+                # OCI-layer rootfs: overlay lowerdir chain; our runner rides
+                # in via a read-only bind (reference mounts the SDK runner
+                # into containers the same way)
+                rootfs_manifest = m
+                lowerdirs = store.lowerdirs(request.image_id)
+                repo_root = os.path.dirname(
+                    os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+                )
+                extra_mounts.append(
+                    {"local_path": repo_root, "mount_path": "/opt/b9", "read_only": True}
+                )
+                env.setdefault("PYTHONPATH", "/opt/b9")
+                for kv in (m.get("config", {}).get("env") or []):
+                    k, _, v = kv.partition("=")
+                    env.setdefault(k, v)
+                env.setdefault(
+                    "PATH", "/usr/local/sbin:/usr/local/bin:/usr/sbin:/usr/bin:/sbin:/bin"
+                )
+            else:
+                # prefix image: site dir onto PYTHONPATH + files tree
+                from ..abstractions.image import image_env
 
-            env.update(image_env(self.image_store_path, request.image_id))
+                env.update(image_env(self.image_store_path, request.image_id))
         if needs_addr:
             inst.port = free_port()
             env["BIND_PORT"] = str(inst.port)
         entry = list(request.entry_point)
+        if not entry and rootfs_manifest is not None:
+            entry = list(rootfs_manifest.get("config", {}).get("entrypoint") or [])
         if not entry:
             module = _RUNNER_MODULES.get(kind, "beta9_amd.runner.function")
-            entry = [sys.executable, "-u", "-m", module]
+            if rootfs_manifest is not None:
+                entry = ["python3", "-u", "-m", module]
+            else:
+                entry = [sys.executable, "-u", "-m", module]
         spec = ContainerSpec(
             container_id=request.container_id,
             entry_point=entry,
             env=env,
+            cwd=(rootfs_manifest or {}).get("config", {}).get("cwd") or "",
             cpu_millicores=request.cpu,
             memory_mb=request.memory_mb,
             hostname=request.hostname or request.container_id,
             block_network=request.block_network,
-            mounts=[m.model_dump() for m in request.mounts],
+            mounts=[m.model_dump() for m in request.mounts] + extra_mounts,
+            image_lowerdirs=lowerdirs,
         )
         return spec, needs_addr
 

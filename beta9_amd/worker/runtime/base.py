@@ -46,6 +46,12 @@ class ContainerSpec:
     mounts: List[Dict] = field(default_factory=list)
     hostname: str = ""
     block_network: bool = False
+    # OCI-layer rootfs: overlayfs lowerdir chain (topmost first) from the
+    # content-addressed image store; empty = host-rootfs overlay mode.
+    image_lowerdirs: List[str] = field(default_factory=list)
+    # host-backed upper/work (build capture: the fs delta survives exit)
+    upper_dir: str = ""
+    work_dir: str = ""
 
 
 @dataclass

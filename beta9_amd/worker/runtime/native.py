@@ -75,6 +75,12 @@ class NativeRuntime(ProcRuntime):
             "--cgroup-parent",
             "b9",
         ]
+        if spec.image_lowerdirs:
+            # rootfs image: pure layer chain (host "/" never mixed in —
+            # overlayfs rejects nested lowerdirs)
+            argv += ["--lower", ":".join(spec.image_lowerdirs)]
+        if spec.upper_dir and spec.work_dir:
+            argv += ["--upper", spec.upper_dir, "--work", spec.work_dir]
         if spec.cpu_millicores > 0:
             argv += ["--cpu", str(spec.cpu_millicores)]
         if spec.memory_mb > 0:

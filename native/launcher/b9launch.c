@@ -17,10 +17,17 @@
  *   b9launch [--rootfs-overlay DIR] [--cpu MILLICORES] [--mem MB]
  *            [--hostname NAME] [--dev /dev/node]... [--bind SRC:DST]...
  *            [--env K=V]... [--cwd DIR] [--block-network]
+ *            [--lower DIR1:DIR2:...] [--upper DIR --work DIR]
  *            [--cgroup-parent NAME] -- CMD [ARGS...]
  *
  * --rootfs-overlay DIR: DIR must contain (or will get) upper/ work/ merged/;
  * lowerdir is "/" unless DIR/lower exists (a prepared image tree).
+ * --lower: explicit overlayfs lowerdir chain (topmost first) — image layers
+ *   from the content-addressed layer store.
+ * --upper/--work: host-disk-backed upper/work instead of the tmpfs scratch,
+ *   so a build container's filesystem delta SURVIVES exit and can be
+ *   captured as a new image layer (reference: build containers +
+ *   captureCheckpointFilesystem upper-layer semantics).
  */
 
 #define _GNU_SOURCE
@@ -55,6 +62,9 @@ static int g_nenvs = 0;
 static const char *g_cwd = NULL;
 static int g_block_network = 0;
 static const char *g_cgroup_parent = "b9";
+static const char *g_lower = NULL;   /* explicit lowerdir chain (topmost first) */
+static const char *g_upper = NULL;   /* host-backed upper (build capture) */
+static const char *g_work = NULL;    /* host-backed workdir */
 static char **g_cmd = NULL;
 
 static void die(const char *msg) {
@@ -130,20 +140,32 @@ static int child_main(void *arg) {
     die("make / rprivate");
 
   if (g_overlay_dir) {
-    char upper[768], work[768], merged[768], lower[768], opts[3072];
-    snprintf(upper, sizeof upper, "%s/upper", g_overlay_dir);
-    snprintf(work, sizeof work, "%s/work", g_overlay_dir);
-    snprintf(merged, sizeof merged, "%s/merged", g_overlay_dir);
-    snprintf(lower, sizeof lower, "%s/lower", g_overlay_dir);
-    /* upper/work must be on tmpfs so nesting on any host fs works */
+    char upper[768], work[768], merged[768], lower[768], opts[4000];
     mkdir(g_overlay_dir, 0755);
-    if (mount("tmpfs", g_overlay_dir, "tmpfs", 0, "size=90%") != 0)
-      die("tmpfs for overlay scratch");
-    mkdir(upper, 0755);
-    mkdir(work, 0755);
-    mkdir(merged, 0755);
+    if (g_upper && g_work) {
+      /* host-disk-backed upper/work (build capture): no tmpfs scratch */
+      snprintf(upper, sizeof upper, "%s", g_upper);
+      snprintf(work, sizeof work, "%s", g_work);
+      snprintf(merged, sizeof merged, "%s/merged", g_overlay_dir);
+      mkdir(upper, 0755);
+      mkdir(work, 0755);
+      mkdir(merged, 0755);
+    } else {
+      snprintf(upper, sizeof upper, "%s/upper", g_overlay_dir);
+      snprintf(work, sizeof work, "%s/work", g_overlay_dir);
+      snprintf(merged, sizeof merged, "%s/merged", g_overlay_dir);
+      /* upper/work must be on tmpfs so nesting on any host fs works */
+      if (mount("tmpfs", g_overlay_dir, "tmpfs", 0, "size=90%") != 0)
+        die("tmpfs for overlay scratch");
+      mkdir(upper, 0755);
+      mkdir(work, 0755);
+      mkdir(merged, 0755);
+    }
+    snprintf(lower, sizeof lower, "%s/lower", g_overlay_dir);
     struct stat st;
-    const char *lowerdir = (stat(lower, &st) == 0) ? lower : "/";
+    const char *lowerdir;
+    if (g_lower) lowerdir = g_lower;            /* image layer chain */
+    else lowerdir = (stat(lower, &st) == 0) ? lower : "/";
     snprintf(opts, sizeof opts, "lowerdir=%s,upperdir=%s,workdir=%s",
              lowerdir, upper, work);
     if (mount("overlay", merged, "overlay", 0, opts) != 0) die("overlay mount");
@@ -241,6 +263,9 @@ int main(int argc, char **argv) {
     else if (strcmp(argv[i], "--env") == 0 && g_nenvs < MAX_ENVS) g_envs[g_nenvs++] = argv[++i];
     else if (strcmp(argv[i], "--cwd") == 0) g_cwd = argv[++i];
     else if (strcmp(argv[i], "--block-network") == 0) g_block_network = 1;
+    else if (strcmp(argv[i], "--lower") == 0) g_lower = argv[++i];
+    else if (strcmp(argv[i], "--upper") == 0) g_upper = argv[++i];
+    else if (strcmp(argv[i], "--work") == 0) g_work = argv[++i];
     else if (strcmp(argv[i], "--cgroup-parent") == 0) g_cgroup_parent = argv[++i];
     else { fprintf(stderr, "b9launch: unknown arg %s\n", argv[i]); return 125; }
   }

@@ -1,0 +1,226 @@
+"""OCI image plane: layer store, whiteouts, OCI layout import, rootfs
+containers via b9launch, rootfs builds capturing the overlay upper.
+
+Reference capability being reproduced: pkg/worker/image.go:278-348 (PullLazy
+content-addressed image mounts), pkg/abstractions/image/build.go:62-343
+(build containers), pkg/common/overlay.go (layered overlayfs roots).
+"""
+
+import hashlib
+import io
+import json
+import os
+import shutil
+import subprocess
+import tarfile
+
+import pytest
+
+from beta9_amd.images.store import ImageStore, LayerStore
+from beta9_amd.worker.runtime.native import namespaces_available
+
+needs_ns = pytest.mark.skipif(not namespaces_available(), reason="no namespace privileges")
+
+
+def _tar_bytes(entries):
+    """entries: list of (name, content|None for dir)"""
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w") as tf:
+        for name, content in entries:
+            if content is None:
+                info = tarfile.TarInfo(name)
+                info.type = tarfile.DIRTYPE
+                info.mode = 0o755
+                tf.addfile(info)
+            else:
+                data = content.encode() if isinstance(content, str) else content
+                info = tarfile.TarInfo(name)
+                info.size = len(data)
+                info.mode = 0o644
+                tf.addfile(info, io.BytesIO(data))
+    return buf.getvalue()
+
+
+def test_layer_extract_and_whiteouts(tmp_path):
+    store = LayerStore(str(tmp_path))
+    l1 = _tar_bytes([("etc", None), ("etc/keep.conf", "keep"), ("etc/gone.conf", "bye"),
+                     ("usr", None), ("usr/f1", "one")])
+    d1 = store.add_layer_from_tar(io.BytesIO(l1))
+    assert store.has_layer(d1)
+    fs1 = store.layer_fs(d1)
+    assert open(os.path.join(fs1, "etc/keep.conf")).read() == "keep"
+
+    # layer 2 whites out etc/gone.conf
+    l2 = _tar_bytes([("etc", None), ("etc/.wh.gone.conf", ""), ("etc/new.conf", "new")])
+    d2 = store.add_layer_from_tar(io.BytesIO(l2))
+    fs2 = store.layer_fs(d2)
+    wh = os.path.join(fs2, "etc/gone.conf")
+    st = os.stat(wh)
+    assert st.st_rdev == 0 and not st.st_size  # char 0:0 whiteout
+
+    # dedupe: same bytes -> same digest, no re-extract
+    assert store.add_layer_from_tar(io.BytesIO(l1)) == d1
+
+    # path traversal rejected
+    evil = _tar_bytes([("../../escape", "nope")])
+    d3 = store.add_layer_from_tar(io.BytesIO(evil))
+    assert not os.path.exists(tmp_path.parent / "escape")
+    assert store.has_layer(d3)
+
+
+def _make_oci_layout(root, layers, config=None):
+    """Minimal OCI image layout."""
+    blobs = os.path.join(root, "blobs", "sha256")
+    os.makedirs(blobs, exist_ok=True)
+
+    def put_blob(data):
+        h = hashlib.sha256(data).hexdigest()
+        with open(os.path.join(blobs, h), "wb") as f:
+            f.write(data)
+        return "sha256:" + h, len(data)
+
+    layer_descs = []
+    for tar in layers:
+        dig, size = put_blob(tar)
+        layer_descs.append(
+            {"mediaType": "application/vnd.oci.image.layer.v1.tar", "digest": dig, "size": size}
+        )
+    cfg = config or {"config": {"Env": ["MARKER=oci"], "Entrypoint": ["/bin/sh"], "WorkingDir": "/srv"}}
+    cfg_bytes = json.dumps(cfg).encode()
+    cfg_dig, cfg_size = put_blob(cfg_bytes)
+    manifest = {
+        "schemaVersion": 2,
+        "config": {"mediaType": "application/vnd.oci.image.config.v1+json", "digest": cfg_dig, "size": cfg_size},
+        "layers": layer_descs,
+    }
+    mbytes = json.dumps(manifest).encode()
+    mdig, msize = put_blob(mbytes)
+    with open(os.path.join(root, "index.json"), "w") as f:
+        json.dump({"schemaVersion": 2, "manifests": [
+            {"mediaType": "application/vnd.oci.image.manifest.v1+json", "digest": mdig, "size": msize}
+        ]}, f)
+    with open(os.path.join(root, "oci-layout"), "w") as f:
+        json.dump({"imageLayoutVersion": "1.0.0"}, f)
+
+
+def test_import_oci_layout(tmp_path):
+    layout = tmp_path / "layout"
+    os.makedirs(layout)
+    l1 = _tar_bytes([("srv", None), ("srv/base.txt", "base")])
+    l2 = _tar_bytes([("srv", None), ("srv/top.txt", "top")])
+    _make_oci_layout(str(layout), [l1, l2])
+
+    store = ImageStore(str(tmp_path / "store"))
+    image_id = store.import_oci(str(layout))
+    m = store.get_manifest(image_id)
+    assert m["kind"] == "rootfs" and len(m["layers"]) == 2
+    assert m["config"]["env"] == ["MARKER=oci"]
+    assert m["config"]["entrypoint"] == ["/bin/sh"]
+    assert m["config"]["cwd"] == "/srv"
+    # lowerdirs: topmost first (layer 2 shadows layer 1)
+    dirs = store.lowerdirs(image_id)
+    assert os.path.exists(os.path.join(dirs[0], "srv/top.txt"))
+    assert os.path.exists(os.path.join(dirs[1], "srv/base.txt"))
+    # re-import is idempotent
+    assert store.import_oci(str(layout)) == image_id
+
+
+def _mini_rootfs_tar():
+    """A runnable rootfs: /bin/sh plus the shared libs it needs, as a tar."""
+    buf = io.BytesIO()
+    sh = os.path.realpath("/bin/sh")
+    libs = set()
+    out = subprocess.run(["ldd", sh], capture_output=True, text=True).stdout
+    for line in out.splitlines():
+        parts = line.split()
+        for p in parts:
+            if p.startswith("/") and os.path.exists(p):
+                libs.add(p)  # ORIGINAL path: the ELF interp header uses it
+
+    def add_file(tf, src, arcname):
+        with open(src, "rb") as f:
+            data = f.read()
+        info = tarfile.TarInfo(arcname.lstrip("/"))
+        info.size = len(data)
+        info.mode = 0o755
+        tf.addfile(info, io.BytesIO(data))
+
+    with tarfile.open(fileobj=buf, mode="w") as tf:
+        add_file(tf, sh, "bin/sh")
+        for lib in libs:
+            add_file(tf, os.path.realpath(lib), lib)
+        for d in ("etc", "srv", "root"):
+            info = tarfile.TarInfo(d)
+            info.type = tarfile.DIRTYPE
+            info.mode = 0o755
+            tf.addfile(info)
+    return buf.getvalue()
+
+
+@needs_ns
+@pytest.mark.asyncio
+async def test_rootfs_container_runs_from_layers(tmp_path):
+    """A container boots from pure image layers (its own rootfs, not the
+    host's) with host files invisible."""
+    from beta9_amd.worker.runtime.base import ContainerSpec
+    from beta9_amd.worker.runtime.native import NativeRuntime
+
+    store = ImageStore(str(tmp_path / "store"))
+    base = store.layers.add_layer_from_tar(io.BytesIO(_mini_rootfs_tar()))
+    top = store.layers.add_layer_from_tar(
+        io.BytesIO(_tar_bytes([("srv", None), ("srv/hello.txt", "from-layer")]))
+    )
+    store.put_manifest("img-test", [base, top], {"env": [], "entrypoint": []}, kind="rootfs")
+
+    rt = NativeRuntime(scratch_path=str(tmp_path / "scratch"))
+    lines = []
+    spec = ContainerSpec(
+        container_id="rootfs-1",
+        entry_point=["/bin/sh", "-c",
+                     "read line < /srv/hello.txt; echo $line; "
+                     "if [ -e /root/repo ]; then echo host-visible; else echo isolated; fi; "
+                     ": > /srv/scratch"],
+        env={"PATH": "/bin"},
+        image_lowerdirs=store.lowerdirs("img-test"),
+        hostname="rootfs-test",
+    )
+    await rt.run(spec, on_log=lambda s, l: lines.append(l))
+    code = await rt.wait("rootfs-1")
+    await rt.delete("rootfs-1")
+    assert code == 0, lines
+    assert lines[0] == "from-layer"
+    assert lines[1] == "isolated"  # host filesystem is NOT visible: own rootfs
+    # write went to the overlay upper, not the layer store
+    assert not os.path.exists(os.path.join(store.layers.layer_fs(top), "srv/scratch"))
+
+
+@needs_ns
+@pytest.mark.asyncio
+async def test_rootfs_build_captures_upper(tmp_path):
+    """A rootfs build runs steps inside the base image and captures the fs
+    delta as a new content-addressed layer."""
+    import asyncio
+
+    import beta9_amd
+
+    from .helpers import cluster
+
+    async with cluster(n_workers=1) as c:
+        svc = c.gateway.image_service
+        base_id = svc.store.put_manifest(
+            "mini-base",
+            [svc.store.layers.add_layer_from_tar(io.BytesIO(_mini_rootfs_tar()))],
+            {"env": [], "entrypoint": ["/bin/sh"]},
+            kind="rootfs",
+        )["image_id"]
+
+        img = beta9_amd.Image(
+            base_image="mini-base",
+            commands=["echo built-content > /srv/built.txt"],
+        )
+        image_id = await asyncio.to_thread(img.build)
+        m = svc.store.get_manifest(image_id)
+        assert m["kind"] == "rootfs"
+        assert len(m["layers"]) == 2  # base + captured delta
+        top_fs = svc.store.layers.layer_fs(m["layers"][-1])
+        assert open(os.path.join(top_fs, "srv/built.txt")).read().strip() == "built-content"
