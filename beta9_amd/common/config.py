@@ -19,8 +19,9 @@ from pydantic import BaseModel, Field
 class DatabaseConfig(BaseModel):
     # sqlite path; ":memory:" for tests. (The reference uses Postgres; this
     # framework embeds its backend store so a single-node MI355X deployment
-    # has no external dependencies.)
-    path: str = "/var/lib/b9/backend.db"
+    # has no external dependencies.) Lives in its own subdir so the worker
+    # can tmpfs-mask it out of containers' view of the host rootfs.
+    path: str = "/var/lib/b9/db/backend.db"
 
 
 class GatewayConfig(BaseModel):
@@ -166,7 +167,7 @@ class AppConfig(BaseModel):
     def with_root(self, root: str) -> "AppConfig":
         """Re-root all on-disk paths under `root` (tests, per-rank benches)."""
         cfg = self.model_copy(deep=True)
-        cfg.database.path = os.path.join(root, "backend.db")
+        cfg.database.path = os.path.join(root, "db/backend.db")
         cfg.image_service.store_path = os.path.join(root, "images")
         cfg.storage.root = os.path.join(root, "workspaces")
         cfg.cache.root = os.path.join(root, "cache")

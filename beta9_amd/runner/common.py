@@ -95,6 +95,13 @@ def setup_environment(env: RunnerEnv) -> None:
         if env.workspace_dir not in sys.path:
             sys.path.insert(0, env.workspace_dir)
         os.chdir(env.workspace_dir)
+    elif os.getcwd() == "/":
+        # native runtime pivots to "/": give the runner a private scratch
+        # workspace instead of the container root
+        ws = "/tmp/workspace"
+        os.makedirs(ws, exist_ok=True)
+        os.chdir(ws)
+        sys.path.insert(0, ws)
     else:
         sys.path.insert(0, os.getcwd())
 

@@ -75,6 +75,8 @@ class Lifecycle:
         worker_host: str = "127.0.0.1",
         prewarm=None,  # PrewarmPool
         image_store_path: str = "",
+        isolation: str = "native",
+        mask_paths=None,
     ):
         self.worker_id = worker_id
         self.repo = repo
@@ -86,14 +88,19 @@ class Lifecycle:
         self.prewarm = prewarm
         self.image_store_path = image_store_path
         self.images = None  # ImageStore, lazy (store may not exist yet)
+        self.isolation = isolation
+        self.mask_paths = list(mask_paths or [])
         self.containers: Dict[str, ContainerInstance] = {}
         from .usage import UsageTracker
 
         self.usage = UsageTracker(repo)
-        # native runtime for containers needing mount/network isolation
-        # (volumes bind-mounted into an overlay rootfs; --block-network)
+        # b9launch-backed runtime: THE DEFAULT for user code when the host
+        # permits namespaces (reference: runc is the default; bare host
+        # processes are the explicit opt-out via isolation="proc")
         self.native_runtime = None
-        if type(runtime).__name__ != "NativeRuntime":
+        if type(runtime).__name__ == "NativeRuntime":
+            self.native_runtime = runtime
+        else:
             try:
                 from .runtime.native import NativeRuntime, namespaces_available
 
@@ -103,12 +110,16 @@ class Lifecycle:
                 self.native_runtime = None
         self._container_runtime: Dict[str, object] = {}
 
+    def _native_default(self) -> bool:
+        return self.native_runtime is not None and self.isolation == "native"
+
     def _runtime_for(self, spec) -> "Runtime":
         needs_native = (
             spec.mounts
             or spec.block_network
             or spec.image_lowerdirs
             or spec.upper_dir
+            or self._native_default()
         )
         if self.native_runtime is not None and needs_native:
             self._container_runtime[spec.container_id] = self.native_runtime
@@ -157,12 +168,20 @@ class Lifecycle:
                 handle = None
                 runtime = self._runtime_for(spec)
                 # Prewarmed shell path: torch+HIP context already up on the
-                # assigned GPU; the shell becomes the container.
-                if (
+                # assigned GPU; the shell becomes the container. (Shells are
+                # themselves b9launch containers when isolation is on, so a
+                # prewarm hit never bypasses the isolation default.)
+                user_mounts = [m for m in spec.mounts if not m.get("system")]
+                prewarm_ok = (
                     self.prewarm is not None
-                    and runtime is self.runtime
                     and hasattr(runtime, "adopt")
-                ):
+                    and not user_mounts
+                    and not spec.image_lowerdirs
+                    and not spec.upper_dir
+                    and not spec.block_network
+                    and self.prewarm.sees_path(spec.env.get("B9_WORKSPACE_DIR", ""))
+                )
+                if prewarm_ok:
                     from .prewarm import runner_module_from_entry
 
                     module = runner_module_from_entry(spec.entry_point)
@@ -367,6 +386,23 @@ class Lifecycle:
                 from ..abstractions.image import image_env
 
                 env.update(image_env(self.image_store_path, request.image_id))
+                if self._native_default() and self.mask_paths:
+                    # re-expose the image store through the data-root mask
+                    # ("system" mounts don't disqualify the prewarm path)
+                    extra_mounts.append(
+                        {
+                            "local_path": self.image_store_path,
+                            "mount_path": self.image_store_path,
+                            "system": True,
+                        }
+                    )
+        wd = env.get("B9_WORKSPACE_DIR", "")
+        if wd and self._native_default() and os.path.isdir(wd):
+            # fresh /tmp + masks hide host paths: the synced workspace code
+            # rides in on an explicit system bind
+            extra_mounts.append(
+                {"local_path": wd, "mount_path": wd, "system": True}
+            )
         if needs_addr:
             inst.port = free_port()
             env["BIND_PORT"] = str(inst.port)
@@ -390,6 +426,7 @@ class Lifecycle:
             block_network=request.block_network,
             mounts=[m.model_dump() for m in request.mounts] + extra_mounts,
             image_lowerdirs=lowerdirs,
+            mask_paths=self.mask_paths,
         )
         return spec, needs_addr
 

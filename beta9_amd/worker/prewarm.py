@@ -68,22 +68,40 @@ class PrewarmPool:
         cpu_shells: int = 2,
         base_env: Optional[Dict[str, str]] = None,
         visible_env_names: Optional[List[str]] = None,
+        shells_per_gpu: int = 2,
+        isolate: bool = False,
+        device_paths: Optional[Dict[int, List[str]]] = None,
+        mask_paths: Optional[List[str]] = None,
+        bind_paths: Optional[List[str]] = None,
     ):
+        """isolate=True parks each shell INSIDE a b9launch container
+        (namespaces + overlay root + only its own GPU nodes in /dev), so a
+        prewarm hit is an isolated container too — the cold-start headline
+        never bypasses the isolation default. The shell's unix socket dir is
+        bind-mounted through; activation works unchanged."""
         self.gpu_ordinals = gpu_ordinals
         self.cpu_shells = cpu_shells
+        self.shells_per_gpu = max(1, shells_per_gpu)
         self.base_env = base_env or {}
         self.visible_env_names = visible_env_names or [
             "HIP_VISIBLE_DEVICES",
             "ROCR_VISIBLE_DEVICES",
         ]
+        self.isolate = isolate
+        self.device_paths = device_paths or {}
+        self.mask_paths = mask_paths or []
+        self.bind_paths = bind_paths or []
         self._shells: Dict[Tuple[int, ...], List[PrewarmedShell]] = {}
         self._dir = tempfile.mkdtemp(prefix="b9-prewarm-")
+        self._scratch = tempfile.mkdtemp(prefix="b9-prewarm-ov-")
         self._replenish_lock = asyncio.Lock()
         self._n = 0
         self._stopping = False
 
     async def start(self) -> None:
-        tasks = [self._spawn((o,)) for o in self.gpu_ordinals]
+        tasks = []
+        for o in self.gpu_ordinals:
+            tasks += [self._spawn((o,)) for _ in range(self.shells_per_gpu)]
         tasks += [self._spawn(()) for _ in range(self.cpu_shells)]
         await asyncio.gather(*tasks)
 
@@ -103,11 +121,29 @@ class PrewarmPool:
             env["WORKER_GPU_DEVICES"] = vis
         else:
             env.pop("WORKER_GPU_DEVICES", None)
+        argv = [sys.executable, "-u", "-m", "beta9_amd.runner.prewarm"]
+        if self.isolate:
+            from .runtime.native import LAUNCHER
+
+            wrap = [
+                LAUNCHER,
+                "--rootfs-overlay",
+                os.path.join(self._scratch, f"shell-{self._n}"),
+                "--hostname",
+                f"prewarm-{self._n}",
+                "--bind",
+                f"{self._dir}:{self._dir}",  # the park socket crosses the ns
+            ]
+            for mask in self.mask_paths:
+                wrap += ["--mask", mask]
+            for b in self.bind_paths:
+                wrap += ["--bind", f"{b}:{b}"]
+            for o in key:
+                for dev in self.device_paths.get(o, []):
+                    wrap += ["--dev", dev]
+            argv = wrap + ["--"] + argv
         proc = await asyncio.create_subprocess_exec(
-            sys.executable,
-            "-u",
-            "-m",
-            "beta9_amd.runner.prewarm",
+            *argv,
             env=env,
             stdout=asyncio.subprocess.PIPE,
             stderr=asyncio.subprocess.PIPE,
@@ -115,6 +151,16 @@ class PrewarmPool:
         )
         shell = PrewarmedShell(proc, sock_path, key)
         self._shells.setdefault(key, []).append(shell)
+
+    def sees_path(self, path: str) -> bool:
+        """Whether a parked shell's mount namespace can see `path` — shells
+        bind a fixed set of host dirs at spawn; activation cannot add
+        binds, so containers needing other paths skip the prewarm pool."""
+        if not path or not self.isolate:
+            return True
+        return any(
+            os.path.commonpath([path, b]) == b for b in self.bind_paths if b
+        )
 
     async def acquire(self, ordinals: List[int]) -> Optional[PrewarmedShell]:
         """Take a ready shell matching the GPU set (or CPU for [])."""

@@ -33,9 +33,17 @@ def launcher_available() -> bool:
     return os.access(LAUNCHER, os.X_OK)
 
 
+_NS_AVAILABLE: Optional[bool] = None
+
+
 def namespaces_available() -> bool:
-    """Probe whether this host permits the namespace/overlay setup."""
+    """Probe whether this host permits the namespace/overlay setup
+    (cached: one b9launch round-trip per process)."""
+    global _NS_AVAILABLE
+    if _NS_AVAILABLE is not None:
+        return _NS_AVAILABLE
     if not launcher_available():
+        _NS_AVAILABLE = False
         return False
     import subprocess
 
@@ -45,9 +53,10 @@ def namespaces_available() -> bool:
             capture_output=True,
             timeout=15,
         )
-        return r.returncode == 0
+        _NS_AVAILABLE = r.returncode == 0
     except (subprocess.TimeoutExpired, OSError):
-        return False
+        _NS_AVAILABLE = False
+    return _NS_AVAILABLE
 
 
 class NativeRuntime(ProcRuntime):
@@ -81,6 +90,10 @@ class NativeRuntime(ProcRuntime):
             argv += ["--lower", ":".join(spec.image_lowerdirs)]
         if spec.upper_dir and spec.work_dir:
             argv += ["--upper", spec.upper_dir, "--work", spec.work_dir]
+        if not spec.image_lowerdirs:
+            # host-rootfs mode: mask the platform's data roots
+            for mask in spec.mask_paths:
+                argv += ["--mask", mask]
         if spec.cpu_millicores > 0:
             argv += ["--cpu", str(spec.cpu_millicores)]
         if spec.memory_mb > 0:
@@ -101,6 +114,12 @@ class NativeRuntime(ProcRuntime):
         from .base import container_base_env
 
         env = container_base_env(spec.env)
+        if not spec.image_lowerdirs:
+            # host-rootfs mode: the runner package must be importable even
+            # though the container's cwd is / after pivot (the worker may
+            # rely on cwd for imports; containers cannot)
+            extra = env.get("PYTHONPATH", "")
+            env["PYTHONPATH"] = (extra + os.pathsep if extra else "") + REPO_ROOT
         proc = await asyncio.create_subprocess_exec(
             *argv,
             env=env,

@@ -76,3 +76,19 @@ def batched_echo(batch=None, **kw):
 
 
 batched_echo._b9_batch = {"max_size": 16, "wait_ms": 20.0}
+
+
+def probe_paths(paths=None, **kw):
+    """Existence probe used by isolation tests."""
+    import os as _os
+
+    return {p: _os.path.exists(p) for p in (paths or [])}
+
+
+def pidns_probe(**kw):
+    """PID-namespace + overlay-write probe used by isolation tests."""
+    import os as _os
+
+    with open("/isolation-canary", "w") as f:
+        f.write("x")
+    return {"pid": _os.getpid(), "canary": _os.path.exists("/isolation-canary")}

@@ -61,6 +61,7 @@ class Cluster:
                 gateway_host=self.cfg.gateway.host,
                 gateway_grpc_port=self.gateway.grpc_port,
                 image_store_path=self.cfg.image_service.store_path,
+                data_root=self.tmpdir,
             )
             await w.start()
             self.workers.append(w)

@@ -49,7 +49,7 @@ async def test_native_runtime_isolation(tmp_path):
     await rt.delete("nat-1")
     assert code == 0
     assert lines[0] == "nat-host"  # UTS namespace
-    assert lines[1] == "1"  # PID namespace: shell is pid 1
+    assert lines[1] == "2"  # PID namespace: b9launch mini-init is 1, payload is 2
     assert not os.path.exists("/isolated-write")  # overlay upper, not host
     assert int(lines[2]) < 15  # minimal /dev
 
