@@ -3,17 +3,25 @@
 through the full platform path.
 
 BASELINE.json metric: "p50 GPU-container cold-start (ms) + tasks/sec at
-1/2/4/8 concurrent MI355X". Each rank (one per GPU, launched by
-torch.distributed.run for N>1) runs a complete in-process node — gateway +
-scheduler + dispatcher + embedded worker pinned to its GPU — deploys the
-bf16 synthetic MLP endpoint (random-init weights; no network for datasets),
-and drives requests through the REAL dispatch path: gateway task dispatch ->
-request buffer -> container subprocess (HTTP) -> hand-written gfx950 MFMA
-kernels. A step = `--concurrency` requests per rank completed end-to-end.
+1/2/4/8 concurrent MI355X".
 
-Cold start (scale 0 -> first healthy response, containing HIP init + HBM
-arena prealloc + weight init) is measured per rank during setup and reported
-as cold_start_p50_ms; the timed region measures serving tasks/sec.
+Topology (round-2 rework, VERDICT weak #3): ONE gateway + scheduler +
+N-GPU worker live on rank 0 — the scheduler bin-packs N containers onto the
+one node, exactly BASELINE's 1/2/4/8-concurrent shape. Ranks 1..N-1 (one
+per GPU, launched by torch.distributed.run) are load generators driving
+their own stub through the rank-0 gateway over real gRPC. Collectives
+(barriers, timing reduction) run over RCCL ("nccl" backend on ROCm) so the
+N>1 bench itself executes RCCL over xGMI.
+
+Cold start reporting (VERDICT weak #1): three populations —
+  - cold_start_concurrent_ms: N stubs scale 0->1 simultaneously (the
+    N-concurrent storm; one sample per rank);
+  - cold_start_all_ms: >=10 sequential prewarm-HIT cold starts (fresh stub,
+    ready shell) on rank 0;
+  - cold_start_miss_all_ms: >=10 sequential prewarm-MISS cold starts
+    (B9_DISABLE_PREWARM=1: full b9launch + python + torch + HIP init +
+    model load path).
+The timed region measures serving tasks/sec only.
 """
 
 from __future__ import annotations
@@ -39,50 +47,13 @@ def get_dist():
     return rank, world, local_rank
 
 
-async def build_node(tmpdir: str, local_rank: int, args):
-    """One in-process node: gateway + embedded worker pinned to one GPU."""
-    from beta9_amd.common.config import AppConfig
-    from beta9_amd.gateway.gateway import Gateway
-    from beta9_amd.worker.gpu import GPUManager, discover_devices, _torch_fallback_devices
-    from beta9_amd.worker.repo_client import DirectRepoClient
-    from beta9_amd.worker.worker import Worker
+def stub_cfg(args, has_gpu: bool, name_env: dict | None = None):
+    from beta9_amd.types import AutoscalerConfig, RuntimeConfig, StubConfig, TaskPolicy
 
-    cfg = AppConfig().with_root(tmpdir)
-    cfg.gateway.grpc_port = 0
-    cfg.gateway.http_port = 0
-    cfg.database.path = ":memory:"
-    gw = Gateway(cfg)
-    await gw.start()
-    creds = gw.bootstrap_workspace("bench")
-
-    devices = discover_devices(cfg.gpu.kfd_topology_path) or _torch_fallback_devices()
-    mine = [d for d in devices if d.ordinal == local_rank]
-    gpu_manager = GPUManager(devices=mine) if mine else None
-    repo = DirectRepoClient(gw.worker_repo, gw.container_repo, events=gw.events)
-    worker = Worker(
-        worker_id=f"bench-w{local_rank}",
-        repo=repo,
-        cpu_limit=64000,
-        memory_limit_mb=262144,
-        gpu_type=gpu_manager.gpu_type if gpu_manager else "",
-        gpu_manager=gpu_manager,
-        gateway_host=cfg.gateway.host,
-        gateway_grpc_port=gw.grpc_port,
-        # GPU bench: shells for the GPU only (no CPU shells; at N=8 ranks
-        # that is 16 fewer concurrent torch imports), CPU dev box: one
-        prewarm_cpu_shells=0 if (gpu_manager and gpu_manager.total) else 1,
-    )
-    await worker.start()
-    return gw, worker, creds, gpu_manager
-
-
-async def deploy_endpoint(gw, creds, args, has_gpu: bool):
-    from beta9_amd.types import RuntimeConfig, StubConfig, TaskPolicy
-
-    cfg = StubConfig(
+    return StubConfig(
         runtime=RuntimeConfig(
-            cpu=8000,
-            memory_mb=65536,
+            cpu=4000,
+            memory_mb=24576,
             gpu="MI355X" if has_gpu else "",
             gpus=["MI355X"] if has_gpu else [],
             gpu_count=1 if has_gpu else 0,
@@ -99,57 +70,267 @@ async def deploy_endpoint(gw, creds, args, has_gpu: bool):
             "B9_BENCH_INTER": str(args.inter),
             "B9_BENCH_LAYERS": str(args.layers),
             "B9_BENCH_MODEL": args.model,
+            **(name_env or {}),
         },
-        autoscaler=__import__("beta9_amd.types", fromlist=["AutoscalerConfig"]).AutoscalerConfig(
+        autoscaler=AutoscalerConfig(
             max_containers=1, tasks_per_container=max(32, args.concurrency * 2)
         ),
         timeout=600.0,
         task_policy=TaskPolicy(max_retries=0, timeout=600.0),
     )
-    ws = creds["workspace_id"]
-    app = gw.db.get_or_create_app(ws, "bench")
-    stub = gw.db.get_or_create_stub(ws, "bench-endpoint", "endpoint/deployment", cfg, app_id=app["id"])
-    gw.db.create_deployment(ws, "bench-endpoint", stub["id"], "endpoint/deployment", app_id=app["id"])
-    return stub["id"], ws
 
 
-async def one_request(gw, ws: str, stub_id: str, tokens: int) -> dict:
-    task = await gw.dispatcher.send_and_execute(
-        "endpoint", ws, stub_id, {"method": "POST", "path": "/", "json": {"tokens": tokens}}
+async def build_node(tmpdir: str, args, n_gpus_wanted: int):
+    """THE node: one gateway + scheduler + one worker owning every GPU."""
+    from beta9_amd.common.config import AppConfig
+    from beta9_amd.gateway.gateway import Gateway
+    from beta9_amd.worker.gpu import GPUManager, _torch_fallback_devices, discover_devices
+    from beta9_amd.worker.repo_client import DirectRepoClient
+    from beta9_amd.worker.worker import Worker
+
+    cfg = AppConfig().with_root(tmpdir)
+    cfg.gateway.grpc_port = 0
+    cfg.gateway.http_port = 0
+    cfg.database.path = ":memory:"
+    gw = Gateway(cfg)
+    await gw.start()
+    creds = gw.bootstrap_workspace("bench")
+
+    devices = discover_devices(cfg.gpu.kfd_topology_path) or _torch_fallback_devices()
+    devices = devices[:n_gpus_wanted] if n_gpus_wanted else devices
+    gpu_manager = GPUManager(devices=devices) if devices else None
+    repo = DirectRepoClient(gw.worker_repo, gw.container_repo, events=gw.events)
+    n = len(devices) if devices else 0
+    worker = Worker(
+        worker_id="bench-node",
+        repo=repo,
+        cpu_limit=max(64000, 8000 * max(n, 1)),
+        memory_limit_mb=max(262144, 32768 * max(n, 1)),
+        gpu_type=gpu_manager.gpu_type if gpu_manager else "",
+        gpu_manager=gpu_manager,
+        gateway_host=cfg.gateway.host,
+        gateway_grpc_port=gw.grpc_port,
+        prewarm_cpu_shells=0 if (gpu_manager and gpu_manager.total) else 1,
+        prewarm_shells_per_gpu=2,
+        image_store_path=cfg.image_service.store_path,
+        data_root=tmpdir,
     )
-    done = await gw.dispatcher.wait(task.task_id, timeout=600.0)
-    result = await gw.dispatcher.get_result(task.task_id)
-    ok = done.status.value == "COMPLETE" and isinstance(result, dict) and result.get("status_code") == 200
-    return {"ok": ok, "result": result}
+    await worker.start()
+    return gw, worker, creds, gpu_manager
 
 
-async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
+def deploy_stub(gw, ws: str, args, has_gpu: bool, name: str):
+    cfg = stub_cfg(args, has_gpu)
+    app = gw.db.get_or_create_app(ws, "bench")
+    stub = gw.db.get_or_create_stub(ws, name, "endpoint/deployment", cfg, app_id=app["id"])
+    gw.db.create_deployment(ws, name, stub["id"], "endpoint/deployment", app_id=app["id"])
+    return stub["id"]
+
+
+async def rpc_invoke(client, stub_id: str, tokens: int, affinity: str) -> dict:
+    r = await client.call(
+        "EndpointService",
+        "EndpointInvoke",
+        {"stub_id": stub_id, "args": {"tokens": tokens}, "affinity": affinity,
+         "timeout": 600.0},
+        timeout=620.0,
+    )
+    ok = (
+        r.get("status") == "COMPLETE"
+        and isinstance(r.get("result"), dict)
+        and r["result"].get("status_code") == 200
+    )
+    return {"ok": ok, "result": r.get("result")}
+
+
+async def cold_iterations(gw, worker, ws, args, has_gpu: bool, n_iter: int,
+                          miss: bool) -> list:
+    """Sequential single cold starts on rank 0: fresh stub -> first healthy
+    response -> stop. miss=True bypasses the prewarm pool."""
+    from beta9_amd.abstractions.endpoint import get_instance
+
+    samples = []
+    for i in range(n_iter):
+        name = f"cold-{'miss' if miss else 'hit'}-{i}"
+        cfg = stub_cfg(args, has_gpu,
+                       name_env={"B9_DISABLE_PREWARM": "1"} if miss else None)
+        app = gw.db.get_or_create_app(ws, "bench")
+        stub = gw.db.get_or_create_stub(ws, name, "endpoint/deployment", cfg, app_id=app["id"])
+        stub_id = stub["id"]
+        if not miss and worker.prewarm_pool is not None:
+            # hit iterations must find a ready shell (replenish is async)
+            deadline = time.monotonic() + 240.0
+            while time.monotonic() < deadline:
+                shells = [
+                    s
+                    for lst in worker.prewarm_pool._shells.values()
+                    for s in lst
+                    if (not has_gpu) == (not s.key)
+                ]
+                if any(s.ready or os.path.exists(s.sock_path) for s in shells):
+                    break
+                await asyncio.sleep(0.1)
+        t0 = time.perf_counter()
+        task = await gw.dispatcher.send_and_execute(
+            "endpoint", ws, stub_id, {"method": "POST", "path": "/",
+                                      "json": {"tokens": args.tokens}},
+        )
+        done = await gw.dispatcher.wait(task.task_id, timeout=600.0)
+        result = await gw.dispatcher.get_result(task.task_id)
+        dt = (time.perf_counter() - t0) * 1000.0
+        ok = done.status.value == "COMPLETE" and isinstance(result, dict) and result.get("status_code") == 200
+        if not ok:
+            raise RuntimeError(f"cold iteration {name} failed: {str(result)[:300]}")
+        samples.append(dt)
+        # stop the container and wait for it to exit
+        inst = await get_instance(gw, stub_id)
+        cids = list((await gw.container_repo.containers_for_stub(stub_id)).keys())
+        await inst.shutdown()
+        for cid in cids:
+            await gw.scheduler.stop(cid)
+        deadline = time.monotonic() + 60.0
+        while time.monotonic() < deadline:
+            left = await gw.container_repo.containers_for_stub(stub_id)
+            if not left:
+                break
+            await asyncio.sleep(0.1)
+    return samples
+
+
+async def run_bench(args, rank: int, world: int, local_rank: int) -> dict:
     import torch
 
-    with tempfile.TemporaryDirectory() as tmpdir:
-        gw, worker, creds, gpu_manager = await build_node(tmpdir, local_rank, args)
+    use_dist = world > 1
+    dist = None
+    if use_dist:
+        import torch.distributed as dist  # noqa: F811
+
+    gw = worker = gpu_manager = None
+    tmpdir_ctx = None
+    info = {}
+    if rank == 0:
+        tmpdir_ctx = tempfile.TemporaryDirectory()
+        gw, worker, creds, gpu_manager = await build_node(tmpdir_ctx.name, args, args.gpus)
         has_gpu = gpu_manager is not None and gpu_manager.total > 0
-        stub_id, ws = await deploy_endpoint(gw, creds, args, has_gpu)
-        # prewarm shells (torch+HIP context) park before the measured cold
-        # start — they are part of worker boot, not of a container start
-        await worker.wait_prewarmed(timeout=180.0)
+        ws = creds["workspace_id"]
+        stub_ids = [
+            deploy_stub(gw, ws, args, has_gpu, f"bench-endpoint-r{r}")
+            for r in range(world if use_dist else 1)
+        ]
+        await worker.wait_prewarmed(timeout=420.0)
+        info = {
+            "addr": f"{gw.cfg.gateway.host}:{gw.grpc_port}",
+            "token": creds["token"],
+            "ws": ws,
+            "stub_ids": stub_ids,
+            "has_gpu": has_gpu,
+        }
+    if use_dist:
+        lst = [info]
+        # collectives run OFF the event loop: rank 0's loop must keep
+        # serving the gateway while other ranks are mid-request
+        await asyncio.to_thread(dist.broadcast_object_list, lst, 0)
+        info = lst[0]
+    has_gpu = info["has_gpu"]
+    my_stub = info["stub_ids"][rank]
 
-        # ---- cold start: scale 0 -> first healthy response -----------------
-        t0 = time.perf_counter()
-        first = await one_request(gw, ws, stub_id, args.tokens)
-        cold_ms = (time.perf_counter() - t0) * 1000.0
-        if not first["ok"]:
-            raise RuntimeError(f"cold-start request failed: {str(first)[:400]}")
+    from beta9_amd.rpc.channel import RpcClient
+
+    client = RpcClient(info["addr"], token=info["token"])
+
+    def dbg(msg):
+        if os.environ.get("B9_BENCH_DEBUG"):
+            print(f"[bench r{rank}] {msg}", file=sys.stderr, flush=True)
+
+    # ---- N-concurrent cold-start storm ------------------------------------
+    if use_dist:
+        await asyncio.to_thread(dist.barrier)
+    dbg(f"storm start stub={my_stub} addr={info['addr']}")
+    t0 = time.perf_counter()
+    first = await rpc_invoke(client, my_stub, args.tokens, str(rank))
+    cold_ms = (time.perf_counter() - t0) * 1000.0
+    dbg(f"storm done {cold_ms:.1f}ms ok={first['ok']}")
+    if not first["ok"]:
+        raise RuntimeError(f"rank {rank} cold-start failed: {str(first)[:400]}")
+    try:
+        first_load = first["result"]["body"]["result"]["load"]
+    except (KeyError, TypeError):
+        first_load = {}
+
+    async def step() -> int:
+        results = await asyncio.gather(
+            *(rpc_invoke(client, my_stub, args.tokens, str(rank))
+              for _ in range(args.concurrency))
+        )
+        failed = sum(1 for r in results if not r["ok"])
+        if failed:
+            raise RuntimeError(f"{failed}/{len(results)} requests failed in step")
+        return len(results)
+
+    dbg("warmup begin")
+    for _ in range(args.warmup):
+        await step()
+    dbg("warmup done")
+
+    # ---- timed region ------------------------------------------------------
+    if use_dist:
+        if torch.cuda.is_available() and local_rank < torch.cuda.device_count():
+            torch.cuda.set_device(local_rank)
+        await asyncio.to_thread(dist.barrier)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    dbg("timed begin")
+    t_start = time.perf_counter()
+    done_tasks = 0
+    for _ in range(args.steps):
+        done_tasks += await step()
+    dbg("timed end")
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t_start
+    if use_dist:
+        await asyncio.to_thread(dist.barrier)
+        dev = "cuda" if (dist.get_backend() == "nccl") else "cpu"
+        # float64 collectives: RCCL prefers f32/f64 sums; these run on the
+        # GPU under the nccl backend = real RCCL over xGMI
+        def _reduce():
+            t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            c = torch.tensor([cold_ms], dtype=torch.float64, device=dev)
+            colds = [torch.zeros(1, dtype=torch.float64, device=dev) for _ in range(world)]
+            dist.all_gather(colds, c)
+            n_tasks = torch.tensor([float(done_tasks)], dtype=torch.float64, device=dev)
+            dist.all_reduce(n_tasks, op=dist.ReduceOp.SUM)
+            if dev == "cuda":
+                torch.cuda.synchronize()
+            return float(t.item()), [float(x.item()) for x in colds], int(n_tasks.item())
+        elapsed_max, cold_list, total_tasks = await asyncio.to_thread(_reduce)
+        cold_concurrent = sorted(cold_list)
+    else:
+        elapsed_max = elapsed
+        cold_concurrent = [cold_ms]
+        total_tasks = done_tasks
+
+    dbg("reduction done")
+    # ---- sequential hit/miss cold-start iterations (rank 0, untimed) ------
+    cold_hit = []
+    cold_miss = []
+    spans = {}
+    boot_phases = {}
+    if rank == 0:
+        cold_hit = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
+                                         args.cold_iters, miss=False)
+        cold_miss = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
+                                          args.cold_miss_iters, miss=True)
         try:
-            first_load = first["result"]["body"]["result"]["load"]
-        except (KeyError, TypeError):
-            first_load = {}
-        boot_phases = {}
-        try:
-            addrs = await gw.container_repo.addresses_for_stub(stub_id)
+            cids = list((await gw.container_repo.containers_for_stub(my_stub)).keys())
+            if cids:
+                spans = {
+                    s["name"]: round(s["duration_ms"], 2)
+                    for s in await gw.container_repo.get_spans(cids[0])
+                }
+            addrs = await gw.container_repo.addresses_for_stub(my_stub)
             if addrs:
-                from beta9_amd.rpc.channel import RpcClient
-
                 rc = RpcClient(list(addrs.values())[0])
                 h = await rc.call("Runner", "Health", {}, timeout=5.0)
                 boot_phases = h.get("gpu_phases", {})
@@ -157,75 +338,27 @@ async def run_rank(args, rank: int, world: int, local_rank: int) -> dict:
         except Exception:
             pass
 
-        async def step() -> int:
-            results = await asyncio.gather(
-                *(one_request(gw, ws, stub_id, args.tokens) for _ in range(args.concurrency))
-            )
-            failed = sum(1 for r in results if not r["ok"])
-            if failed:
-                raise RuntimeError(f"{failed}/{len(results)} requests failed in step")
-            return len(results)
-
-        # ---- warmup --------------------------------------------------------
-        for _ in range(args.warmup):
-            await step()
-
-        # ---- timed region --------------------------------------------------
-        use_dist = world > 1
-        if use_dist:
-            import torch.distributed as dist
-
-            if torch.cuda.is_available() and local_rank < torch.cuda.device_count():
-                torch.cuda.set_device(local_rank)
-            dist.barrier()
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        t_start = time.perf_counter()
-        done_tasks = 0
-        for _ in range(args.steps):
-            done_tasks += await step()
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        elapsed = time.perf_counter() - t_start
-        if use_dist:
-            import torch.distributed as dist
-
-            dist.barrier()
-            t = torch.tensor([elapsed], dtype=torch.float64)
-            dist.all_reduce(t, op=dist.ReduceOp.MAX)
-            elapsed_max = float(t.item())
-            c = torch.tensor([cold_ms], dtype=torch.float64)
-            colds = [torch.zeros(1, dtype=torch.float64) for _ in range(world)]
-            dist.all_gather(colds, c)
-            cold_list = [float(x.item()) for x in colds]
-        else:
-            elapsed_max = elapsed
-            cold_list = [cold_ms]
-
-        spans = {}
-        try:
-            cids = list((await gw.container_repo.containers_for_stub(stub_id)).keys())
-            if cids:
-                spans = {
-                    s["name"]: round(s["duration_ms"], 2)
-                    for s in await gw.container_repo.get_spans(cids[0])
-                }
-        except Exception:
-            pass
-
-        for w in [worker]:
-            await w.shutdown()
+    dbg("cold iterations done")
+    await client.close()
+    if use_dist:
+        await asyncio.to_thread(dist.barrier)
+    dbg("final barrier done")
+    if rank == 0:
+        await worker.shutdown()
         await gw.shutdown()
+        tmpdir_ctx.cleanup()
 
-        return {
-            "elapsed_max": elapsed_max,
-            "tasks": done_tasks,
-            "cold_list": cold_list,
-            "native_gpu": has_gpu,
-            "spans": spans,
-            "first_load": first_load,
-            "boot_phases": boot_phases,
-        }
+    return {
+        "elapsed_max": elapsed_max,
+        "total_tasks": total_tasks,
+        "cold_concurrent": cold_concurrent,
+        "cold_hit": cold_hit,
+        "cold_miss": cold_miss,
+        "native_gpu": has_gpu,
+        "spans": spans,
+        "first_load": first_load,
+        "boot_phases": boot_phases,
+    }
 
 
 def main() -> int:
@@ -239,26 +372,34 @@ def main() -> int:
     p.add_argument("--inter", type=int, default=14336)
     p.add_argument("--layers", type=int, default=8)
     p.add_argument("--model", choices=["mlp", "transformer"], default="mlp")
+    p.add_argument("--cold-iters", type=int, default=10,
+                   help="sequential prewarm-hit cold starts to sample")
+    p.add_argument("--cold-miss-iters", type=int, default=10,
+                   help="sequential prewarm-miss cold starts to sample")
     p.add_argument("--dyn-batch", action="store_true",
-                   help="serve through the dynamic-batching handler "
-                        "(one coalesced forward per group of concurrent "
-                        "requests; same total tokens)")
+                   help="serve through the dynamic-batching handler")
     args = p.parse_args()
 
     rank, world, local_rank = get_dist()
     if world > 1:
+        import torch
+
         import torch.distributed as dist
 
-        dist.init_process_group(backend="gloo", init_method="env://")
+        backend = "nccl" if (
+            torch.cuda.is_available() and torch.cuda.device_count() >= world
+        ) else "gloo"
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend, init_method="env://")
 
-    r = asyncio.run(run_rank(args, rank, world, local_rank))
+    r = asyncio.run(run_bench(args, rank, world, local_rank))
 
     if rank == 0:
         n = world if world > 1 else args.gpus
-        total_tasks = r["tasks"] * (world if world > 1 else 1)
-        value = total_tasks / r["elapsed_max"]
-        cold_sorted = sorted(r["cold_list"])
-        p50 = statistics.median(cold_sorted)
+        value = r["total_tasks"] / r["elapsed_max"]
+        hit = sorted(r["cold_hit"]) or sorted(r["cold_concurrent"])
+        miss = sorted(r["cold_miss"])
         out = {
             "metric": "tasks_per_sec",
             "value": round(value, 3),
@@ -272,8 +413,12 @@ def main() -> int:
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
-            "cold_start_p50_ms": round(p50, 1),
-            "cold_start_all_ms": [round(x, 1) for x in cold_sorted],
+            "cold_start_p50_ms": round(statistics.median(hit), 1),
+            "cold_start_p95_ms": round(hit[min(len(hit) - 1, max(0, -(-len(hit) * 95 // 100) - 1))], 1),
+            "cold_start_all_ms": [round(x, 1) for x in hit],
+            "cold_start_miss_p50_ms": round(statistics.median(miss), 1) if miss else None,
+            "cold_start_miss_all_ms": [round(x, 1) for x in miss],
+            "cold_start_concurrent_ms": [round(x, 1) for x in sorted(r["cold_concurrent"])],
             "cold_start_phases_ms": r["spans"],
             "first_container_load": r["first_load"],
             "container_boot_phases_ms": r["boot_phases"],
@@ -282,7 +427,11 @@ def main() -> int:
                 "model": f"synthetic-{args.model}-{args.layers}L-{args.hidden}h-{args.inter}i",
                 "global_batch": args.concurrency * (world if world > 1 else 1),
                 "seq_len": args.tokens,
-                "dyn_batch": bool(getattr(args, "dyn_batch", False)), "parallelism": f"serving dp{world if world > 1 else 1} (1 GPU-container per rank)",
+                "dyn_batch": bool(getattr(args, "dyn_batch", False)),
+                "parallelism": (
+                    f"serving dp{world if world > 1 else 1} "
+                    "(ONE scheduler bin-packing 1 container/GPU on one node)"
+                ),
                 "requests_per_step_per_gpu": args.concurrency,
                 "baseline_metric": "p50 GPU-container cold-start (ms) + tasks/sec at 1/2/4/8 concurrent MI355X",
             },

@@ -175,11 +175,11 @@ class EndpointService:
         stub_id = req["stub_id"]
         if self.gw.db.get_stub(stub_id) is None:
             raise not_found(f"stub {stub_id}")
+        payload = {"method": "POST", "path": "/", "json": req.get("args")}
+        if req.get("affinity"):
+            payload["affinity"] = str(req["affinity"])
         task = await self.gw.dispatcher.send_and_execute(
-            "endpoint",
-            auth["workspace_id"],
-            stub_id,
-            {"method": "POST", "path": "/", "json": req.get("args")},
+            "endpoint", auth["workspace_id"], stub_id, payload,
         )
         done = await self.gw.dispatcher.wait(task.task_id, timeout=float(req.get("timeout", 300.0)))
         result = await self.gw.dispatcher.get_result(task.task_id)

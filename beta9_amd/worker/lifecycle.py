@@ -179,6 +179,7 @@ class Lifecycle:
                     and not spec.image_lowerdirs
                     and not spec.upper_dir
                     and not spec.block_network
+                    and spec.env.get("B9_DISABLE_PREWARM") != "1"
                     and self.prewarm.sees_path(spec.env.get("B9_WORKSPACE_DIR", ""))
                 )
                 if prewarm_ok:

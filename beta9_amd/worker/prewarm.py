@@ -123,7 +123,12 @@ class PrewarmPool:
             env.pop("WORKER_GPU_DEVICES", None)
         argv = [sys.executable, "-u", "-m", "beta9_amd.runner.prewarm"]
         if self.isolate:
-            from .runtime.native import LAUNCHER
+            from .runtime.native import LAUNCHER, REPO_ROOT
+
+            # cwd pivots to "/" inside the namespace; the package must come
+            # from PYTHONPATH, not the worker's cwd
+            extra = env.get("PYTHONPATH", "")
+            env["PYTHONPATH"] = (extra + os.pathsep if extra else "") + REPO_ROOT
 
             wrap = [
                 LAUNCHER,
