@@ -145,6 +145,13 @@ class CheckpointConfig(BaseModel):
     staging_buffer_mb: int = 256
 
 
+class StateConfig(BaseModel):
+    # Control-plane state durability: "" = in-memory (dies with the
+    # gateway); a path = WAL + snapshot persistence so backlog/claims
+    # survive gateway restarts (reference keeps this in Redis).
+    persist_path: str = "/var/lib/b9/state"
+
+
 class MonitoringConfig(BaseModel):
     metrics_enabled: bool = True
     # lifecycle span sink: "state" (queryable via API) and/or "jsonl" file
@@ -162,6 +169,7 @@ class AppConfig(BaseModel):
     task: TaskConfig = Field(default_factory=TaskConfig)
     gpu: GPUConfig = Field(default_factory=GPUConfig)
     checkpoint: CheckpointConfig = Field(default_factory=CheckpointConfig)
+    state: StateConfig = Field(default_factory=StateConfig)
     monitoring: MonitoringConfig = Field(default_factory=MonitoringConfig)
 
     def with_root(self, root: str) -> "AppConfig":
@@ -174,6 +182,7 @@ class AppConfig(BaseModel):
         cfg.checkpoint.store_path = os.path.join(root, "checkpoints")
         cfg.worker.images_path = os.path.join(root, "worker/images")
         cfg.worker.scratch_path = os.path.join(root, "worker/scratch")
+        cfg.state.persist_path = os.path.join(root, "state")
         return cfg
 
 

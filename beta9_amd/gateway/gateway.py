@@ -31,7 +31,12 @@ log = logging.getLogger(__name__)
 class Gateway:
     def __init__(self, cfg: Optional[AppConfig] = None):
         self.cfg = cfg or load_config()
-        self.store = StateStore()
+        if self.cfg.state.persist_path:
+            from ..common.persist import PersistentStateStore
+
+            self.store = PersistentStateStore(self.cfg.state.persist_path)
+        else:
+            self.store = StateStore()
         self.db = BackendDB(self.cfg.database.path)
         self.worker_repo = WorkerRepo(self.store, keepalive_ttl_s=self.cfg.worker.keepalive_ttl_s)
         self.container_repo = ContainerRepo(self.store)

@@ -195,6 +195,18 @@ class Scheduler:
                     except asyncio.TimeoutError:
                         pass
                     continue
+                # entries scored in the future are durable retry-backoffs
+                # (survive a gateway crash, unlike an in-memory sleep):
+                # push them back and only process what is ready
+                now = time.time()
+                ready = [(m, s) for m, s in popped if s <= now]
+                for m, s in popped:
+                    if s > now:
+                        await self.store.zadd(keys.scheduler_backlog(), m, s)
+                if not ready:
+                    await asyncio.sleep(min(poll if poll > 0 else 0.05, 0.05))
+                    continue
+                popped = ready
                 workers = await self._live_workers()
                 for container_id, _score in popped:
                     d = await self.store.get(keys.scheduler_request(container_id))
@@ -349,8 +361,7 @@ class Scheduler:
             self.cfg.scheduler.retry_backoff_max_s,
         )
 
-        async def _requeue() -> None:
-            await asyncio.sleep(backoff)
-            await self._enqueue(request, priority=request.timestamp)
-
-        asyncio.create_task(_requeue())
+        # durable backoff: the entry goes straight back into the persisted
+        # backlog with a future ready-time score; the loop skips it until
+        # then. A crash mid-backoff no longer loses the request.
+        await self._enqueue(request, priority=time.time() + backoff)
