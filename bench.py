@@ -318,10 +318,7 @@ async def run_bench(args, rank: int, world: int, local_rank: int) -> dict:
     spans = {}
     boot_phases = {}
     if rank == 0:
-        cold_hit = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
-                                         args.cold_iters, miss=False)
-        cold_miss = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
-                                          args.cold_miss_iters, miss=True)
+        # grab the serving container's spans/phases BEFORE stopping it
         try:
             cids = list((await gw.container_repo.containers_for_stub(my_stub)).keys())
             if cids:
@@ -331,12 +328,39 @@ async def run_bench(args, rank: int, world: int, local_rank: int) -> dict:
                 }
             addrs = await gw.container_repo.addresses_for_stub(my_stub)
             if addrs:
-                rc = RpcClient(list(addrs.values())[0])
-                h = await rc.call("Runner", "Health", {}, timeout=5.0)
+                rc2 = RpcClient(list(addrs.values())[0])
+                h = await rc2.call("Runner", "Health", {}, timeout=5.0)
                 boot_phases = h.get("gpu_phases", {})
-                await rc.close()
+                await rc2.close()
         except Exception:
             pass
+        # free the GPUs: the serving containers hold every device and the
+        # cold-start iterations each need one
+        from beta9_amd.abstractions.endpoint import get_instance
+
+        for sid in info["stub_ids"]:
+            try:
+                inst = await get_instance(gw, sid)
+                cids = list((await gw.container_repo.containers_for_stub(sid)).keys())
+                await inst.shutdown()
+                for cid in cids:
+                    await gw.scheduler.stop(cid)
+            except Exception:
+                pass
+        deadline = time.monotonic() + 90.0
+        while time.monotonic() < deadline:
+            busy = False
+            for sid in info["stub_ids"]:
+                if await gw.container_repo.containers_for_stub(sid):
+                    busy = True
+            if not busy:
+                break
+            await asyncio.sleep(0.2)
+        dbg("serving stubs stopped; cold iterations begin")
+        cold_hit = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
+                                         args.cold_iters, miss=False)
+        cold_miss = await cold_iterations(gw, worker, info["ws"], args, has_gpu,
+                                          args.cold_miss_iters, miss=True)
 
     dbg("cold iterations done")
     await client.close()
