@@ -94,6 +94,20 @@ def main() -> int:
     if os.environ.get("B9_BOOT_TRACE"):
         import time as _t
         print(f"[boottrace] shell spec received {_t.time():.6f}", flush=True)
+    if spec.get("want_netns"):
+        # the worker passes a network-namespace fd over the socket
+        # (SCM_RIGHTS); join it before any server binds happen
+        import ctypes
+
+        _msg, fds, _flags, _addr = socket.recv_fds(conn, 16, 1)
+        if fds:
+            libc = ctypes.CDLL("libc.so.6", use_errno=True)
+            CLONE_NEWNET = 0x40000000
+            if libc.setns(fds[0], CLONE_NEWNET) != 0:
+                err = ctypes.get_errno()
+                conn.sendall(f"err setns: {os.strerror(err)}\n".encode())
+                return 1
+            os.close(fds[0])
     os.environ.update({str(k): str(v) for k, v in spec.get("env", {}).items()})
     conn.sendall(b"ok\n")
     conn.close()

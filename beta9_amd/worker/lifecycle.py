@@ -61,6 +61,8 @@ class ContainerInstance:
         self.exit_code: Optional[int] = None
         self.started_at = 0.0
         self.done = asyncio.Event()
+        self.net_ip: str = ""  # per-container netns IP ("" = host network)
+        self.exposed_ports: list = []  # (host_port, container_port)
 
 
 class Lifecycle:
@@ -77,6 +79,7 @@ class Lifecycle:
         image_store_path: str = "",
         isolation: str = "native",
         mask_paths=None,
+        netplane=None,  # NetworkPlane
     ):
         self.worker_id = worker_id
         self.repo = repo
@@ -90,6 +93,7 @@ class Lifecycle:
         self.images = None  # ImageStore, lazy (store may not exist yet)
         self.isolation = isolation
         self.mask_paths = list(mask_paths or [])
+        self.netplane = netplane
         self.containers: Dict[str, ContainerInstance] = {}
         from .usage import UsageTracker
 
@@ -163,6 +167,35 @@ class Lifecycle:
                 spec.env["B9_CHECKPOINT_KEY"] = request.checkpoint.remote_key
                 spec.env["B9_CHECKPOINT_ID"] = request.checkpoint.checkpoint_id
 
+            # per-container network namespace (pods/sandboxes + any request
+            # exposing ports): preallocated slot with veth/bridge/IP
+            # (reference: network.go Setup + the slot pool :542-624)
+            net_slot = None
+            kind = StubType(request.stub_type).kind if request.stub_type else "function"
+            if (
+                self.netplane is not None
+                and self._native_default()
+                and not request.block_network
+                and (request.ports or kind in ("pod", "sandbox"))
+            ):
+                with rec.span("worker.network_setup"):
+                    net_slot, pool_hit = await self.netplane.acquire(cid)
+                    spec.netns_fd = net_slot.ns_fd()
+                    inst.net_ip = net_slot.ip
+                    if pool_hit:
+                        rec.record("worker.network_slot_hit", time.time(), time.time())
+                if request.ports:
+                    with rec.span("worker.network_expose_ports"):
+                        for cport in request.ports:
+                            hport = free_port()
+                            await self.netplane.expose_port(hport, net_slot.ip, cport)
+                            inst.exposed_ports.append((hport, cport))
+                if inst.port and self.worker_host != "127.0.0.1":
+                    # remote gateways can't route to the bridge subnet:
+                    # proxy the runner's bind port at the same host port
+                    await self.netplane.expose_port(inst.port, net_slot.ip, inst.port)
+                    inst.exposed_ports.append((inst.port, inst.port))
+
             with rec.span("worker.runtime_start"):
                 on_log = self._make_log_cb(cid)
                 handle = None
@@ -190,7 +223,8 @@ class Lifecycle:
                     if module is not None and len(ordinals) <= 1:
                         shell = await self.prewarm.acquire(ordinals)
                         if shell is not None:
-                            if await shell.activate(spec.env, module):
+                            nfd = net_slot.ns_fd() if net_slot is not None else None
+                            if await shell.activate(spec.env, module, netns_fd=nfd):
                                 handle = await runtime.adopt(spec, shell.proc, on_log)
                                 rec.record("worker.prewarm_hit", time.time(), time.time())
                             else:
@@ -216,7 +250,11 @@ class Lifecycle:
             )
             if needs_addr:
                 with rec.span("worker.publish_address"):
-                    addr = f"{self.worker_host}:{inst.port}"
+                    # netns containers are addressed by their bridge IP on
+                    # this host; remote gateways reach them through the
+                    # exposed-port proxies
+                    host = inst.net_ip if (inst.net_ip and self.worker_host == "127.0.0.1") else self.worker_host
+                    addr = f"{host}:{inst.port}"
                     await self.repo.set_container_address(cid, request.stub_id, addr)
 
             rec.record("worker.cold_start", start_total, time.time())
@@ -250,6 +288,16 @@ class Lifecycle:
             except Exception:
                 pass
             with rec.span("worker.cleanup"):
+                if self.netplane is not None:
+                    for hport, cport in inst.exposed_ports:
+                        try:
+                            await self.netplane.unexpose_port(hport, cport)
+                        except Exception:
+                            pass
+                    try:
+                        await self.netplane.release(cid)
+                    except Exception:
+                        pass
                 if gpu_assignment is not None and self.gpu is not None:
                     self.gpu.unassign(cid)
                 await self.runtime_of(cid).delete(cid)

@@ -107,6 +107,10 @@ class NativeRuntime(ProcRuntime):
                 argv += ["--bind", f"{local}:{target}"]
         if spec.block_network:
             argv += ["--block-network"]
+        pass_fds = ()
+        if spec.netns_fd >= 0 and not spec.block_network:
+            argv += ["--netns-fd", str(spec.netns_fd)]
+            pass_fds = (spec.netns_fd,)
         if spec.cwd:
             argv += ["--cwd", spec.cwd]
         argv += ["--"] + list(spec.entry_point)
@@ -126,6 +130,7 @@ class NativeRuntime(ProcRuntime):
             stdout=asyncio.subprocess.PIPE,
             stderr=asyncio.subprocess.PIPE,
             start_new_session=True,
+            pass_fds=pass_fds,
         )
         from .proc import _Proc
 

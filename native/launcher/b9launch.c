@@ -67,6 +67,7 @@ static const char *g_cgroup_parent = "b9";
 static const char *g_lower = NULL;   /* explicit lowerdir chain (topmost first) */
 static const char *g_upper = NULL;   /* host-backed upper (build capture) */
 static const char *g_work = NULL;    /* host-backed workdir */
+static int g_netns_fd = -1;          /* join a preallocated network namespace */
 static char **g_cmd = NULL;
 
 static void die(const char *msg) {
@@ -139,6 +140,13 @@ static void forward_to_payload(int sig) {
 
 static int child_main(void *arg) {
   (void)arg;
+  if (g_netns_fd >= 0) {
+    /* join the preallocated slot namespace (veth/IP already programmed by
+     * the worker's network plane) — the slot-pool fast path */
+    if (setns(g_netns_fd, CLONE_NEWNET) != 0)
+      die("setns netns");
+    close(g_netns_fd);
+  }
   if (g_hostname && sethostname(g_hostname, strlen(g_hostname)) != 0)
     fprintf(stderr, "b9launch: sethostname: %s\n", strerror(errno));
 
@@ -312,6 +320,7 @@ int main(int argc, char **argv) {
     else if (strcmp(argv[i], "--env") == 0 && g_nenvs < MAX_ENVS) g_envs[g_nenvs++] = argv[++i];
     else if (strcmp(argv[i], "--cwd") == 0) g_cwd = argv[++i];
     else if (strcmp(argv[i], "--block-network") == 0) g_block_network = 1;
+    else if (strcmp(argv[i], "--netns-fd") == 0) g_netns_fd = atoi(argv[++i]);
     else if (strcmp(argv[i], "--lower") == 0) g_lower = argv[++i];
     else if (strcmp(argv[i], "--upper") == 0) g_upper = argv[++i];
     else if (strcmp(argv[i], "--work") == 0) g_work = argv[++i];
