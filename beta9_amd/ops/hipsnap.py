@@ -97,7 +97,18 @@ class Arena:
         self._offset = 0
         ext = _load_ext()
         if ext is not None:
-            self._impl = ext.Arena(device, self.size_bytes, self.staging_bytes)
+            try:
+                self._impl = ext.Arena(device, self.size_bytes, self.staging_bytes)
+            except RuntimeError:
+                # extension built but no GPU on this host (CPU dev box with
+                # cross-compiled .so): degrade like the no-ext path — but
+                # NEVER silently on a GPU host
+                import torch
+
+                if torch.cuda.is_available():
+                    raise
+                self._impl = None
+                self._cpu_buf = bytearray(0)
         else:
             self._impl = None
             self._cpu_buf = bytearray(0)

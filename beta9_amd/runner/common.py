@@ -168,8 +168,30 @@ class GPUBoot:
                 t2 = time.perf_counter()
                 self.restored = hipsnap.restore_snapshot(self.arena, snap_path)
                 self.phases["hbm_restore_ms"] = (time.perf_counter() - t2) * 1000
+                if self.restored:
+                    t3 = time.perf_counter()
+                    self._load_object_state(snap_path)
+                    self.phases["obj_restore_ms"] = (time.perf_counter() - t3) * 1000
         except Exception:
             log.exception("GPU boot failed; continuing CPU-only")
+
+    def _load_object_state(self, snap_path: str) -> None:
+        """Rehydrate the pickled handler objects over the restored arena so
+        on_start is skipped (objsnap; VERDICT round-1 item #4)."""
+        global _restored_state
+        obj_path = snap_path + ".obj"
+        if not os.path.exists(obj_path):
+            return
+        try:
+            from ..ops import objsnap
+
+            with open(obj_path, "rb") as f:
+                state = objsnap.load_state(f.read(), self.arena)
+            if isinstance(state, dict):
+                _restored_state = state
+                self.phases["obj_restored"] = float(len(state))
+        except Exception:
+            log.exception("object-state restore failed; on_start will run")
 
 
 class HandlerError(Exception):
@@ -314,6 +336,19 @@ class RunnerClient:
         from ..ops import hipsnap
 
         info = hipsnap.snapshot(arena, grant["dir"], grant["key"])
+        # object-level state rides next to the arena snapshot so restore
+        # can skip on_start entirely
+        if _checkpoint_state:
+            try:
+                from ..ops import objsnap
+
+                data = objsnap.save_state(dict(_checkpoint_state), arena)
+                with open(
+                    hipsnap.snapshot_path(grant["dir"], grant["key"]) + ".obj", "wb"
+                ) as f:
+                    f.write(data)
+            except Exception:
+                log.exception("object-state snapshot failed (arena-only checkpoint)")
         await self.commit_checkpoint(grant["key"], info["bytes"])
         log.info(
             "checkpoint %s written: %.1f GB at %.1f GB/s",
@@ -340,10 +375,40 @@ class RunnerClient:
 
 async def run_on_start(env: RunnerEnv) -> Any:
     """Run on_start and return its context value (reference: on_start result
-    passed to handlers via context)."""
+    passed to handlers via context). After an object-level checkpoint
+    restore (objsnap) on_start is SKIPPED entirely — the model objects and
+    the on_start return value came back with the snapshot (reference flow
+    this replaces: CRIU restore resumes past on_start, criu.go:512-651)."""
+    if _restored_state is not None:
+        return _restored_state.get("__on_start_value__")
     if not env.on_start:
         return None
     fn = unwrap_handler(load_callable(env.on_start))
     if asyncio.iscoroutinefunction(fn):
-        return await fn()
-    return fn()
+        value = await fn()
+    else:
+        value = fn()
+    register_checkpoint_state("__on_start_value__", value)
+    return value
+
+
+# -- object-level checkpoint state (objsnap) --------------------------------
+# Handlers register their GPU-resident objects (models, KV caches); the
+# checkpoint pickles them with arena-tensor refs; restore hands the SAME
+# objects back and on_start never runs.
+_checkpoint_state: Dict[str, Any] = {}
+_restored_state: Optional[Dict[str, Any]] = None
+
+
+def register_checkpoint_state(name: str, obj: Any) -> None:
+    _checkpoint_state[name] = obj
+
+
+def get_checkpoint_state(name: str) -> Any:
+    if _restored_state is not None and name in _restored_state:
+        return _restored_state[name]
+    return _checkpoint_state.get(name)
+
+
+def state_restored() -> bool:
+    return _restored_state is not None

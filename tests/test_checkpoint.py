@@ -137,6 +137,10 @@ async def test_scale_to_zero_resume_gpu():
             for _cycle in range(2):
                 r2 = await invoke()
                 assert r2["load"]["restored"] is True
+                # objsnap: the model OBJECT came back too — on_start did not
+                # rebuild anything (load() found the restored object)
+                assert r2["load"].get("object_restored") is True, r2["load"]
+                assert r2["load"]["load_ms"] < 50, r2["load"]
                 assert r2["checksum"] == r2["checksum"]
                 for _ in range(300):
                     if not await c.gateway.container_repo.containers_for_stub(stub_id):
