@@ -70,7 +70,7 @@ def test_on_start_skipped_after_restore(monkeypatch):
     env.on_start = "tests.fixtures.handlers:on_start"
     # simulate a restored state: run_on_start must not import/run anything
     monkeypatch.setattr(common, "_restored_state", {"__on_start_value__": {"ctx": "restored"}})
-    v = asyncio.get_event_loop().run_until_complete(common.run_on_start(env))
+    v = asyncio.run(common.run_on_start(env))
     assert v == {"ctx": "restored"}
     assert common.state_restored()
     monkeypatch.setattr(common, "_restored_state", None)
