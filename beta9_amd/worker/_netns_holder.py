@@ -27,6 +27,12 @@ CLONE_NEWNET = 0x40000000
 
 def main() -> int:
     libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    # die with the worker (survives the exec into `sleep`): crashed workers
+    # must not leak namespace holders
+    PR_SET_PDEATHSIG = 1
+    libc.prctl(PR_SET_PDEATHSIG, 9, 0, 0, 0)
+    if os.getppid() == 1:
+        return 1  # worker already gone
     if libc.unshare(CLONE_NEWNET) != 0:
         err = ctypes.get_errno()
         print(f"ERR unshare: {os.strerror(err)}", flush=True)

