@@ -399,7 +399,10 @@ class Lifecycle:
         env.setdefault("STUB_ID", request.stub_id)
         env.setdefault("STUB_TYPE", request.stub_type)
         env.setdefault("WORKSPACE_ID", request.workspace_id)
-        env.setdefault("B9_GATEWAY_HOST", self.gateway_host)
+        # the gateway may bind 0.0.0.0 (multi-node); containers must dial
+        # the address THIS worker reaches it at
+        if env.get("B9_GATEWAY_HOST", "") in ("", "0.0.0.0", "::"):
+            env["B9_GATEWAY_HOST"] = self.gateway_host
         env.setdefault("B9_GATEWAY_PORT", str(self.gateway_grpc_port))
         env.setdefault("CONTAINER_HOSTNAME", request.hostname or request.container_id)
         env.setdefault("B9_LIGHT_IMPORT", "1")  # runners skip the SDK surface
