@@ -1254,12 +1254,13 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_gate_up_silu_256_kernel(
 
 typedef unsigned char fp8_t;
 
-template <int BNJ>
+template <int BNJ, bool Q8 = false>
 __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
     const fp8_t* __restrict__ A, const fp8_t* __restrict__ Wg,
     const fp8_t* __restrict__ Wu, bf16_t* __restrict__ Out,
     const float* __restrict__ g_scale, const float* __restrict__ u_scale,
-    float a_scale, int M, int N, int K) {
+    float a_scale, int M, int N, int K,
+    fp8_t* __restrict__ Out8 = nullptr, float inv_act2 = 1.0f) {
   constexpr int BNW = BNJ * 16;
   constexpr int BNT = BNW * 2;
   constexpr int BK8 = 128;  // fp8 elements per LDS row = 128 B (2 k-tiles)
@@ -1379,9 +1380,179 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
         int row = tile_m + wr * 64 + i * 16 + c_row0 + r;
         float g = acc_g[i][j][r] * gs;
         float u = acc_u[i][j][r] * us;
-        Out[(int64_t)row * N + col] = (bf16_t)(g / (1.0f + __expf(-g)) * u);
+        float v = g / (1.0f + __expf(-g)) * u;
+        if constexpr (Q8) {
+          // w8a8 chain: emit the NEXT projection's fp8 activation directly
+          // (static calibrated scale) — zero eager quant kernels between
+          // gate-up and down-proj
+          __hip_fp8_e4m3 q(v * inv_act2);
+          Out8[(int64_t)row * N + col] = q.__x;
+        } else {
+          Out[(int64_t)row * N + col] = (bf16_t)v;
+        }
       }
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FP8 w8a8 down-projection with split-K: partials = Aq[M,K] @ Wq[N,K]^T in
+// raw f32 (scales applied in the reduce, where the residual is fused too).
+// Completes the fp8 serving chain: rmsnorm_fp8 -> gate_up_fp8(Q8) ->
+// down_fp8+residual — zero bf16 GEMMs, zero eager quant kernels
+// (VERDICT round-1 item #8). Same LDS geometry as the gate-up fp8 kernel
+// (128-B rows, 16-B-chunk XOR swizzle, glds width 16).
+// ---------------------------------------------------------------------------
+
+template <int BNJ>
+__global__ __launch_bounds__(THREADS, 1) void gemm_down_fp8_kernel(
+    const fp8_t* __restrict__ A, const fp8_t* __restrict__ W,
+    float* __restrict__ Cpart, int M, int N, int K, int splitk) {
+  constexpr int BNW = BNJ * 16;
+  constexpr int BNT = BNW * 2;
+  constexpr int BK8 = 128;
+  __shared__ fp8_t lds[2 * (BM + BNT) * BK8];
+
+  int nwg_n = N / BNT;
+  int nwg_m = M / BM;
+  int wg = blockIdx.x;
+  int slice = wg / (nwg_m * nwg_n);
+  int t = wg % (nwg_m * nwg_n);
+  int tile_m = (t / nwg_n) * BM;
+  int tile_n = (t % nwg_n) * BNT;
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6;
+  int lane = tid & 63;
+  int wr = wave >> 1, wc = wave & 1;
+
+  int ld_row = lane >> 3;
+  int ld_chunk = lane & 7;
+  auto stage = [&](int buf, int kt) {
+    fp8_t* Ab = lds + buf * (BM + BNT) * BK8;
+    fp8_t* Wb = Ab + BM * BK8;
+    const fp8_t* a_src = A + (int64_t)(tile_m)*K + (int64_t)kt * BK8;
+    const fp8_t* w_src = W + (int64_t)(tile_n)*K + (int64_t)kt * BK8;
+#pragma unroll
+    for (int call = 0; call < BM / 8 / WAVES; call++) {
+      int p = wave + WAVES * call;
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(a_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 16)),
+          (__attribute__((address_space(3))) uint32_t*)(Ab + p * 1024), 16, 0, 0);
+    }
+#pragma unroll
+    for (int call = 0; call < BNT / 8 / WAVES; call++) {
+      int p = wave + WAVES * call;
+      int row = p * 8 + ld_row;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(w_src + (int64_t)row * K + ((ld_chunk ^ (row & 7)) * 16)),
+          (__attribute__((address_space(3))) uint32_t*)(Wb + p * 1024), 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][BNJ] = {};
+
+  int ntiles_all = K / BK8;
+  int per = ntiles_all / splitk;
+  int kt0 = slice * per;
+  int kt_end = (slice == splitk - 1) ? ntiles_all : kt0 + per;
+
+  stage(0, kt0);
+
+  int fr_row = lane & 15;
+  int fr_q = lane >> 4;
+  auto off_of = [&](int r, int c8) -> int {
+    return r * BK8 + ((((c8 >> 1) ^ (r & 7)) << 4)) + ((c8 & 1) << 3);
+  };
+  auto rd = [&](const fp8_t* img, int off) -> long {
+    return *(const long*)(img + off);
+  };
+
+  constexpr int GLDS = BM / 8 / WAVES + BNT / 8 / WAVES;
+  for (int kt = kt0; kt < kt_end; kt++) {
+    int buf = (kt - kt0) & 1;
+    fp8_t* Ab = lds + buf * (BM + BNT) * BK8;
+    fp8_t* Wb = Ab + BM * BK8;
+    if (kt + 1 < kt_end) {
+      stage(buf ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(GLDS) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      long a_frag[4], w_frag[BNJ];
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+        a_frag[i] = rd(Ab, off_of(wr * 64 + i * 16 + fr_row, ks * 4 + fr_q));
+#pragma unroll
+      for (int j = 0; j < BNJ; j++)
+        w_frag[j] = rd(Wb, off_of(wc * BNW + j * 16 + fr_row, ks * 4 + fr_q));
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < BNJ; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a_frag[i], w_frag[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  float* out = Cpart + (int64_t)slice * M * N;
+  int c_col = lane & 15;
+  int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < BNJ; j++) {
+      int col = tile_n + wc * BNW + j * 16 + c_col;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = tile_m + wr * 64 + i * 16 + c_row0 + r;
+        out[(int64_t)row * N + col] = acc[i][j][r];
+      }
+    }
+  }
+}
+
+// split-K reduce for the fp8 path: out = R + sum(partials) * a_scale *
+// w_scale[col] — the dequant scales fold into the reduce pass for free.
+__global__ __launch_bounds__(256) void splitk_reduce_scale_kernel(
+    const float* __restrict__ Cpart, bf16_t* __restrict__ C, int64_t mn,
+    int N, int splitk, float a_scale, const float* __restrict__ w_scale,
+    const bf16_t* __restrict__ R) {
+  int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  typedef float f32x4v __attribute__((ext_vector_type(4)));
+  for (; idx + 8 <= mn; idx += stride) {
+    float sum[8] = {};
+#pragma unroll 4
+    for (int s = 0; s < splitk; s++) {
+      const f32x4v* p = (const f32x4v*)(Cpart + (int64_t)s * mn + idx);
+      f32x4v lo = p[0], hi = p[1];
+#pragma unroll
+      for (int k = 0; k < 4; k++) {
+        sum[k] += lo[k];
+        sum[4 + k] += hi[k];
+      }
+    }
+    int col0 = (int)(idx % N);  // rows are N-aligned (idx%8==0, N%8==0)
+    bf16x8 o;
+    bf16x8 rv;
+    if (R != nullptr) rv = *(const bf16x8*)(R + idx);
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      float v = sum[k] * a_scale * w_scale[col0 + k];
+      if (R != nullptr) v += (float)((bf16_t*)&rv)[k];
+      ((bf16_t*)&o)[k] = (bf16_t)v;
+    }
+    *(bf16x8*)(C + idx) = o;
   }
 }
 
@@ -1409,7 +1580,7 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
                        (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
                        (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
                        (const float*)u_scale.data_ptr(), (float)a_scale,
-                       (int)M, (int)N, (int)K);
+                       (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
   } else {
     dim3 grid((M / BM) * (N / 64));
     hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<2>), grid, dim3(THREADS),
@@ -1417,8 +1588,96 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
                        (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
                        (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
                        (const float*)u_scale.data_ptr(), (float)a_scale,
-                       (int)M, (int)N, (int)K);
+                       (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
   }
+  return Out;
+}
+
+torch::Tensor gemm_gate_up_silu_fp8_q8(torch::Tensor A, torch::Tensor Wg,
+                                       torch::Tensor Wu, torch::Tensor g_scale,
+                                       torch::Tensor u_scale, double a_scale,
+                                       double act2_scale) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.dim() == 2);
+  TORCH_CHECK(Wg.dtype() == torch::kUInt8 && Wu.dtype() == torch::kUInt8);
+  int64_t M = A.size(0), K = A.size(1), N = Wg.size(0);
+  TORCH_CHECK(Wg.size(1) == K && Wu.size(1) == K && g_scale.size(0) == N);
+  TORCH_CHECK(u_scale.size(0) == N, "u_scale length must equal N");
+  TORCH_CHECK(A.is_contiguous() && Wg.is_contiguous() && Wu.is_contiguous());
+  TORCH_CHECK(M % BM == 0 && K % 128 == 0 && N % 64 == 0);
+  TORCH_CHECK(act2_scale > 0.0);
+  auto Out8 = torch::empty({M, N}, A.options().dtype(torch::kUInt8));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  float inv = (float)(1.0 / act2_scale);
+  int64_t tiles128 = (M / BM) * (N / 128);
+  if (N % 128 == 0 && tiles128 >= 256) {
+    dim3 grid(tiles128);
+    hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<4, true>), grid,
+                       dim3(THREADS), 0, stream, (const fp8_t*)A.data_ptr(),
+                       (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                       (bf16_t*)nullptr, (const float*)g_scale.data_ptr(),
+                       (const float*)u_scale.data_ptr(), (float)a_scale,
+                       (int)M, (int)N, (int)K, (fp8_t*)Out8.data_ptr(), inv);
+  } else {
+    dim3 grid((M / BM) * (N / 64));
+    hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<2, true>), grid,
+                       dim3(THREADS), 0, stream, (const fp8_t*)A.data_ptr(),
+                       (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                       (bf16_t*)nullptr, (const float*)g_scale.data_ptr(),
+                       (const float*)u_scale.data_ptr(), (float)a_scale,
+                       (int)M, (int)N, (int)K, (fp8_t*)Out8.data_ptr(), inv);
+  }
+  return Out8;
+}
+
+torch::Tensor gemm_down_fp8(torch::Tensor A, torch::Tensor W,
+                            torch::Tensor w_scale, double a_scale,
+                            torch::Tensor R, int64_t splitk) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.dim() == 2);
+  TORCH_CHECK(W.is_cuda() && W.dtype() == torch::kUInt8);
+  TORCH_CHECK(w_scale.dtype() == torch::kFloat32 && w_scale.is_cuda());
+  int64_t M = A.size(0), K = A.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K && w_scale.size(0) == N);
+  TORCH_CHECK(A.is_contiguous() && W.is_contiguous() && w_scale.is_contiguous());
+  TORCH_CHECK(M % BM == 0 && K % 128 == 0 && N % 64 == 0,
+              "fp8 down needs M%128, K%128, N%64 == 0");
+  bool has_r = R.defined() && R.numel() > 0;
+  if (has_r) {
+    TORCH_CHECK(R.is_cuda() && R.dtype() == torch::kBFloat16 &&
+                R.is_contiguous() && R.size(0) == M && R.size(1) == N);
+  }
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  // pick split so the grid covers the 256 CUs (~2 blocks/CU)
+  bool wide = (N % 128 == 0);
+  int64_t bnt = wide ? 128 : 64;
+  int64_t base_tiles = (M / BM) * (N / bnt);
+  if (splitk <= 0) {
+    splitk = 1;
+    while (base_tiles * splitk < 512 && splitk < 16 &&
+           (K / 128) % (splitk * 2) == 0)
+      splitk *= 2;
+  }
+  auto Cpart = torch::empty({splitk, M, N}, A.options().dtype(torch::kFloat32));
+  dim3 grid(base_tiles * splitk);
+  if (wide) {
+    hipLaunchKernelGGL((gemm_down_fp8_kernel<4>), grid, dim3(THREADS), 0,
+                       stream, (const fp8_t*)A.data_ptr(),
+                       (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                       (int)M, (int)N, (int)K, (int)splitk);
+  } else {
+    hipLaunchKernelGGL((gemm_down_fp8_kernel<2>), grid, dim3(THREADS), 0,
+                       stream, (const fp8_t*)A.data_ptr(),
+                       (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                       (int)M, (int)N, (int)K, (int)splitk);
+  }
+  auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
+  int64_t mn = M * N;
+  TORCH_CHECK(mn % 8 == 0 && N % 8 == 0);
+  int blocks = (int)std::min<int64_t>(2048, (mn / 8 + 255) / 256);
+  hipLaunchKernelGGL(splitk_reduce_scale_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (const float*)Cpart.data_ptr(),
+                     (bf16_t*)Out.data_ptr(), mn, (int)N, (int)splitk,
+                     (float)a_scale, (const float*)w_scale.data_ptr(),
+                     has_r ? (const bf16_t*)R.data_ptr() : (const bf16_t*)nullptr);
   return Out;
 }
 
@@ -1510,6 +1769,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("xcd_swizzle") = true, py::arg("splitk") = -1);
   m.def("rmsnorm_fp8", &rmsnorm_fp8, py::arg("x"), py::arg("w"),
         py::arg("eps") = 1e-6, py::arg("act_scale") = 1.0);
+  m.def("gemm_gate_up_silu_fp8_q8", &gemm_gate_up_silu_fp8_q8, py::arg("A"),
+        py::arg("Wg"), py::arg("Wu"), py::arg("g_scale"), py::arg("u_scale"),
+        py::arg("a_scale"), py::arg("act2_scale"));
+  m.def("gemm_down_fp8", &gemm_down_fp8, py::arg("A"), py::arg("W"),
+        py::arg("w_scale"), py::arg("a_scale"), py::arg("R"),
+        py::arg("splitk") = -1);
   m.def("gemm_gate_up_silu_fp8", &gemm_gate_up_silu_fp8, py::arg("A"),
         py::arg("Wg"), py::arg("Wu"), py::arg("g_scale"), py::arg("u_scale"),
         py::arg("a_scale"));

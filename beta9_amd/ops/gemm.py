@@ -39,12 +39,18 @@ def has_native() -> bool:
 def gemm_nt(a: torch.Tensor, w: torch.Tensor, xcd_swizzle: bool = False) -> torch.Tensor:
     """C = a @ w.T — a [M,K] bf16, w [N,K] bf16 (torch Linear weight layout).
 
-    Hand-written MFMA path for aligned shapes (M%128, N%128, K%64 == 0);
-    torch fallback only on CPU."""
+    Best-kernel-per-shape dispatch (VERDICT round-1 item #7): serving
+    shapes (small M) run the hand-written MFMA kernels (they beat the
+    library there — profiles/KERNELS.md); large plain square-ish GEMMs go
+    to hipBLASLt via torch.matmul, which is the intended division of
+    labor (fused hot ops hand-written, plain library GEMMs to the
+    library: hipBLASLt 1587 vs ours 1180 TF @8192^3)."""
     ext = _load()
     if ext is not None and a.is_cuda:
         M, K = a.shape
         N = w.shape[0]
+        if M >= 1024 and N >= 2048:
+            return a @ w.t()  # hipBLASLt territory
         if M % 128 == 0 and N % 128 == 0 and K % 64 == 0:
             return ext.gemm_bf16_nt(a.contiguous(), w.contiguous(), xcd_swizzle)
         # pad M up to 128 (weights stay aligned by construction)
@@ -146,6 +152,8 @@ def warm_all(device: str = "cuda:0") -> bool:
     q = torch.ones(128, 128, dtype=torch.uint8, device=device)
     sc = torch.ones(128, dtype=torch.float32, device=device)
     ext.gemm_gate_up_silu_fp8(q, q, q, sc, sc, 1.0)
+    ext.gemm_gate_up_silu_fp8_q8(q, q, q, sc, sc, 1.0, 1.0)
+    ext.gemm_down_fp8(q, q, sc, 1.0, torch.empty(0, device=device), -1)
     ext.rmsnorm_fp8(a, torch.ones(64, dtype=torch.bfloat16, device=device), 1e-6, 1.0)
     # torch EAGER kernels the serving path touches (residual add, checksum
     # cast/abs/reduce, softmax, cat): each costs 10-40 ms at first launch on
@@ -207,6 +215,37 @@ def mlp_gate_up_fp8(a_q, w_gate_q, w_up_q, g_scale, u_scale, a_scale: float):
     return ext.gemm_gate_up_silu_fp8(
         a_q.contiguous(), w_gate_q.contiguous(), w_up_q.contiguous(),
         g_scale.contiguous(), u_scale.contiguous(), a_scale,
+    )
+
+
+def mlp_gate_up_fp8_q8(a_q, w_gate_q, w_up_q, g_scale, u_scale,
+                       a_scale: float, act2_scale: float):
+    """FP8 gate-up with the NEXT projection's activation quantization fused
+    into the epilogue: emits uint8 e4m3 of (silu(g)*u / act2_scale). With
+    rmsnorm_fp8_out and gemm_down_fp8 this makes the whole serving block
+    w8a8 with zero eager quant kernels and zero bf16 GEMMs (VERDICT
+    round-1 item #8)."""
+    ext = _load()
+    if ext is None or not a_q.is_cuda:
+        raise RuntimeError("fp8 gate-up q8 requires the native extension + GPU")
+    return ext.gemm_gate_up_silu_fp8_q8(
+        a_q.contiguous(), w_gate_q.contiguous(), w_up_q.contiguous(),
+        g_scale.contiguous(), u_scale.contiguous(), a_scale, act2_scale,
+    )
+
+
+def mlp_down_fp8(a_q, w_down_q, w_scale, a_scale: float,
+                 residual: Optional[torch.Tensor] = None):
+    """FP8 w8a8 down-projection with split-K and the residual + dequant
+    scales fused into the reduce: out = residual + (a_q @ w_q.T) * a_scale
+    * w_scale[col], bf16 out."""
+    ext = _load()
+    if ext is None or not a_q.is_cuda:
+        raise RuntimeError("fp8 down requires the native extension + GPU")
+    r = residual if residual is not None else torch.empty(0)
+    return ext.gemm_down_fp8(
+        a_q.contiguous(), w_down_q.contiguous(), w_scale.contiguous(),
+        a_scale, r.contiguous() if r.numel() else r, -1,
     )
 
 

@@ -284,3 +284,84 @@ def test_rmsnorm_fp8_fused_quant():
     # and close to the reference quantization (within ~one quant step rel.)
     rel2 = ((out - ref).abs().mean() / (ref.abs().mean() + 1e-6)).item()
     assert rel2 < 0.08, rel2
+
+
+@pytest.mark.gpu
+def test_down_fp8_numerics():
+    """FP8 w8a8 split-K down-projection + fused residual/dequant vs the
+    dequantized fp32 reference (VERDICT round-1 item #8)."""
+    import torch
+
+    from beta9_amd.ops import gemm as g
+
+    torch.manual_seed(5)
+    M, K, N = 256, 1024, 512  # K%128==0, N%64==0
+    a = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(N, K, device="cuda") * 0.5).to(torch.bfloat16)
+    r = torch.randn(M, N, device="cuda").to(torch.bfloat16)
+    aq, a_s = g.quantize_fp8_tensor(a)
+    wq, w_s = g.quantize_fp8_rowwise(w)
+    out = g.mlp_down_fp8(aq, wq, w_s, a_s, r)
+    adq = aq.view(torch.float8_e4m3fn).float() * a_s
+    wdq = wq.view(torch.float8_e4m3fn).float() * w_s[:, None]
+    ref = r.float() + adq @ wdq.t()
+    err = ((out.float() - ref).abs().max() / ref.abs().max()).item()
+    assert err < 0.02, err
+    # no-residual path
+    out2 = g.mlp_down_fp8(aq, wq, w_s, a_s, None)
+    ref2 = adq @ wdq.t()
+    err2 = ((out2.float() - ref2).abs().max() / ref2.abs().max()).item()
+    assert err2 < 0.02, err2
+
+
+@pytest.mark.gpu
+def test_gate_up_fp8_q8_epilogue():
+    """Q8 epilogue emits e4m3 of silu(g)*u/act2 matching the bf16 kernel's
+    output after quantization."""
+    import torch
+
+    from beta9_amd.ops import gemm as g
+
+    torch.manual_seed(7)
+    a = (torch.randn(128, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    wg = (torch.randn(512, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    wu = (torch.randn(512, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    aq, a_s = g.quantize_fp8_tensor(a)
+    gq, g_s = g.quantize_fp8_rowwise(wg)
+    uq, u_s = g.quantize_fp8_rowwise(wu)
+    bf = g.mlp_gate_up_fp8(aq, gq, uq, g_s, u_s, a_s)
+    act2 = (bf.float().abs().amax() * 1.1 / g.E4M3_MAX).item()
+    q8 = g.mlp_gate_up_fp8_q8(aq, gq, uq, g_s, u_s, a_s, act2)
+    deq = q8.view(torch.float8_e4m3fn).float() * act2
+    err = ((deq - bf.float()).abs().max() / bf.float().abs().max()).item()
+    assert err < 0.05, err  # one extra e4m3 rounding
+
+
+@pytest.mark.gpu
+def test_fp8_backend_steady_state_full_chain():
+    """Second fp8 forward call runs the FULL w8a8 chain (gate-up Q8 ->
+    fp8 down + fused residual) and stays at quantization-level error."""
+    import os
+
+    import torch
+
+    from beta9_amd.models.mlp import SyntheticServingModel
+
+    m = SyntheticServingModel(512, 1024, 2, "cuda:0", use_arena=False)
+    x = torch.randn(64, 512, dtype=torch.bfloat16, device="cuda:0")
+    old = os.environ.get("B9_GEMM_BACKEND")
+    try:
+        os.environ["B9_GEMM_BACKEND"] = "native"
+        ref = m.forward(x).float()
+        os.environ["B9_GEMM_BACKEND"] = "fp8"
+        m.forward(x)  # calibration call (bf16 down)
+        for blk in m.blocks:
+            assert blk._act2_scale is not None
+        out = m.forward(x).float()  # steady state: full fp8 chain
+    finally:
+        if old is None:
+            os.environ.pop("B9_GEMM_BACKEND", None)
+        else:
+            os.environ["B9_GEMM_BACKEND"] = old
+    rel = ((out - ref).abs().mean() / ref.abs().mean()).item()
+    assert rel < 0.12, rel
