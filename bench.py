@@ -70,6 +70,7 @@ def stub_cfg(args, has_gpu: bool, name_env: dict | None = None):
             "B9_BENCH_INTER": str(args.inter),
             "B9_BENCH_LAYERS": str(args.layers),
             "B9_BENCH_MODEL": args.model,
+            **({"B9_GEMM_BACKEND": args.backend} if getattr(args, "backend", "") else {}),
             **(name_env or {}),
         },
         autoscaler=AutoscalerConfig(
@@ -396,6 +397,7 @@ def main() -> int:
     p.add_argument("--inter", type=int, default=14336)
     p.add_argument("--layers", type=int, default=8)
     p.add_argument("--model", choices=["mlp", "transformer"], default="mlp")
+    p.add_argument("--backend", default="", help="B9_GEMM_BACKEND for the serving container (e.g. fp8)")
     p.add_argument("--cold-iters", type=int, default=10,
                    help="sequential prewarm-hit cold starts to sample")
     p.add_argument("--cold-miss-iters", type=int, default=10,
