@@ -78,9 +78,15 @@ def test_asan_volcache_interposer(tmp_path):
     real = tmp_path / "real.txt"
     real.write_text("cached-bytes")
     vmap = f"{tmp_path}/virtual.txt={real}"
+    # the asan runtime must be first in the preload chain when injecting
+    # an asan-built .so into a non-asan binary
+    libasan = subprocess.run(
+        ["gcc", "-print-file-name=libasan.so"], capture_output=True, text=True
+    ).stdout.strip()
     r = subprocess.run(
         ["cat", str(tmp_path / "virtual.txt")],
-        env={**ENV, "LD_PRELOAD": ASAN_VOLCACHE, "VOLUME_CACHE_MAP": vmap},
+        env={**ENV, "LD_PRELOAD": f"{libasan}:{ASAN_VOLCACHE}",
+             "VOLUME_CACHE_MAP": vmap},
         capture_output=True, text=True, timeout=30,
     )
     assert "AddressSanitizer" not in r.stderr, r.stderr[-2000:]
