@@ -283,6 +283,27 @@ def workspace_create(name: str):
     typer.echo(json.dumps(r))
 
 
+image_app = typer.Typer(help="container images")
+app.add_typer(image_app, name="image")
+
+
+@image_app.command("import")
+def image_import(path: str, image_id: str = typer.Option("", help="pin a specific id")):
+    """Import an OCI image layout dir / docker-save tar into the layer
+    store (the offline `from_registry`). Admin token required."""
+    payload = {"path": path}
+    if image_id:
+        payload["image_id"] = image_id
+    r = _client().call("ImageService", "ImportImage", payload)
+    typer.echo(json.dumps(r))
+
+
+@image_app.command("get")
+def image_get(image_id: str):
+    r = _client().call("ImageService", "GetImage", {"image_id": image_id})
+    typer.echo(json.dumps(r))
+
+
 # -- secrets --------------------------------------------------------------------
 
 
