@@ -224,9 +224,26 @@ class PodService:
 
             self.gw.tcp_ingress = TcpIngress()
         port = await self.gw.tcp_ingress.expose(cid, host, int(req["port"]))
-        return {
+        out = {
             "url": f"http://{self.gw.cfg.gateway.host}:{port}",
             "host": self.gw.cfg.gateway.host,
             "port": port,
             "target": f"{host}:{req['port']}",
         }
+        if req.get("tls"):
+            # SNI TLS ingress (reference: pod tcp.go TLS listener with
+            # per-stub SNI certs): one shared TLS port, routed by hostname
+            # {container_id}.tcp.{base_domain}, self-signed per-host cert
+            if getattr(self.gw, "sni_ingress", None) is None:
+                from ..gateway.tcp_ingress import SniTlsIngress
+
+                self.gw.sni_ingress = SniTlsIngress(
+                    base_domain=self.gw.cfg.gateway.base_domain
+                )
+                await self.gw.sni_ingress.start()
+            self.gw.sni_ingress.add_route(cid, host, int(req["port"]))
+            sni_host = f"{cid}.tcp.{self.gw.cfg.gateway.base_domain}"
+            out["tls_port"] = self.gw.sni_ingress.port
+            out["tls_sni"] = sni_host
+            out["tls_url"] = f"https://{sni_host}:{self.gw.sni_ingress.port}"
+        return out

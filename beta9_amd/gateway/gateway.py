@@ -365,6 +365,8 @@ class Gateway:
                 pass
         if getattr(self, "tcp_ingress", None) is not None:
             await self.tcp_ingress.shutdown()
+        if getattr(self, "sni_ingress", None) is not None:
+            await self.sni_ingress.stop()
         if getattr(self, "cache_transport", None) is not None:
             await self.cache_transport.stop()
         await self.rpc_server.stop()
