@@ -27,8 +27,11 @@ else:
         PodInstance,
         QueueClient,
         QueueDepthAutoscaler,
+        AsyncSandbox,
+        AsyncSandboxInstance,
         Sandbox,
         SandboxInstance,
+        SandboxTerminal,
         SignalClient,
         TaskQueue,
         Volume,

@@ -121,6 +121,27 @@ class PodService:
     async def SandboxExec(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
         return await self._pm_call(req["container_id"], "Exec", req)
 
+    # -- interactive terminal (PTY in the PM; the dropbear/shell analog) ----
+    @rpc()
+    async def SandboxTerminalOpen(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "TerminalOpen", req)
+
+    @rpc()
+    async def SandboxTerminalRead(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "TerminalRead", req)
+
+    @rpc()
+    async def SandboxTerminalWrite(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "TerminalWrite", req)
+
+    @rpc()
+    async def SandboxTerminalResize(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "TerminalResize", req)
+
+    @rpc()
+    async def SandboxTerminalClose(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        return await self._pm_call(req["container_id"], "TerminalClose", req)
+
     @rpc()
     async def SandboxRunCode(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
         return await self._pm_call(req["container_id"], "RunCode", req)

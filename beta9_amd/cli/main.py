@@ -373,11 +373,77 @@ def worker_drain(worker_id: str, timeout: float = typer.Option(120.0), remove: b
     typer.echo("ok")
 
 
+def _pty_shell(container_id: str) -> None:
+    """Raw-mode terminal bridged to the container PM's pty over gRPC."""
+    import select
+    import shutil
+    import termios
+    import threading
+    import tty
+
+    client = _client()
+    size = shutil.get_terminal_size()
+    r = client.call(
+        "PodService", "SandboxTerminalOpen",
+        {"container_id": container_id, "rows": size.lines, "cols": size.columns},
+    )
+    term_id = r["term_id"]
+    typer.echo(f"[b9 shell] pty attached to {container_id} (exit the shell to detach)")
+    stop = threading.Event()
+
+    def reader():
+        while not stop.is_set():
+            try:
+                rr = client.call(
+                    "PodService", "SandboxTerminalRead",
+                    {"container_id": container_id, "term_id": term_id, "timeout": 5.0},
+                    timeout=20.0,
+                )
+            except Exception:
+                break
+            data = rr.get("data") or b""
+            if data:
+                os.write(1, data if isinstance(data, bytes) else data.encode())
+            if rr.get("closed"):
+                break
+        stop.set()
+
+    old = termios.tcgetattr(0)
+    t = threading.Thread(target=reader, daemon=True)
+    try:
+        tty.setraw(0)
+        t.start()
+        while not stop.is_set():
+            ready, _, _ = select.select([0], [], [], 0.1)
+            if not ready:
+                continue
+            data = os.read(0, 4096)
+            if not data:
+                break
+            client.call(
+                "PodService", "SandboxTerminalWrite",
+                {"container_id": container_id, "term_id": term_id, "data": data},
+            )
+    finally:
+        stop.set()
+        termios.tcsetattr(0, termios.TCSADRAIN, old)
+        try:
+            client.call(
+                "PodService", "SandboxTerminalClose",
+                {"container_id": container_id, "term_id": term_id},
+            )
+        except Exception:
+            pass
+        typer.echo("\n[b9 shell] detached")
+
+
 @app.command()
-def shell(container_id: str):
+def shell(container_id: str, pty: bool = typer.Option(True, help="full PTY terminal (raw mode); --no-pty for the line shell")):
     """Interactive shell into a sandbox/pod container (reference: beta9
-    shell via in-container dropbear; here: the container's process-manager
-    Exec surface, line-oriented)."""
+    shell via in-container dropbear; here: a PTY owned by the container's
+    process manager — full terminal semantics, no SSH daemon)."""
+    if pty and sys.stdin.isatty():
+        return _pty_shell(container_id)
     client = _client()
     typer.echo(f"connected to {container_id} (line shell; 'exit' to quit)")
     cwd = "/"

@@ -37,12 +37,47 @@ class _Process:
         self.started_at = time.time()
 
 
+class _Terminal:
+    """A PTY-backed shell session (the dropbear/SSH terminal-semantics
+    equivalent — reference: pkg/abstractions/shell + in-container dropbear;
+    here the PM owns the pty directly, no SSH daemon needed)."""
+
+    def __init__(self, term_id: int, pid: int, master_fd: int):
+        self.term_id = term_id
+        self.pid = pid
+        self.master_fd = master_fd
+        self.buf = b""
+        self.closed = False
+
+    def pump(self) -> None:
+        """Drain available pty output into the buffer (non-blocking)."""
+        import errno
+
+        while True:
+            try:
+                data = os.read(self.master_fd, 65536)
+            except BlockingIOError:
+                return
+            except OSError as e:
+                if e.errno in (errno.EIO, errno.EBADF):  # shell exited
+                    self.closed = True
+                return
+            if not data:
+                self.closed = True
+                return
+            self.buf += data
+            if len(self.buf) > (4 << 20):
+                self.buf = self.buf[-(4 << 20):]
+
+
 class ProcessManager:
     service_name = "ProcessManager"
 
     def __init__(self, root: str = "/"):
         self.root = root
         self.procs: Dict[int, _Process] = {}
+        self.terms: Dict[int, _Terminal] = {}
+        self._next_term = 0
 
     def _path(self, p: str) -> str:
         if not os.path.isabs(p):
@@ -101,6 +136,95 @@ class ProcessManager:
             "stdout": entry.stdout.decode(errors="replace"),
             "stderr": entry.stderr.decode(errors="replace"),
         }
+
+    # -- interactive terminal (PTY) -----------------------------------------
+    @rpc()
+    async def TerminalOpen(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        import fcntl
+        import pty
+        import shutil
+        import struct
+        import termios
+
+        shell = req.get("command") or shutil.which("bash") or "/bin/sh"
+        rows = int(req.get("rows", 24))
+        cols = int(req.get("cols", 80))
+        pid, master_fd = pty.fork()
+        if pid == 0:  # the shell
+            env = dict(os.environ)
+            env["TERM"] = req.get("term", "xterm-256color")
+            try:
+                os.chdir(req.get("cwd") or os.getcwd())
+            except OSError:
+                pass
+            os.execvpe(shell, [shell], env)
+            os._exit(127)
+        os.set_blocking(master_fd, False)
+        fcntl.ioctl(master_fd, termios.TIOCSWINSZ, struct.pack("HHHH", rows, cols, 0, 0))
+        self._next_term += 1
+        t = _Terminal(self._next_term, pid, master_fd)
+        self.terms[t.term_id] = t
+        return {"term_id": t.term_id, "pid": pid}
+
+    def _term(self, req) -> _Terminal:
+        t = self.terms.get(int(req.get("term_id", 0)))
+        if t is None:
+            raise invalid(f"no terminal {req.get('term_id')}")
+        return t
+
+    @rpc()
+    async def TerminalRead(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Long-poll read: returns buffered pty output (waits up to
+        `timeout` for the first byte)."""
+        t = self._term(req)
+        deadline = time.time() + float(req.get("timeout", 10.0))
+        while True:
+            t.pump()
+            if t.buf or t.closed or time.time() >= deadline:
+                data, t.buf = t.buf, b""
+                return {"data": data, "closed": t.closed}
+            await asyncio.sleep(0.02)
+
+    @rpc()
+    async def TerminalWrite(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        t = self._term(req)
+        data = req.get("data", b"")
+        if isinstance(data, str):
+            data = data.encode()
+        try:
+            os.write(t.master_fd, data)
+        except OSError:
+            t.closed = True
+        return {"ok": not t.closed}
+
+    @rpc()
+    async def TerminalResize(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        import fcntl
+        import struct
+        import termios
+
+        t = self._term(req)
+        fcntl.ioctl(
+            t.master_fd, termios.TIOCSWINSZ,
+            struct.pack("HHHH", int(req.get("rows", 24)), int(req.get("cols", 80)), 0, 0),
+        )
+        return {"ok": True}
+
+    @rpc()
+    async def TerminalClose(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        t = self.terms.pop(int(req.get("term_id", 0)), None)
+        if t is not None:
+            import signal as _signal
+
+            try:
+                os.kill(t.pid, _signal.SIGHUP)
+            except ProcessLookupError:
+                pass
+            try:
+                os.close(t.master_fd)
+            except OSError:
+                pass
+        return {"ok": True}
 
     @rpc()
     async def RunCode(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:

@@ -786,8 +786,108 @@ class SandboxInstance:
     def snapshot_memory(self) -> Dict:
         return self._call("SandboxSnapshotMemory", {})
 
+    def terminal(self, rows: int = 24, cols: int = 80) -> "SandboxTerminal":
+        return SandboxTerminal(self, rows=rows, cols=cols)
+
     def terminate(self) -> None:
         self.client.call("PodService", "StopPod", {"container_id": self.container_id})
+
+
+class SandboxTerminal:
+    """Interactive PTY session in a sandbox (reference: shell-over-dropbear;
+    here the PM owns the pty — `b9 shell` uses the same RPCs)."""
+
+    def __init__(self, inst: "SandboxInstance", rows: int = 24, cols: int = 80):
+        self.inst = inst
+        r = inst._call("SandboxTerminalOpen", {"rows": rows, "cols": cols})
+        self.term_id = r["term_id"]
+
+    def write(self, data) -> None:
+        self.inst._call("SandboxTerminalWrite", {"term_id": self.term_id, "data": data})
+
+    def read(self, timeout: float = 5.0):
+        r = self.inst._call("SandboxTerminalRead", {"term_id": self.term_id, "timeout": timeout})
+        return r.get("data") or b"", bool(r.get("closed"))
+
+    def resize(self, rows: int, cols: int) -> None:
+        self.inst._call("SandboxTerminalResize", {"term_id": self.term_id, "rows": rows, "cols": cols})
+
+    def close(self) -> None:
+        self.inst._call("SandboxTerminalClose", {"term_id": self.term_id})
+
+
+class AsyncSandboxInstance:
+    """Asyncio variant of SandboxInstance (reference: the SDK's sandbox
+    async APIs, sdk sandbox.py sync+asyncio surfaces). Wraps the same
+    PodService RPCs on the caller's event loop."""
+
+    def __init__(self, config: Optional["ClientConfig"], container_id: str, addr: str):
+        from .rpc.channel import RpcClient
+
+        cfg = config or get_config()
+        self._rpc = RpcClient(cfg.gateway_addr, token=cfg.token)
+        self.container_id = container_id
+        self.addr = addr
+
+    async def _call(self, method: str, payload: Dict[str, Any]) -> Any:
+        payload = {"container_id": self.container_id, **payload}
+        return await self._rpc.call(
+            "PodService", method, payload,
+            timeout=float(payload.get("timeout", 120.0)) + 15.0,
+        )
+
+    async def exec(self, command, timeout: float = 120.0, background: bool = False,
+                   cwd: str = "", env: Optional[Dict] = None) -> Dict:
+        return await self._call("SandboxExec", {
+            "command": command, "timeout": timeout, "background": background,
+            "cwd": cwd, "env": env,
+        })
+
+    async def run_code(self, code: str, timeout: float = 120.0) -> Dict:
+        return await self._call("SandboxRunCode", {"code": code, "timeout": timeout})
+
+    async def status(self) -> Dict:
+        return await self._call("SandboxStatus", {})
+
+    async def upload(self, path: str, data: bytes) -> Dict:
+        return await self._call("SandboxUploadFile", {"path": path, "data": data})
+
+    async def download(self, path: str) -> bytes:
+        return (await self._call("SandboxDownloadFile", {"path": path}))["data"]
+
+    async def ls(self, path: str = ".") -> list:
+        return await self._call("SandboxListFiles", {"path": path})
+
+    async def expose_port(self, port: int, tls: bool = False) -> Dict:
+        return await self._call("SandboxExposePort", {"port": port, "tls": tls})
+
+    async def terminate(self) -> None:
+        await self._rpc.call("PodService", "StopPod", {"container_id": self.container_id})
+
+    async def close(self) -> None:
+        await self._rpc.close()
+
+
+class AsyncSandbox:
+    """`await AsyncSandbox(...).create()` — async twin of Sandbox."""
+
+    def __init__(self, **kwargs):
+        self._sync = Sandbox(**kwargs)
+
+    async def create(self, timeout: float = 60.0) -> AsyncSandboxInstance:
+        import asyncio as _a
+
+        def _mk():
+            stub_id = self._sync.prepare_runtime(stub_type="sandbox", force=True)
+            return self._sync.client.call(
+                "PodService", "CreatePod", {"stub_id": stub_id, "timeout": timeout},
+                timeout=timeout + 10.0,
+            )
+
+        r = await _a.to_thread(_mk)
+        if not r.get("ok"):
+            raise RuntimeError("sandbox did not become ready")
+        return AsyncSandboxInstance(self._sync.client.config, r["container_id"], r["addr"])
 
 
 class SandboxFS:

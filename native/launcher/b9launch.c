@@ -219,6 +219,10 @@ static int child_main(void *arg) {
     snprintf(path, sizeof path, "%s/dev/pts", merged);
     mkdir(path, 0755);
     mount("devpts", path, "devpts", 0, "newinstance,ptmxmode=0666");
+    /* ptys (sandbox terminals) open /dev/ptmx */
+    snprintf(path, sizeof path, "%s/dev/ptmx", merged);
+    if (symlink("pts/ptmx", path) != 0 && errno != EEXIST)
+      fprintf(stderr, "b9launch: ptmx symlink: %s\n", strerror(errno));
     snprintf(path, sizeof path, "%s/tmp", merged);
     mkdir(path, 01777);
     mount("tmpfs", path, "tmpfs", 0, "mode=1777");

@@ -96,3 +96,62 @@ async def test_gpu_function_in_native_runtime():
             assert r["vol"] is True
         finally:
             await w.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_sandbox_pty_terminal():
+    """PTY terminal in the sandbox PM: real terminal semantics (prompt,
+    echo, interactive state) — the dropbear/shell parity path."""
+    import beta9_amd
+
+    from .helpers import cluster
+
+    async with cluster(n_workers=1) as c:
+        sb = await asyncio.to_thread(beta9_amd.Sandbox(cpu=100, memory=128).create)
+        try:
+            term = await asyncio.to_thread(sb.terminal, 24, 80)
+            await asyncio.to_thread(term.write, "echo T$((40+2))T\n")
+            out = b""
+            for _ in range(20):
+                data, closed = await asyncio.to_thread(term.read, 2.0)
+                out += data
+                if b"T42T" in out or closed:
+                    break
+            assert b"T42T" in out, out[-500:]
+            # interactive state persists across writes (shell variable)
+            await asyncio.to_thread(term.write, "X=hello\n")
+            await asyncio.to_thread(term.write, "echo GOT:$X\n")
+            out2 = b""
+            for _ in range(20):
+                data, closed = await asyncio.to_thread(term.read, 2.0)
+                out2 += data
+                if b"GOT:hello" in out2 or closed:
+                    break
+            assert b"GOT:hello" in out2, out2[-500:]
+            await asyncio.to_thread(term.close)
+        finally:
+            await asyncio.to_thread(sb.terminate)
+
+
+@pytest.mark.asyncio
+async def test_async_sandbox_api():
+    """AsyncSandbox: the asyncio twin of the Sandbox surface (reference:
+    sdk sandbox.py sync+asyncio APIs)."""
+    import beta9_amd
+
+    from .helpers import cluster
+
+    async with cluster(n_workers=1) as c:
+        inst = await beta9_amd.AsyncSandbox(cpu=100, memory=128).create()
+        try:
+            r = await inst.run_code("print(6*7)")
+            assert "42" in r["stdout"]
+            await inst.upload("/tmp/workspace/a.txt", b"async-bytes")
+            assert await inst.download("/tmp/workspace/a.txt") == b"async-bytes"
+            names = [f["name"] for f in await inst.ls("/tmp/workspace")]
+            assert "a.txt" in names
+            st = await inst.status()
+            assert st.get("ok") or st.get("running", True)
+        finally:
+            await inst.terminate()
+            await inst.close()
