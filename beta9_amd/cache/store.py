@@ -26,6 +26,8 @@ class ContentStore:
         quota_bytes: int = 200 * (1 << 30),
         eviction_watermark: float = 0.80,
         hot_guard_s: float = 600.0,
+        memory_tier_bytes: int = 512 * (1 << 20),
+        memory_max_item: int = 64 * (1 << 20),
     ):
         self.root = root
         self.page_size = page_size
@@ -33,7 +35,18 @@ class ContentStore:
         self.eviction_watermark = eviction_watermark
         self.hot_guard_s = hot_guard_s
         os.makedirs(root, exist_ok=True)
-        self._lock = threading.Lock()
+        self._lock = threading.RLock()  # remove() runs under _maybe_evict's lock
+        # memory tier (reference: pkg/cache disk+memory tiers,
+        # storage.go): hot content cached in RAM, LRU-bounded; reads of
+        # resident content never touch the disk
+        self.memory_tier_bytes = memory_tier_bytes
+        self.memory_max_item = memory_max_item
+        from collections import OrderedDict
+
+        self._mem: "OrderedDict[str, bytes]" = OrderedDict()
+        self._mem_used = 0
+        self.mem_hits = 0
+        self.mem_misses = 0
 
     # -- paths -------------------------------------------------------------
     def _dir(self, digest: str) -> str:
@@ -131,12 +144,44 @@ class ContentStore:
         return self.content_path(digest)
 
     def read(self, digest: str, offset: int = 0, length: int = 0) -> Optional[bytes]:
+        with self._lock:
+            data = self._mem.get(digest)
+            if data is not None:
+                self._mem.move_to_end(digest)  # LRU refresh
+                self.mem_hits += 1
+                if offset or length:
+                    return data[offset: offset + length] if length else data[offset:]
+                return data
         p = self.get_path(digest)
         if p is None:
             return None
+        self.mem_misses += 1
         with open(p, "rb") as f:
             f.seek(offset)
-            return f.read(length) if length else f.read()
+            out = f.read(length) if length else f.read()
+        if offset == 0 and not length:
+            self._mem_promote(digest, out)
+        return out
+
+    def _mem_promote(self, digest: str, data: bytes) -> None:
+        if len(data) > self.memory_max_item or self.memory_tier_bytes <= 0:
+            return
+        with self._lock:
+            if digest in self._mem:
+                return
+            self._mem[digest] = data
+            self._mem_used += len(data)
+            while self._mem_used > self.memory_tier_bytes and self._mem:
+                _, old = self._mem.popitem(last=False)
+                self._mem_used -= len(old)
+
+    def mem_stats(self) -> dict:
+        return {
+            "items": len(self._mem),
+            "bytes": self._mem_used,
+            "hits": self.mem_hits,
+            "misses": self.mem_misses,
+        }
 
     def _touch(self, digest: str) -> None:
         # refresh LRU stamp, throttled (reference: 5 min throttle)
@@ -195,6 +240,10 @@ class ContentStore:
                 self.remove(digest)
 
     def remove(self, digest: str) -> None:
+        with self._lock:
+            old = self._mem.pop(digest, None)
+            if old is not None:
+                self._mem_used -= len(old)
         shutil.rmtree(self._dir(digest), ignore_errors=True)
 
 

@@ -174,3 +174,38 @@ async def test_reconciler_skips_dead_peer(tmp_path):
         assert await rb.ensure(digest) is not None
     finally:
         await srv_a.stop()
+
+
+def test_memory_tier_lru(tmp_path):
+    """Hot content serves from the RAM tier; LRU bounded by budget
+    (reference: pkg/cache memory tier + eviction watermarks)."""
+    from beta9_amd.cache.store import ContentStore
+
+    s = ContentStore(str(tmp_path), memory_tier_bytes=3000, memory_max_item=1500)
+    d1 = s.add_bytes(b"a" * 1000)
+    d2 = s.add_bytes(b"b" * 1000)
+    d3 = s.add_bytes(b"c" * 1000)
+    big = s.add_bytes(b"X" * 2000)  # > memory_max_item: never promoted
+
+    assert s.read(d1) == b"a" * 1000  # miss -> promote
+    assert s.read(d1) == b"a" * 1000  # hit
+    st = s.mem_stats()
+    assert st["hits"] == 1 and st["items"] == 1
+
+    s.read(d2); s.read(d3)           # fills to 3000
+    s.read(big)                      # too big: not promoted
+    assert s.mem_stats()["items"] == 3
+    # a 4th small item evicts the LRU (d1)
+    d4 = s.add_bytes(b"d" * 1000)
+    s.read(d4)
+    st = s.mem_stats()
+    assert st["items"] == 3 and st["bytes"] <= 3000
+    hits_before = st["hits"]
+    s.read(d1)  # evicted -> miss again
+    assert s.mem_stats()["hits"] == hits_before
+    # ranged reads from memory
+    s.read(d4)
+    assert s.read(d4, offset=10, length=5) == b"d" * 5
+    # removal drops the memory copy
+    s.remove(d4)
+    assert s.read(d4) is None
