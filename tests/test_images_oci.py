@@ -41,6 +41,18 @@ def _tar_bytes(entries):
     return buf.getvalue()
 
 
+def _mknod_available(tmp_path) -> bool:
+    import stat as _stat
+
+    probe = str(tmp_path / ".mknod-probe")
+    try:
+        os.mknod(probe, _stat.S_IFCHR | 0o600, os.makedev(0, 0))
+        os.unlink(probe)
+        return True
+    except (OSError, PermissionError):
+        return False
+
+
 def test_layer_extract_and_whiteouts(tmp_path):
     store = LayerStore(str(tmp_path))
     l1 = _tar_bytes([("etc", None), ("etc/keep.conf", "keep"), ("etc/gone.conf", "bye"),
@@ -55,8 +67,13 @@ def test_layer_extract_and_whiteouts(tmp_path):
     d2 = store.add_layer_from_tar(io.BytesIO(l2))
     fs2 = store.layer_fs(d2)
     wh = os.path.join(fs2, "etc/gone.conf")
-    st = os.stat(wh)
-    assert st.st_rdev == 0 and not st.st_size  # char 0:0 whiteout
+    if _mknod_available(tmp_path):
+        st = os.stat(wh)
+        assert st.st_rdev == 0 and st.st_size == 0  # char 0:0 whiteout
+    else:
+        # locked-down containers (e.g. the GPU pool) deny mknod: the
+        # extractor skips whiteout materialization rather than failing
+        assert not os.path.exists(wh)
 
     # dedupe: same bytes -> same digest, no re-extract
     assert store.add_layer_from_tar(io.BytesIO(l1)) == d1
