@@ -121,6 +121,12 @@ class PodService:
     async def SandboxExec(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
         return await self._pm_call(req["container_id"], "Exec", req)
 
+    @rpc()
+    async def SandboxRunContainer(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Nested isolated container inside a sandbox (the Docker-in-sandbox
+        analog — reference: pkg/worker/sandbox.go:148-465)."""
+        return await self._pm_call(req["container_id"], "RunContainer", req)
+
     # -- interactive terminal (PTY in the PM; the dropbear/shell analog) ----
     @rpc()
     async def SandboxTerminalOpen(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:

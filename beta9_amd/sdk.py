@@ -789,6 +789,18 @@ class SandboxInstance:
     def terminal(self, rows: int = 24, cols: int = 80) -> "SandboxTerminal":
         return SandboxTerminal(self, rows=rows, cols=cols)
 
+    def run_container(self, command, image_id: str = "", cpu: int = 0,
+                      memory_mb: int = 0, block_network: bool = False,
+                      timeout: float = 120.0) -> Dict:
+        """Run a nested ISOLATED container inside this sandbox (the
+        Docker-in-sandbox analog): own namespaces + overlay rootfs
+        (optionally from an imported image's layers)."""
+        return self._call("SandboxRunContainer", {
+            "command": command, "image_id": image_id, "cpu": cpu,
+            "memory_mb": memory_mb, "block_network": block_network,
+            "timeout": timeout,
+        })
+
     def terminate(self) -> None:
         self.client.call("PodService", "StopPod", {"container_id": self.container_id})
 

@@ -406,6 +406,8 @@ class Lifecycle:
         env.setdefault("B9_GATEWAY_PORT", str(self.gateway_grpc_port))
         env.setdefault("CONTAINER_HOSTNAME", request.hostname or request.container_id)
         env.setdefault("B9_LIGHT_IMPORT", "1")  # runners skip the SDK surface
+        if self.image_store_path:
+            env.setdefault("B9_IMAGE_STORE", self.image_store_path)
         if request.task_id:
             env.setdefault("TASK_ID", request.task_id)
         lowerdirs: list = []

@@ -139,3 +139,33 @@ async def test_exposed_port_proxy_reaches_container():
             assert "served" in rr.get("stdout", "")
         finally:
             await asyncio.to_thread(client.call, "PodService", "StopPod", {"container_id": cid})
+
+
+@needs_net
+@pytest.mark.asyncio
+async def test_nested_container_in_sandbox():
+    """Docker-in-sandbox analog: a sandbox launches a NESTED isolated
+    container (namespaces-in-namespaces via b9launch) whose writes and
+    hostname don't leak into the sandbox."""
+    import beta9_amd
+
+    async with cluster(n_workers=1) as c:
+        sb = await asyncio.to_thread(beta9_amd.Sandbox(cpu=200, memory=256).create)
+        try:
+            r = await asyncio.to_thread(
+                sb.run_container,
+                "hostname; echo nested-says-$((6*7)); touch /nested-canary",
+            )
+            assert r.get("exit_code") == 0, r
+            assert "nested" in r["stdout"] and "nested-says-42" in r["stdout"]
+            # the nested container's write stayed in ITS overlay
+            chk = await asyncio.to_thread(sb.run_code, "import os; print(os.path.exists('/nested-canary'))")
+            assert "False" in chk["stdout"]
+            # block-network nested container has no interfaces beyond lo
+            r2 = await asyncio.to_thread(
+                sb.run_container, "ls /sys/class/net", block_network=True,
+            )
+            assert r2.get("exit_code") == 0, r2
+            assert r2["stdout"].split() == ["lo"], r2["stdout"]
+        finally:
+            await asyncio.to_thread(sb.terminate)
