@@ -328,7 +328,16 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256_kernel(
 // race-screens at 256/512/4096 (guide §5.4 two-lane discipline).
 // ---------------------------------------------------------------------------
 
-template <bool XCD_SWIZZLE>
+__device__ __forceinline__ void vmcnt_wait(int allowed) {
+  switch (allowed) {
+    case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+    case 2: asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); break;
+    case 4: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+    default: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+  }
+}
+
+template <bool XCD_SWIZZLE, int LEAD = 5>
 __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ Bt,
     bf16_t* __restrict__ C, int M, int N, int K) {
@@ -376,11 +385,17 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
   f32x4 acc[8][4] = {};
   int ntiles = K / BK;
   int total_halves = ntiles * 4;
-  const int LEAD = 5;
 
-  // prologue: 5 halves staged, first 4 (tile 0) landed
-  for (int h = 0; h < LEAD && h < total_halves; h++) stage_half(h);
-  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  // prologue: LEAD halves staged, first 4 (tile 0) landed. LEAD>5 stages
+  // into LDS regions whose last reader is 1-2 phases away — covered by
+  // HBM landing latency (>=900 cyc vs ~300-cyc phases) and REQUIRED to be
+  // race-screened per §5.4 two-lane discipline for each LEAD value.
+  {
+    int staged = total_halves < LEAD ? total_halves : LEAD;
+    for (int h = 0; h < staged; h++) stage_half(h);
+    int allowed = 2 * (staged - 4);
+    vmcnt_wait(allowed < 0 ? 0 : allowed);
+  }
   __builtin_amdgcn_s_barrier();
 
   int fr_row = lane & 15;
@@ -414,14 +429,14 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_bf16_nt_256p8_kernel(
       int h = kt * 4 + q + LEAD;
       if (h < total_halves) stage_half(h);
       if (q == 3) {
-        // next K-tile's 4 halves must land before its first phase reads:
-        // allow only the newest stage in flight (2 glds per wave). At the
-        // tail (kt == ntiles-2) the h==total stage above was guarded off, so
-        // the newest in-flight glds ARE the last tile's B1 — drain fully.
-        if (kt == ntiles - 2)
-          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        else
-          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        // next K-tile's 4 halves must land before its first phase reads.
+        // Per-wave in-flight glds allowed = 2 x (halves staged BEYOND tile
+        // kt+1's last half), clamped at the tail where stage guards turn
+        // off (generalizes the LEAD=5 vmcnt(2)/vmcnt(0) pair).
+        int beyond = total_halves - (kt + 2) * 4;
+        if (beyond > LEAD - 4) beyond = LEAD - 4;
+        int allowed = 2 * (beyond < 0 ? 0 : beyond);
+        vmcnt_wait(allowed);
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
@@ -1047,26 +1062,30 @@ torch::Tensor gemm_bf16_nt_residual(torch::Tensor A, torch::Tensor Bt,
 }
 
 torch::Tensor gemm_bf16_nt_p8(torch::Tensor A, torch::Tensor Bt,
-                              bool xcd_swizzle) {
+                              bool xcd_swizzle, int64_t lead) {
   check_bf16_2d(A, "A");
   check_bf16_2d(Bt, "Bt");
   int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
   TORCH_CHECK(Bt.size(1) == K, "K mismatch");
   TORCH_CHECK(M % BM2 == 0 && N % BN2 == 0 && K % BK == 0,
               "p8 kernel needs M,N %% 256 == 0 and K %% 64 == 0");
+  TORCH_CHECK(lead >= 5 && lead <= 7, "lead in [5,7]");
   auto C = torch::empty({M, N}, A.options());
   dim3 grid((M / BM2) * (N / BN2));
   auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(THREADS2), 0, stream,
+                       (const bf16_t*)A.data_ptr(), (const bf16_t*)Bt.data_ptr(),
+                       (bf16_t*)C.data_ptr(), (int)M, (int)N, (int)K);
+  };
   if (xcd_swizzle) {
-    hipLaunchKernelGGL((gemm_bf16_nt_256p8_kernel<true>), grid, dim3(THREADS2),
-                       0, stream, (const bf16_t*)A.data_ptr(),
-                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
-                       (int)M, (int)N, (int)K);
+    if (lead == 6) launch(gemm_bf16_nt_256p8_kernel<true, 6>);
+    else if (lead == 7) launch(gemm_bf16_nt_256p8_kernel<true, 7>);
+    else launch(gemm_bf16_nt_256p8_kernel<true, 5>);
   } else {
-    hipLaunchKernelGGL((gemm_bf16_nt_256p8_kernel<false>), grid,
-                       dim3(THREADS2), 0, stream, (const bf16_t*)A.data_ptr(),
-                       (const bf16_t*)Bt.data_ptr(), (bf16_t*)C.data_ptr(),
-                       (int)M, (int)N, (int)K);
+    if (lead == 6) launch(gemm_bf16_nt_256p8_kernel<false, 6>);
+    else if (lead == 7) launch(gemm_bf16_nt_256p8_kernel<false, 7>);
+    else launch(gemm_bf16_nt_256p8_kernel<false, 5>);
   }
   return C;
 }
@@ -1783,7 +1802,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_nt_p2", &gemm_bf16_nt_p2, py::arg("A"), py::arg("Bt"),
         py::arg("xcd_swizzle") = false);
   m.def("gemm_bf16_nt_p8", &gemm_bf16_nt_p8, py::arg("A"), py::arg("Bt"),
-        py::arg("xcd_swizzle") = false);
+        py::arg("xcd_swizzle") = false, py::arg("lead") = 5);
   m.def("silu_mul", &silu_mul);
   m.def("rmsnorm", &rmsnorm, py::arg("x"), py::arg("w"),
         py::arg("eps") = 1e-6);
