@@ -196,6 +196,23 @@ class ContainerRepo:
     async def stop_requested(self, container_id: str) -> bool:
         return await self.store.get(keys.container_stop_requested(container_id)) is not None
 
+    async def stop_events(self):
+        """Async iterator of container ids whose stop key was just set —
+        the keyspace-event-driven stop path (reference: worker
+        processStopContainerEvents via Redis keyspace events; replaces
+        per-container 5 Hz polling)."""
+        sub = await self.store.subscribe("container:stop:*")
+        try:
+            while True:
+                ev = await sub.get()
+                if ev is None:
+                    continue
+                event, key = ev
+                if event == "set":
+                    yield key.rsplit(":", 1)[-1]
+        finally:
+            sub.close()
+
     async def append_log(self, container_id: str, line: str, cap: int = 2000) -> None:
         key = keys.container_logs(container_id)
         await self.store.rpush(key, line)

@@ -95,6 +95,14 @@ class WorkerRepositoryService:
     async def StopRequested(self, req: Dict[str, Any], context, auth) -> bool:
         return await self._c.stop_requested(req["container_id"])
 
+    @rpc(kind="unary_stream")
+    async def StreamStopEvents(self, req: Dict[str, Any], context, auth):
+        """Keyspace-event-driven container stops (reference: worker
+        processStopContainerEvents via Redis keyspace events) — one stream
+        per worker instead of 5 Hz polling per container."""
+        async for cid in self._c.stop_events():
+            yield {"container_id": cid}
+
     @rpc()
     async def AppendLog(self, req: Dict[str, Any], context, auth) -> bool:
         await self._c.append_log(req["container_id"], req["line"])

@@ -490,18 +490,28 @@ class Lifecycle:
 
         return cb
 
+    async def stop_container(self, cid: str) -> None:
+        """Graceful-then-hard kill (driven by the worker's stop-event
+        stream; also the slow-poll fallback's action)."""
+        log.info("stop requested for %s", cid)
+        rt = self.runtime_of(cid)
+        await rt.kill(cid, 15)
+        await asyncio.sleep(5.0)
+        if cid in self.containers:
+            await rt.kill(cid, 9)
+
     async def _watch_stop(self, cid: str) -> None:
-        """Poll for stop requests (reference: worker
-        processStopContainerEvents via keyspace events)."""
+        """SLOW-poll safety net for stop requests that raced the event
+        stream (the fast path is the worker's single stop-event stream —
+        reference: processStopContainerEvents via keyspace events)."""
         try:
+            if await self.repo.stop_requested(cid):  # raced container start
+                await self.stop_container(cid)
+                return
             while True:
-                await asyncio.sleep(0.2)
+                await asyncio.sleep(2.0)
                 if await self.repo.stop_requested(cid):
-                    log.info("stop requested for %s", cid)
-                    rt = self.runtime_of(cid)
-                    await rt.kill(cid, 15)
-                    await asyncio.sleep(5.0)
-                    await rt.kill(cid, 9)
+                    await self.stop_container(cid)
                     return
         except asyncio.CancelledError:
             pass

@@ -72,6 +72,9 @@ class DirectRepoClient:
     async def stop_requested(self, container_id: str) -> bool:
         return await self._c.stop_requested(container_id)
 
+    def stream_stop_events(self):
+        return self._c.stop_events()
+
     async def append_log(self, container_id: str, line: str) -> None:
         await self._c.append_log(container_id, line)
 
@@ -150,6 +153,12 @@ class RemoteRepoClient:
 
     async def stop_requested(self, container_id: str) -> bool:
         return bool(await self._rpc.call(SVC, "StopRequested", {"container_id": container_id}))
+
+    async def stream_stop_events(self):
+        async for msg in self._rpc.stream(SVC, "StreamStopEvents", {}):
+            cid = msg.get("container_id") if isinstance(msg, dict) else None
+            if cid:
+                yield cid
 
     async def append_log(self, container_id: str, line: str) -> None:
         await self._rpc.call(SVC, "AppendLog", {"container_id": container_id, "line": line})

@@ -171,6 +171,20 @@ class Worker:
         self._run_task = asyncio.create_task(self._run())
         ka = asyncio.create_task(self._keepalive_loop())
         self._tasks.add(ka)
+        self._tasks.add(asyncio.create_task(self._stop_event_loop()))
+
+    async def _stop_event_loop(self) -> None:
+        """ONE stop-event stream per worker (keyspace events) instead of
+        per-container polling; reconnects on stream failure."""
+        while not self._stopping.is_set():
+            try:
+                async for cid in self.repo.stream_stop_events():
+                    if cid in self.lifecycle.containers:
+                        asyncio.create_task(self.lifecycle.stop_container(cid))
+            except asyncio.CancelledError:
+                return
+            except Exception:
+                await asyncio.sleep(1.0)  # gateway restart etc: retry
 
     async def wait_prewarmed(self, timeout: float = 120.0) -> bool:
         """Block until every prewarm shell is parked (bench setup)."""
