@@ -102,7 +102,9 @@ CREATE TABLE IF NOT EXISTS checkpoints (
   accelerator TEXT DEFAULT '',
   gpu_count INTEGER DEFAULT 0,
   size_bytes INTEGER DEFAULT 0,
-  created_at REAL NOT NULL
+  created_at REAL NOT NULL,
+  cache_digest TEXT DEFAULT '',
+  obj_digest TEXT DEFAULT ''
 );
 CREATE TABLE IF NOT EXISTS images (
   id TEXT PRIMARY KEY,
@@ -134,12 +136,15 @@ class BackendDB:
             self._conn.execute("PRAGMA journal_mode=WAL")
             self._conn.execute("PRAGMA synchronous=NORMAL")
             self._conn.executescript(_SCHEMA)
-            try:  # migrate pre-token_type DB files in place
-                self._conn.execute(
-                    "ALTER TABLE tokens ADD COLUMN token_type TEXT DEFAULT 'workspace'"
-                )
-            except sqlite3.OperationalError:
-                pass
+            for mig in (
+                "ALTER TABLE tokens ADD COLUMN token_type TEXT DEFAULT 'workspace'",
+                "ALTER TABLE checkpoints ADD COLUMN cache_digest TEXT DEFAULT ''",
+                "ALTER TABLE checkpoints ADD COLUMN obj_digest TEXT DEFAULT ''",
+            ):
+                try:  # migrate older DB files in place
+                    self._conn.execute(mig)
+                except sqlite3.OperationalError:
+                    pass
             self._conn.commit()
         # write-behind queue for hot-path task rows (authoritative task
         # state lives in the state store; rows are the durable audit log)
@@ -466,13 +471,14 @@ class BackendDB:
     # -- checkpoints ---------------------------------------------------------------
     def upsert_checkpoint(self, cp: Dict[str, Any]) -> None:
         self._exec(
-            "INSERT INTO checkpoints (id, workspace_id, stub_id, source_container_id, status, remote_key, stub_type, accelerator, gpu_count, size_bytes, created_at)"
-            " VALUES (?,?,?,?,?,?,?,?,?,?,?) ON CONFLICT(id) DO UPDATE SET status=excluded.status, remote_key=excluded.remote_key, size_bytes=excluded.size_bytes",
+            "INSERT INTO checkpoints (id, workspace_id, stub_id, source_container_id, status, remote_key, stub_type, accelerator, gpu_count, size_bytes, created_at, cache_digest, obj_digest)"
+            " VALUES (?,?,?,?,?,?,?,?,?,?,?,?,?) ON CONFLICT(id) DO UPDATE SET status=excluded.status, remote_key=excluded.remote_key, size_bytes=excluded.size_bytes, cache_digest=excluded.cache_digest, obj_digest=excluded.obj_digest",
             (
                 cp["id"], cp["workspace_id"], cp["stub_id"], cp.get("source_container_id", ""),
                 cp["status"], cp.get("remote_key", ""), cp.get("stub_type", ""),
                 cp.get("accelerator", ""), cp.get("gpu_count", 0), cp.get("size_bytes", 0),
                 cp.get("created_at", time.time()),
+                cp.get("cache_digest", ""), cp.get("obj_digest", ""),
             ),
         )
 

@@ -275,6 +275,8 @@ class Gateway:
         )
         self.cache_coordinator = CacheCoordinator(self.store)
         self.cache_service = CacheService(self.content_store, self.cache_coordinator)
+        # durable checkpoint tier rides on the content store
+        runner_svc.content_store = self.content_store
 
         from ..abstractions.disk import DiskService
         from ..worker.disk import DiskManager

@@ -496,6 +496,20 @@ class RunnerService:
         stub_id = req["stub_id"]
         ws = auth["workspace_id"]
         stub = self.db.get_stub(stub_id)
+        # durable tier: ingest the snapshot (+ objsnap) into the
+        # content-addressed store so ANY node can materialize it over the
+        # cache transport (reference: checkpoint upload to cache/S3,
+        # pkg/worker/criu.go:320-430)
+        cache_digest = obj_digest = ""
+        cs = getattr(self, "content_store", None)
+        if cs is not None and self.checkpoint_dir:
+            import asyncio as _a
+
+            snap = os.path.join(self.checkpoint_dir, f"{req['key']}.hipsnap")
+            if os.path.exists(snap):
+                cache_digest = await _a.to_thread(cs.add_file, snap)
+            if os.path.exists(snap + ".obj"):
+                obj_digest = await _a.to_thread(cs.add_file, snap + ".obj")
         self.db.upsert_checkpoint(
             {
                 "id": req["key"],
@@ -509,6 +523,8 @@ class RunnerService:
                 "gpu_count": int(req.get("gpu_count", 1)),
                 "size_bytes": int(req.get("size_bytes", 0)),
                 "created_at": time.time(),
+                "cache_digest": cache_digest,
+                "obj_digest": obj_digest,
             }
         )
         await self.store.delete(f"checkpoint:taking:{stub_id}")

@@ -123,6 +123,8 @@ class Scheduler:
                     accelerator=cp.get("accelerator", ""),
                     gpu_count=cp.get("gpu_count", 0),
                     size_bytes=cp.get("size_bytes", 0),
+                    cache_digest=cp.get("cache_digest", ""),
+                    obj_digest=cp.get("obj_digest", ""),
                 )
         st = ContainerState(
             container_id=request.container_id,

@@ -182,6 +182,12 @@ class CheckpointState(BaseModel):
     gpu_count: int = 0
     size_bytes: int = 0
     created_at: float = 0.0
+    # durable tier: content-store digests so any node can materialize the
+    # snapshot over the cache transport (reference: checkpoint cache_hash,
+    # pkg/types/backend.go:444-464)
+    cache_digest: str = ""
+    obj_digest: str = ""
+
 
 
 class PricingPolicy(BaseModel):

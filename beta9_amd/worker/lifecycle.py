@@ -80,6 +80,7 @@ class Lifecycle:
         isolation: str = "native",
         mask_paths=None,
         netplane=None,  # NetworkPlane
+        data_root: str = "",
     ):
         self.worker_id = worker_id
         self.repo = repo
@@ -94,6 +95,7 @@ class Lifecycle:
         self.isolation = isolation
         self.mask_paths = list(mask_paths or [])
         self.netplane = netplane
+        self.data_root = data_root
         self.containers: Dict[str, ContainerInstance] = {}
         from .usage import UsageTracker
 
@@ -166,6 +168,10 @@ class Lifecycle:
                 # HBM state back through pinned staging during boot.
                 spec.env["B9_CHECKPOINT_KEY"] = request.checkpoint.remote_key
                 spec.env["B9_CHECKPOINT_ID"] = request.checkpoint.checkpoint_id
+                with rec.span("worker.checkpoint_materialize"):
+                    local_dir = await self._materialize_checkpoint(request.checkpoint, spec.env)
+                    if local_dir:
+                        spec.env["B9_CHECKPOINT_DIR"] = local_dir
 
             # per-container network namespace (pods/sandboxes + any request
             # exposing ports): preallocated slot with veth/bridge/IP
@@ -308,6 +314,57 @@ class Lifecycle:
                 await self.repo.delete_container_state(cid)
             inst.done.set()
             self.containers.pop(cid, None)
+
+    async def _materialize_checkpoint(self, cp, env: Dict[str, str]) -> str:
+        """Make the hipsnap snapshot (+objsnap) present on THIS node.
+
+        Single-node: the gateway-provided B9_CHECKPOINT_DIR is a shared
+        path and the files already exist — no work. Remote workers pull
+        the content-store digests over the gateway's CacheService (the
+        durable tier — reference: checkpoint pulled from cache/S3 on
+        restore, pkg/worker/criu.go:147-233)."""
+        gw_dir = env.get("B9_CHECKPOINT_DIR", "")
+        snap_name = f"{cp.remote_key}.hipsnap"
+        if gw_dir and os.path.exists(os.path.join(gw_dir, snap_name)):
+            return ""  # shared filesystem: nothing to do
+        if not cp.cache_digest:
+            return ""
+        base = self.data_root or "/var/lib/b9/worker"
+        local_dir = os.path.join(base, "pulled-checkpoints")
+        os.makedirs(local_dir, exist_ok=True)
+        dst = os.path.join(local_dir, snap_name)
+        if not os.path.exists(dst):
+            ok = await self._cache_pull(cp.cache_digest, dst)
+            if not ok:
+                return ""
+        if cp.obj_digest and not os.path.exists(dst + ".obj"):
+            await self._cache_pull(cp.obj_digest, dst + ".obj")
+        return local_dir
+
+    async def _cache_pull(self, digest: str, dst: str) -> bool:
+        """Stream one content-store object from the gateway's CacheService
+        (over the worker's authenticated gRPC channel)."""
+        rpc = getattr(self.repo, "_rpc", None)
+        if rpc is None:
+            return False  # embedded worker shares the gateway filesystem
+        tmp = dst + ".pull"
+        try:
+            with open(tmp, "wb") as f:
+                async for chunk in rpc.stream(
+                    "CacheService", "GetContentStream", {"hash": digest}
+                ):
+                    if isinstance(chunk, dict):
+                        chunk = chunk.get("data", b"")
+                    f.write(chunk)
+            os.rename(tmp, dst)
+            return True
+        except Exception:
+            log.exception("checkpoint pull %s failed", digest)
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            return False
 
     def _image_store(self):
         if self.images is None and self.image_store_path:

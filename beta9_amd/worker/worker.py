@@ -146,6 +146,7 @@ class Worker:
             isolation=isolation,
             mask_paths=mask_paths,
             netplane=self.netplane,
+            data_root=data_root,
         )
         self._sema = asyncio.Semaphore(container_start_limit)
         self._keepalive_interval_s = keepalive_interval_s

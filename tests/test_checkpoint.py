@@ -149,3 +149,65 @@ async def test_scale_to_zero_resume_gpu():
                 assert not await c.gateway.container_repo.containers_for_stub(stub_id)
         finally:
             await w.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_checkpoint_durable_tier_and_remote_materialize(tmp_path):
+    """CommitCheckpoint ingests the snapshot into the content store
+    (durable tier); a worker WITHOUT the shared checkpoint dir pulls it
+    over the gateway CacheService (reference: checkpoint upload to
+    cache/S3 + restore prefetch, pkg/worker/criu.go:320-430, :147-233)."""
+    import os
+
+    import beta9_amd
+    from beta9_amd.types import CheckpointState, CheckpointStatus
+    from beta9_amd.worker.repo_client import RemoteRepoClient
+
+    async with cluster(n_workers=1) as c:
+        gw = c.gateway
+        ws = c.creds["workspace_id"]
+        stub = gw.db.get_or_create_stub(
+            ws, "ckpt-tier", "endpoint/deployment",
+            gw.db.get_stub_config(gw.db.get_or_create_stub(ws, "seed", "function",
+                __import__("beta9_amd.types", fromlist=["StubConfig"]).StubConfig())["id"]),
+        )
+        key = f"{ws}-{stub['id']}"
+        ckpt_dir = gw.cfg.checkpoint.store_path
+        os.makedirs(ckpt_dir, exist_ok=True)
+        payload = os.urandom(256 * 1024)
+        with open(os.path.join(ckpt_dir, f"{key}.hipsnap"), "wb") as f:
+            f.write(payload)
+        with open(os.path.join(ckpt_dir, f"{key}.hipsnap.obj"), "wb") as f:
+            f.write(b"objsnap-bytes")
+
+        client = beta9_amd.Client()
+        ok = await asyncio.to_thread(
+            client.call, "RunnerService", "CommitCheckpoint",
+            {"stub_id": stub["id"], "key": key, "size_bytes": len(payload)},
+        )
+        assert ok is True
+        row = gw.db.latest_checkpoint(ws, stub["id"])
+        assert row and row["cache_digest"] and row["obj_digest"]
+        assert gw.content_store.has(row["cache_digest"])
+
+        # a "remote" worker: RemoteRepoClient + its own data root, and the
+        # shared checkpoint dir NOT present in its env
+        from beta9_amd.worker.lifecycle import Lifecycle
+        from beta9_amd.worker.runtime.proc import ProcRuntime
+
+        repo = RemoteRepoClient(c.grpc_addr, token=c.token)
+        lc = Lifecycle(
+            worker_id="remote-ck", repo=repo, runtime=ProcRuntime(),
+            gpu_manager=None, gateway_host="127.0.0.1",
+            gateway_grpc_port=gw.grpc_port, data_root=str(tmp_path),
+        )
+        cp = CheckpointState(
+            checkpoint_id=key, remote_key=key, status=CheckpointStatus.available,
+            cache_digest=row["cache_digest"], obj_digest=row["obj_digest"],
+        )
+        local_dir = await lc._materialize_checkpoint(cp, {"B9_CHECKPOINT_DIR": "/nonexistent"})
+        assert local_dir, "remote materialization should produce a local dir"
+        got = open(os.path.join(local_dir, f"{key}.hipsnap"), "rb").read()
+        assert got == payload
+        assert open(os.path.join(local_dir, f"{key}.hipsnap.obj"), "rb").read() == b"objsnap-bytes"
+        await repo.close() if hasattr(repo, "close") else None
