@@ -182,6 +182,79 @@ class Image:
         self.base_image = ref
         return self
 
+    def from_dockerfile(self, path: str, context_dir: str = "") -> "Image":
+        """Build from a Dockerfile (reference: sdk Image.from_dockerfile,
+        SURVEY §A.7). Supported subset — FROM (local OCI layout/archive or
+        imported image id), RUN, ENV, WORKDIR, COPY (from context_dir) —
+        mapped onto the container build pipeline; unsupported instructions
+        raise rather than silently no-op."""
+        import shlex
+
+        ctx = context_dir or os.path.dirname(os.path.abspath(path)) or "."
+        with open(path) as f:
+            lines = f.read().splitlines()
+        # join line continuations
+        joined, buf = [], ""
+        for ln in lines:
+            s = ln.strip()
+            if not s or s.startswith("#"):
+                continue
+            if s.endswith("\\"):
+                buf += s[:-1] + " "
+                continue
+            joined.append(buf + s)
+            buf = ""
+        workdir = ""
+        for ln in joined:
+            op, _, rest = ln.partition(" ")
+            op = op.upper()
+            rest = rest.strip()
+            if op == "FROM":
+                self.base_image = rest.split()[0]
+            elif op == "RUN":
+                cmd = rest
+                if workdir:
+                    cmd = f"cd {shlex.quote(workdir)} && ({cmd})"
+                self.commands.append(cmd)
+            elif op == "ENV":
+                first = rest.split()[0] if rest.split() else ""
+                if "=" in first:  # ENV K=V [K2=V2...]
+                    self.env_vars.extend(rest.split())
+                else:  # legacy ENV K V
+                    k, _, v = rest.partition(" ")
+                    self.env_vars.append(f"{k}={v.strip()}")
+            elif op == "WORKDIR":
+                workdir = rest
+                self.commands.append(f"mkdir -p {shlex.quote(workdir)}")
+            elif op == "COPY":
+                parts = shlex.split(rest)
+                if len(parts) < 2:
+                    raise ValueError(f"COPY needs src dst: {ln}")
+                srcs, dst = parts[:-1], parts[-1]
+                if workdir and not dst.startswith("/"):
+                    dst = os.path.join(workdir, dst)
+                for src in srcs:
+                    full = os.path.join(ctx, src)
+                    if not os.path.exists(full):
+                        raise FileNotFoundError(f"COPY source {full} not in context")
+                    # context files travel as base64 into a build step (the
+                    # build container cannot see the client filesystem)
+                    import base64 as _b64
+
+                    data = open(full, "rb").read() if os.path.isfile(full) else None
+                    if data is None:
+                        raise ValueError(f"COPY of directories not supported yet: {src}")
+                    b = _b64.b64encode(data).decode()
+                    self.commands.append(
+                        f"mkdir -p $(dirname {shlex.quote(dst)}) && "
+                        f"echo {b} | base64 -d > {shlex.quote(dst)}"
+                    )
+            elif op in ("CMD", "ENTRYPOINT", "EXPOSE", "LABEL", "ARG", "USER"):
+                continue  # runtime-level; stubs own entrypoint/ports
+            else:
+                raise ValueError(f"unsupported Dockerfile instruction: {op}")
+        return self
+
     @property
     def is_empty(self) -> bool:
         if getattr(self, "_snapshot_id", ""):

@@ -142,17 +142,25 @@ def test_import_oci_layout(tmp_path):
     assert store.import_oci(str(layout)) == image_id
 
 
-def _mini_rootfs_tar():
-    """A runnable rootfs: /bin/sh plus the shared libs it needs, as a tar."""
+def _mini_rootfs_tar(extra_bins=()):
+    """A runnable rootfs: /bin/sh (+ requested binaries) plus the shared
+    libs they need, as a tar."""
+    import shutil as _sh
+
     buf = io.BytesIO()
-    sh = os.path.realpath("/bin/sh")
+    bins = {"/bin/sh": os.path.realpath("/bin/sh")}
+    for name in extra_bins:
+        p = _sh.which(name)
+        if p:
+            bins["/" + os.path.relpath(p, "/")] = os.path.realpath(p)
     libs = set()
-    out = subprocess.run(["ldd", sh], capture_output=True, text=True).stdout
-    for line in out.splitlines():
-        parts = line.split()
-        for p in parts:
-            if p.startswith("/") and os.path.exists(p):
-                libs.add(p)  # ORIGINAL path: the ELF interp header uses it
+    for real in bins.values():
+        out = subprocess.run(["ldd", real], capture_output=True, text=True).stdout
+        for line in out.splitlines():
+            parts = line.split()
+            for p in parts:
+                if p.startswith("/") and os.path.exists(p):
+                    libs.add(p)  # ORIGINAL path: the ELF interp header uses it
 
     def add_file(tf, src, arcname):
         with open(src, "rb") as f:
@@ -163,7 +171,8 @@ def _mini_rootfs_tar():
         tf.addfile(info, io.BytesIO(data))
 
     with tarfile.open(fileobj=buf, mode="w") as tf:
-        add_file(tf, sh, "bin/sh")
+        for arc, real in bins.items():
+            add_file(tf, real, arc)
         for lib in libs:
             add_file(tf, os.path.realpath(lib), lib)
         for d in ("etc", "srv", "root"):
@@ -241,3 +250,48 @@ async def test_rootfs_build_captures_upper(tmp_path):
         assert len(m["layers"]) == 2  # base + captured delta
         top_fs = svc.store.layers.layer_fs(m["layers"][-1])
         assert open(os.path.join(top_fs, "srv/built.txt")).read().strip() == "built-content"
+
+
+@needs_ns
+@pytest.mark.asyncio
+async def test_from_dockerfile_rootfs_build(tmp_path):
+    """Image.from_dockerfile: FROM local-OCI + RUN/ENV/WORKDIR/COPY mapped
+    onto the in-container build pipeline (reference: sdk from_dockerfile +
+    v2 dockerfile builds, pkg/abstractions/image/build.go:369-628)."""
+    import asyncio
+
+    import beta9_amd
+
+    from .helpers import cluster
+
+    # base image: mini rootfs as an OCI layout on disk
+    layout = tmp_path / "base-layout"
+    os.makedirs(layout)
+    _make_oci_layout(
+        str(layout),
+        [_mini_rootfs_tar(extra_bins=("mkdir", "base64", "dirname", "cat"))],
+        config={"config": {"Entrypoint": ["/bin/sh"]}},
+    )
+    ctx = tmp_path / "ctx"
+    ctx.mkdir()
+    (ctx / "app.txt").write_text("dockerfile-payload")
+    df = ctx / "Dockerfile"
+    df.write_text(
+        f"""
+FROM {layout}
+ENV GREETING=hello MODE=fast
+WORKDIR /srv/app
+COPY app.txt app.txt
+RUN echo built > /srv/app/built.txt
+"""
+    )
+    async with cluster(n_workers=1) as c:
+        img = beta9_amd.Image().from_dockerfile(str(df), context_dir=str(ctx))
+        assert img.base_image == str(layout)
+        assert "GREETING=hello" in img.env_vars and "MODE=fast" in img.env_vars
+        image_id = await asyncio.to_thread(img.build)
+        m = c.gateway.image_service.store.get_manifest(image_id)
+        assert m["kind"] == "rootfs"
+        top = c.gateway.image_service.store.layers.layer_fs(m["layers"][-1])
+        assert open(os.path.join(top, "srv/app/app.txt")).read() == "dockerfile-payload"
+        assert open(os.path.join(top, "srv/app/built.txt")).read().strip() == "built"
