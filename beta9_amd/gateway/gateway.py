@@ -250,7 +250,10 @@ class Gateway:
             return True
 
         self.scheduler.concurrency_check = concurrency_check
-        repo_svc = WorkerRepositoryService(self.worker_repo, self.container_repo, events=self.events)
+        repo_svc = WorkerRepositoryService(
+            self.worker_repo, self.container_repo, events=self.events,
+            storage_root=self.cfg.storage.root,
+        )
 
         fn_exec = FunctionExecutor(self.db, self.scheduler, self.store, self.runner_env)
         self.dispatcher.register(FN_EXEC, fn_exec.execute)
@@ -275,8 +278,10 @@ class Gateway:
         )
         self.cache_coordinator = CacheCoordinator(self.store)
         self.cache_service = CacheService(self.content_store, self.cache_coordinator)
-        # durable checkpoint tier rides on the content store
+        # durable checkpoint tier + remote volume snapshots ride on the
+        # content store
         runner_svc.content_store = self.content_store
+        repo_svc._content_store = self.content_store
 
         from ..abstractions.disk import DiskService
         from ..worker.disk import DiskManager

@@ -6,6 +6,7 @@ Method names mirror beta9_amd/worker/repo_client.py RemoteRepoClient.
 
 from __future__ import annotations
 
+import os
 from typing import Any, Dict, Optional
 
 from ...backend.repos import ContainerRepo, WorkerRepo
@@ -16,10 +17,14 @@ from ...types import ContainerRequest, ContainerState, ContainerStatus, Worker
 class WorkerRepositoryService:
     service_name = "WorkerRepositoryService"
 
-    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo, events=None):
+    def __init__(self, worker_repo: WorkerRepo, container_repo: ContainerRepo, events=None,
+                 content_store=None, storage_root: str = ""):
         self._w = worker_repo
         self._c = container_repo
         self._events = events
+        self._content_store = content_store
+        self._storage_root = storage_root
+        self._volume_snap_cache: Dict[str, tuple] = {}
 
     @rpc()
     async def RegisterWorker(self, req: Dict[str, Any], context, auth) -> bool:
@@ -94,6 +99,55 @@ class WorkerRepositoryService:
     @rpc()
     async def StopRequested(self, req: Dict[str, Any], context, auth) -> bool:
         return await self._c.stop_requested(req["container_id"])
+
+    @rpc()
+    async def SnapshotVolumeContent(self, req: Dict[str, Any], context, auth) -> Dict[str, Any]:
+        """Tar a workspace-storage subtree into the content store so a
+        REMOTE worker can materialize read-only volume mounts over the
+        cache transport (the offline stand-in for GeeseFS/S3 workspace
+        storage — reference: pkg/storage/geese.go; L8). Results cached by
+        tree mtime."""
+        import asyncio as _a
+        import hashlib as _h
+        import io as _io
+        import tarfile as _tar
+
+        path = os.path.realpath(req["path"])
+        root = os.path.realpath(self._storage_root)
+        if os.path.commonpath([root, path]) != root or not os.path.isdir(path):
+            from ...rpc.channel import invalid
+
+            raise invalid(f"not a workspace-storage dir: {req['path']}")
+
+        def latest_mtime(p: str) -> float:
+            mt = os.path.getmtime(p)
+            for dp, _dn, fns in os.walk(p):
+                for fn in fns:
+                    try:
+                        mt = max(mt, os.path.getmtime(os.path.join(dp, fn)))
+                    except OSError:
+                        pass
+            return mt
+
+        mt = await _a.to_thread(latest_mtime, path)
+        cache_key = _h.sha256(f"{path}:{mt}".encode()).hexdigest()
+        cached = self._volume_snap_cache.get(path)
+        if cached and cached[0] == cache_key:
+            return {"digest": cached[1]}
+
+        def build() -> str:
+            buf = _io.BytesIO()
+            with _tar.open(fileobj=buf, mode="w") as tf:
+                tf.add(path, arcname=".")
+            return self._content_store.add_bytes(buf.getvalue())
+
+        if self._content_store is None:
+            from ...rpc.channel import invalid
+
+            raise invalid("no content store on this gateway")
+        digest = await _a.to_thread(build)
+        self._volume_snap_cache[path] = (cache_key, digest)
+        return {"digest": digest}
 
     @rpc(kind="unary_stream")
     async def StreamStopEvents(self, req: Dict[str, Any], context, auth):

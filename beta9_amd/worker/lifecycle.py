@@ -151,6 +151,10 @@ class Lifecycle:
             with rec.span("worker.spec_from_request"):
                 spec, needs_addr = self._spec_from_request(request, inst)
 
+            if any(isinstance(m, dict) and not m.get("system") for m in spec.mounts):
+                with rec.span("worker.setup_mounts"):
+                    await self._materialize_mounts(spec)
+
             if request.requires_gpu():
                 with rec.span("worker.gpu_assignment"):
                     if self.gpu is None or self.gpu.free_count < request.gpu_count:
@@ -340,6 +344,44 @@ class Lifecycle:
         if cp.obj_digest and not os.path.exists(dst + ".obj"):
             await self._cache_pull(cp.obj_digest, dst + ".obj")
         return local_dir
+
+    async def _materialize_mounts(self, spec) -> None:
+        """Make volume mounts present on THIS node. Shared-fs/single-node:
+        paths exist, no-op. Remote workers: read-only volumes are snapshot
+        + pulled over the cache (the offline GeeseFS/S3 stand-in — L8);
+        writable volumes without shared storage fail loudly."""
+        for m in spec.mounts:
+            if not isinstance(m, dict) or m.get("system"):
+                continue
+            local = m.get("local_path", "")
+            if not local or os.path.exists(local):
+                continue
+            if not m.get("read_only"):
+                raise RuntimeError(
+                    f"writable volume {local} is not on this node and has no "
+                    "shared storage; mark it read_only or mount shared storage"
+                )
+            rpc = getattr(self.repo, "_rpc", None)
+            if rpc is None:
+                raise RuntimeError(f"volume {local} missing on embedded worker")
+            r = await rpc.call(
+                "WorkerRepositoryService", "SnapshotVolumeContent", {"path": local}
+            )
+            digest = r["digest"]
+            base = self.data_root or "/var/lib/b9/worker"
+            dst = os.path.join(base, "pulled-volumes", digest)
+            if not os.path.isdir(dst):
+                tarball = dst + ".tar"
+                if not await self._cache_pull(digest, tarball):
+                    raise RuntimeError(f"volume pull failed for {local}")
+                import tarfile as _tar
+
+                os.makedirs(dst + ".tmp", exist_ok=True)
+                with _tar.open(tarball) as tf:
+                    tf.extractall(dst + ".tmp")
+                os.rename(dst + ".tmp", dst)
+                os.unlink(tarball)
+            m["local_path"] = dst
 
     async def _cache_pull(self, digest: str, dst: str) -> bool:
         """Stream one content-store object from the gateway's CacheService

@@ -127,3 +127,85 @@ async def test_worker_on_remote_host():
             await gw.shutdown()
         finally:
             await plane.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_remote_volume_snapshot_and_materialize(tmp_path):
+    """L8: read-only volume content travels to a remote node via
+    SnapshotVolumeContent (content store tar) + cache pull + extract
+    (offline stand-in for GeeseFS/S3 workspace storage)."""
+    import tempfile
+
+    from beta9_amd.common.config import AppConfig
+    from beta9_amd.gateway.gateway import Gateway
+    from beta9_amd.worker.lifecycle import Lifecycle
+    from beta9_amd.worker.repo_client import RemoteRepoClient
+    from beta9_amd.worker.runtime.proc import ProcRuntime
+
+    with tempfile.TemporaryDirectory() as tmp:
+        cfg = AppConfig().with_root(tmp)
+        cfg.gateway.grpc_port = 0
+        cfg.gateway.http_port = 0
+        gw = Gateway(cfg)
+        await gw.start()
+        try:
+            creds = gw.bootstrap_workspace("volsync")
+            vol = os.path.join(cfg.storage.root, creds["workspace_id"], "volumes", "data")
+            os.makedirs(os.path.join(vol, "sub"))
+            with open(os.path.join(vol, "sub", "f.txt"), "w") as f:
+                f.write("volume-bytes")
+
+            repo = RemoteRepoClient(f"127.0.0.1:{gw.grpc_port}", token=creds["token"])
+            lc = Lifecycle(
+                worker_id="rv", repo=repo, runtime=ProcRuntime(), gpu_manager=None,
+                gateway_host="127.0.0.1", gateway_grpc_port=gw.grpc_port,
+                data_root=str(tmp_path),
+            )
+            missing = os.path.join(str(tmp_path), "not-here", "gatewaypath")
+
+            # simulate the remote node: the gateway path is absent locally —
+            # point the mount at the REAL gateway path but pretend it's
+            # missing by using the RPC end-to-end on a spec whose path we
+            # first verify exists gateway-side, then exercise the miss path
+            # via a bind-equivalent copy location
+            class SpecStub:
+                mounts = [{"local_path": vol, "mount_path": "/data", "read_only": True}]
+
+            # same-host: path exists -> no-op
+            await lc._materialize_mounts(SpecStub)
+            assert SpecStub.mounts[0]["local_path"] == vol
+
+            # force the pull path: snapshot via RPC, then materialize by
+            # digest exactly as a remote node would
+            r = await repo._rpc.call(
+                "WorkerRepositoryService", "SnapshotVolumeContent", {"path": vol}
+            )
+            dst_tar = os.path.join(str(tmp_path), "vol.tar")
+            assert await lc._cache_pull(r["digest"], dst_tar)
+            import tarfile
+
+            out = os.path.join(str(tmp_path), "out")
+            with tarfile.open(dst_tar) as tf:
+                tf.extractall(out)
+            assert open(os.path.join(out, "sub", "f.txt")).read() == "volume-bytes"
+
+            # snapshot cache: unchanged tree -> same digest
+            r2 = await repo._rpc.call(
+                "WorkerRepositoryService", "SnapshotVolumeContent", {"path": vol}
+            )
+            assert r2["digest"] == r["digest"]
+
+            # containment: paths outside the storage root are rejected
+            with pytest.raises(Exception):
+                await repo._rpc.call(
+                    "WorkerRepositoryService", "SnapshotVolumeContent", {"path": "/etc"}
+                )
+
+            # writable missing volume fails loudly
+            class SpecRW:
+                mounts = [{"local_path": missing, "mount_path": "/rw"}]
+
+            with pytest.raises(RuntimeError):
+                await lc._materialize_mounts(SpecRW)
+        finally:
+            await gw.shutdown()
