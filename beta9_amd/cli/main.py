@@ -127,6 +127,94 @@ def worker_start():
     asyncio.run(amain())
 
 
+agent_app = typer.Typer(help="BYO-machine agent (reference: pkg/agent)")
+app.add_typer(agent_app, name="agent")
+
+
+@agent_app.command("join")
+def agent_join(
+    gateway_addr: str,
+    token: str = typer.Option(..., help="workspace/worker token"),
+    worker_id: str = typer.Option("", help="stable worker id (default: host-derived)"),
+    cpu_limit: int = typer.Option(0, help="millicores (0 = all cores)"),
+    memory_limit_mb: int = typer.Option(0, help="0 = autodetect"),
+    supervise: bool = typer.Option(True, help="restart the worker on exit"),
+    once: bool = typer.Option(False, help="preflight + single worker run (tests)"),
+):
+    """Join THIS machine to a gateway as a worker node (reference:
+    pkg/agent/agent.go:17-88 — preflight, persist identity, run the worker
+    under supervision). Preflight checks: gateway reachable+token valid,
+    GPU inventory, namespace privileges, disk space."""
+    import json as _json
+    import shutil
+    import socket
+    import subprocess
+    import time
+
+    from ..rpc.channel import RpcClient
+    from ..worker.gpu import discover_devices
+    from ..worker.runtime.native import namespaces_available
+
+    # -- preflight ----------------------------------------------------------
+    import asyncio
+
+    async def _auth_check():
+        c = RpcClient(gateway_addr, token=token)
+        try:
+            return await c.call("GatewayService", "Authorize", {"token": token}, timeout=10.0)
+        finally:
+            await c.close()
+
+    r = asyncio.run(_auth_check())
+    if not r.get("ok"):
+        typer.echo("preflight FAILED: token rejected by gateway")
+        raise typer.Exit(1)
+    typer.echo(f"gateway ok (workspace {r['workspace_name']})")
+    devices = discover_devices()
+    typer.echo(f"GPUs: {len(devices)}" + (f" ({devices[0].name})" if devices else ""))
+    typer.echo(f"namespace isolation: {'available' if namespaces_available() else 'UNAVAILABLE (ProcRuntime fallback)'}")
+    free_gb = shutil.disk_usage("/").free / 1e9
+    typer.echo(f"free disk: {free_gb:.0f} GB")
+    if free_gb < 5:
+        typer.echo("preflight FAILED: <5 GB free disk")
+        raise typer.Exit(1)
+
+    wid = worker_id or f"agent-{socket.gethostname()}"
+    if not cpu_limit:
+        cpu_limit = (os.cpu_count() or 8) * 1000
+    if not memory_limit_mb:
+        try:
+            with open("/proc/meminfo") as f:
+                kb = int(f.readline().split()[1])
+            memory_limit_mb = max(1024, int(kb / 1024 * 0.9))
+        except OSError:
+            memory_limit_mb = 16384
+    state = {
+        "gateway_addr": gateway_addr, "worker_id": wid,
+        "cpu_limit": cpu_limit, "memory_limit_mb": memory_limit_mb,
+        "joined_at": time.time(),
+    }
+    agent_path = os.path.join(os.path.dirname(CONFIG_PATH), "agent.json")
+    os.makedirs(os.path.dirname(agent_path), exist_ok=True)
+    with open(agent_path, "w") as f:
+        _json.dump(state, f)
+    typer.echo(f"agent identity persisted to {agent_path}")
+
+    env = dict(os.environ)
+    env.update({
+        "WORKER_ID": wid, "GATEWAY_ADDR": gateway_addr, "WORKER_TOKEN": token,
+        "CPU_LIMIT": str(cpu_limit), "MEMORY_LIMIT": str(memory_limit_mb),
+        "GPU_TYPE": devices[0].name if devices else "",
+    })
+    while True:
+        typer.echo(f"starting worker {wid} -> {gateway_addr}")
+        p = subprocess.run([sys.executable, "-u", "-m", "beta9_amd.worker"], env=env)
+        typer.echo(f"worker exited rc={p.returncode}")
+        if once or not supervise:
+            raise typer.Exit(p.returncode)
+        time.sleep(2.0)
+
+
 # -- config ----------------------------------------------------------------
 
 

@@ -96,3 +96,56 @@ def test_gateway_process_with_provisioned_workers(tmp_path):
                 os.killpg(proc.pid, signal.SIGKILL)
             except ProcessLookupError:
                 pass
+
+
+@pytest.mark.timeout(180)
+def test_agent_join(tmp_path):
+    """`b9 agent join`: preflight + identity persist + worker registration
+    (reference: pkg/agent/agent.go:17-88)."""
+    import asyncio
+
+    async def amain():
+        from beta9_amd.common.config import AppConfig
+        from beta9_amd.gateway.gateway import Gateway
+
+        cfg = AppConfig().with_root(str(tmp_path / "gw"))
+        cfg.gateway.grpc_port = 0
+        cfg.gateway.http_port = 0
+        gw = Gateway(cfg)
+        await gw.start()
+        creds = gw.bootstrap_workspace("agents")
+        env = dict(os.environ)
+        env["HOME"] = str(tmp_path)
+        env["B9_DATA_ROOT"] = str(tmp_path / "agentdata")
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "beta9_amd.cli", "agent", "join",
+             f"127.0.0.1:{gw.grpc_port}", "--token", creds["token"],
+             "--worker-id", "agent-test-w", "--once"],
+            env=env, cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+            text=True, start_new_session=True,
+        )
+        try:
+            deadline = time.monotonic() + 60
+            seen = False
+            while time.monotonic() < deadline:
+                ws = await gw.worker_repo.list()
+                if any(w.id == "agent-test-w" for w in ws):
+                    seen = True
+                    break
+                if proc.poll() is not None:
+                    raise AssertionError(f"agent exited: {proc.stdout.read()[-2000:]}")
+                await asyncio.sleep(0.3)
+            assert seen, "agent worker never registered"
+            assert os.path.exists(tmp_path / ".b9" / "agent.json")
+        finally:
+            try:
+                os.killpg(proc.pid, signal.SIGTERM)
+                proc.wait(timeout=15)
+            except (subprocess.TimeoutExpired, ProcessLookupError):
+                try:
+                    os.killpg(proc.pid, signal.SIGKILL)
+                except ProcessLookupError:
+                    pass
+            await gw.shutdown()
+
+    asyncio.run(amain())
