@@ -167,6 +167,7 @@ async def cold_iterations(gw, worker, ws, args, has_gpu: bool, n_iter: int,
                     for lst in worker.prewarm_pool._shells.values()
                     for s in lst
                     if (not has_gpu) == (not s.key)
+                    and s.key != worker.prewarm_pool.ZYGOTE_KEY
                 ]
                 if any(s.ready or os.path.exists(s.sock_path) for s in shells):
                     break

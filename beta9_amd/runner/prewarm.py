@@ -31,7 +31,13 @@ def main() -> int:
 
     import torch  # noqa: F401  (the big one)
 
-    if os.environ.get("WORKER_GPU_DEVICES", "") != "":
+    # zygote mode: ALL imports warmed but NO HIP context — the shell stays
+    # GPU-agnostic (HIP reads *_VISIBLE_DEVICES at first cuda.init, which
+    # happens after activation pins the env). Serves prewarm MISSES for any
+    # GPU at import-warm cost instead of full cold python+torch.
+    zygote = os.environ.get("B9_PREWARM_ZYGOTE") == "1"
+
+    if not zygote and os.environ.get("WORKER_GPU_DEVICES", "") != "":
         try:
             torch.cuda.init()
             torch.zeros(1, device="cuda:0")  # force full HIP context + pools
