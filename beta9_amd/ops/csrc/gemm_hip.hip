@@ -1577,7 +1577,8 @@ __global__ __launch_bounds__(256) void splitk_reduce_scale_kernel(
 
 torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
                                     torch::Tensor Wu, torch::Tensor g_scale,
-                                    torch::Tensor u_scale, double a_scale) {
+                                    torch::Tensor u_scale, double a_scale,
+                                    int64_t force_bnj) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.dim() == 2);
   TORCH_CHECK(Wg.dtype() == torch::kUInt8 && Wu.dtype() == torch::kUInt8);
   TORCH_CHECK(g_scale.dtype() == torch::kFloat32 && u_scale.dtype() == torch::kFloat32);
@@ -1592,7 +1593,8 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t tiles128 = (M / BM) * (N / 128);
-  if (N % 128 == 0 && tiles128 >= 256) {
+  bool wide = (N % 128 == 0) && (force_bnj == 4 || (force_bnj == 0 && tiles128 >= 256));
+  if (wide) {
     dim3 grid(tiles128);
     hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<4>), grid, dim3(THREADS),
                        0, stream, (const fp8_t*)A.data_ptr(),
@@ -1796,7 +1798,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("splitk") = -1);
   m.def("gemm_gate_up_silu_fp8", &gemm_gate_up_silu_fp8, py::arg("A"),
         py::arg("Wg"), py::arg("Wu"), py::arg("g_scale"), py::arg("u_scale"),
-        py::arg("a_scale"));
+        py::arg("a_scale"), py::arg("force_bnj") = 0);
   m.def("gemm_gate_up_silu", &gemm_gate_up_silu, py::arg("A"), py::arg("Wg"),
         py::arg("Wu"), py::arg("xcd_swizzle") = true);
   m.def("gemm_bf16_nt_p2", &gemm_bf16_nt_p2, py::arg("A"), py::arg("Bt"),
