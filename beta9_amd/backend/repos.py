@@ -225,6 +225,16 @@ class ContainerRepo:
 
     async def push_span(self, container_id: str, span: Dict[str, Any]) -> None:
         await self.store.rpush(keys.container_lifecycle_spans(container_id), span)
+        # OTLP-JSON export when configured (reference: OTel traces to Tempo,
+        # pkg/common/trace.go)
+        from ..common.otlp import get_exporter
+
+        exp = get_exporter()
+        if exp is not None:
+            try:
+                exp.export({"container_id": container_id, **span})
+            except Exception:
+                pass
 
     async def get_spans(self, container_id: str) -> List[Dict[str, Any]]:
         return await self.store.lrange(keys.container_lifecycle_spans(container_id), 0, -1)

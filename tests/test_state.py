@@ -83,3 +83,36 @@ async def test_keys_pattern():
     await s.set("other", 1)
     ks = await s.keys("task:state:*")
     assert sorted(ks) == ["task:state:1", "task:state:2"]
+
+
+@pytest.mark.asyncio
+async def test_otlp_span_export(tmp_path, monkeypatch):
+    """Lifecycle spans export in OTLP/JSON wire shape (SURVEY §5.1 — the
+    reference ships OTel to Tempo; we emit collector-ingestible lines)."""
+    import json
+
+    from beta9_amd.backend.repos import ContainerRepo
+    from beta9_amd.common import otlp
+    from beta9_amd.common.state import StateStore
+
+    path = str(tmp_path / "traces.jsonl")
+    monkeypatch.setattr(otlp, "_exporter", None)
+    monkeypatch.setenv("B9_OTLP_PATH", path)
+    store = StateStore()
+    repo = ContainerRepo(store)
+    await repo.push_span("c-otlp", {"name": "worker.cold_start", "start": 1.0,
+                                    "end": 1.5, "duration_ms": 500.0})
+    await repo.push_span("c-otlp", {"name": "worker.runtime_start", "start": 1.1,
+                                    "end": 1.2, "duration_ms": 100.0})
+    lines = [json.loads(l) for l in open(path)]
+    assert len(lines) == 2
+    span = lines[0]["resourceSpans"][0]["scopeSpans"][0]["spans"][0]
+    assert span["name"] == "worker.cold_start"
+    assert span["startTimeUnixNano"] == str(int(1.0 * 1e9))
+    t1 = lines[0]["resourceSpans"][0]["scopeSpans"][0]["spans"][0]["traceId"]
+    t2 = lines[1]["resourceSpans"][0]["scopeSpans"][0]["spans"][0]["traceId"]
+    assert t1 == t2  # one trace per container
+    rs = lines[0]["resourceSpans"][0]["resource"]["attributes"][0]
+    assert rs == {"key": "service.name", "value": {"stringValue": "beta9-amd"}}
+    otlp._exporter.close()
+    monkeypatch.setattr(otlp, "_exporter", None)
