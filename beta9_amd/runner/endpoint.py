@@ -334,6 +334,9 @@ def main() -> int:
             log_level="warning",
             lifespan="on",
             access_log=False,
+            # first-party RFC6455 protocol: the image has no
+            # websockets/wsproto, and @realtime stubs need WS routes
+            ws="beta9_amd.common.ws_server:B9WebSocketProtocol",
         )
         return 0
     try:
