@@ -1273,7 +1273,7 @@ __global__ __launch_bounds__(THREADS2, 1) void gemm_gate_up_silu_256_kernel(
 
 typedef unsigned char fp8_t;
 
-template <int BNJ, bool Q8 = false>
+template <int BNJ, bool Q8 = false, bool XCD_SWIZZLE = false>
 __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
     const fp8_t* __restrict__ A, const fp8_t* __restrict__ Wg,
     const fp8_t* __restrict__ Wu, bf16_t* __restrict__ Out,
@@ -1287,6 +1287,12 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
 
   int nwg_n = N / BNT;
   int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
   int tile_m = (wg / nwg_n) * BM;
   int tile_n = (wg % nwg_n) * BNT;
 
@@ -1423,7 +1429,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_gate_up_silu_fp8_kernel(
 // (128-B rows, 16-B-chunk XOR swizzle, glds width 16).
 // ---------------------------------------------------------------------------
 
-template <int BNJ>
+template <int BNJ, bool XCD_SWIZZLE = false>
 __global__ __launch_bounds__(THREADS, 1) void gemm_down_fp8_kernel(
     const fp8_t* __restrict__ A, const fp8_t* __restrict__ W,
     float* __restrict__ Cpart, int M, int N, int K, int splitk) {
@@ -1435,6 +1441,12 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_down_fp8_kernel(
   int nwg_n = N / BNT;
   int nwg_m = M / BM;
   int wg = blockIdx.x;
+  if (XCD_SWIZZLE) {
+    int nwg = gridDim.x;
+    int xcd = wg % 8, idx = wg / 8;
+    int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
   int slice = wg / (nwg_m * nwg_n);
   int t = wg % (nwg_m * nwg_n);
   int tile_m = (t / nwg_n) * BM;
@@ -1578,7 +1590,7 @@ __global__ __launch_bounds__(256) void splitk_reduce_scale_kernel(
 torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
                                     torch::Tensor Wu, torch::Tensor g_scale,
                                     torch::Tensor u_scale, double a_scale,
-                                    int64_t force_bnj) {
+                                    int64_t force_bnj, bool xcd_swizzle) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.dim() == 2);
   TORCH_CHECK(Wg.dtype() == torch::kUInt8 && Wu.dtype() == torch::kUInt8);
   TORCH_CHECK(g_scale.dtype() == torch::kFloat32 && u_scale.dtype() == torch::kFloat32);
@@ -1596,20 +1608,36 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   bool wide = (N % 128 == 0) && (force_bnj == 4 || (force_bnj == 0 && tiles128 >= 256));
   if (wide) {
     dim3 grid(tiles128);
-    hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<4>), grid, dim3(THREADS),
-                       0, stream, (const fp8_t*)A.data_ptr(),
-                       (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
-                       (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
-                       (const float*)u_scale.data_ptr(), (float)a_scale,
-                       (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    if (xcd_swizzle)
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<4, false, true>), grid,
+                         dim3(THREADS), 0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    else
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<4>), grid, dim3(THREADS),
+                         0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
   } else {
     dim3 grid((M / BM) * (N / 64));
-    hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<2>), grid, dim3(THREADS),
-                       0, stream, (const fp8_t*)A.data_ptr(),
-                       (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
-                       (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
-                       (const float*)u_scale.data_ptr(), (float)a_scale,
-                       (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    if (xcd_swizzle)
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<2, false, true>), grid,
+                         dim3(THREADS), 0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    else
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<2>), grid, dim3(THREADS),
+                         0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
   }
   return Out;
 }
@@ -1652,7 +1680,7 @@ torch::Tensor gemm_gate_up_silu_fp8_q8(torch::Tensor A, torch::Tensor Wg,
 
 torch::Tensor gemm_down_fp8(torch::Tensor A, torch::Tensor W,
                             torch::Tensor w_scale, double a_scale,
-                            torch::Tensor R, int64_t splitk) {
+                            torch::Tensor R, int64_t splitk, bool xcd_swizzle) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.dim() == 2);
   TORCH_CHECK(W.is_cuda() && W.dtype() == torch::kUInt8);
   TORCH_CHECK(w_scale.dtype() == torch::kFloat32 && w_scale.is_cuda());
@@ -1680,15 +1708,27 @@ torch::Tensor gemm_down_fp8(torch::Tensor A, torch::Tensor W,
   auto Cpart = torch::empty({splitk, M, N}, A.options().dtype(torch::kFloat32));
   dim3 grid(base_tiles * splitk);
   if (wide) {
-    hipLaunchKernelGGL((gemm_down_fp8_kernel<4>), grid, dim3(THREADS), 0,
-                       stream, (const fp8_t*)A.data_ptr(),
-                       (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
-                       (int)M, (int)N, (int)K, (int)splitk);
+    if (xcd_swizzle)
+      hipLaunchKernelGGL((gemm_down_fp8_kernel<4, true>), grid, dim3(THREADS),
+                         0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                         (int)M, (int)N, (int)K, (int)splitk);
+    else
+      hipLaunchKernelGGL((gemm_down_fp8_kernel<4>), grid, dim3(THREADS), 0,
+                         stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                         (int)M, (int)N, (int)K, (int)splitk);
   } else {
-    hipLaunchKernelGGL((gemm_down_fp8_kernel<2>), grid, dim3(THREADS), 0,
-                       stream, (const fp8_t*)A.data_ptr(),
-                       (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
-                       (int)M, (int)N, (int)K, (int)splitk);
+    if (xcd_swizzle)
+      hipLaunchKernelGGL((gemm_down_fp8_kernel<2, true>), grid, dim3(THREADS),
+                         0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                         (int)M, (int)N, (int)K, (int)splitk);
+    else
+      hipLaunchKernelGGL((gemm_down_fp8_kernel<2>), grid, dim3(THREADS), 0,
+                         stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)W.data_ptr(), (float*)Cpart.data_ptr(),
+                         (int)M, (int)N, (int)K, (int)splitk);
   }
   auto Out = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
   int64_t mn = M * N;
@@ -1795,10 +1835,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("a_scale"), py::arg("act2_scale"));
   m.def("gemm_down_fp8", &gemm_down_fp8, py::arg("A"), py::arg("W"),
         py::arg("w_scale"), py::arg("a_scale"), py::arg("R"),
-        py::arg("splitk") = -1);
+        py::arg("splitk") = -1, py::arg("xcd_swizzle") = false);
   m.def("gemm_gate_up_silu_fp8", &gemm_gate_up_silu_fp8, py::arg("A"),
         py::arg("Wg"), py::arg("Wu"), py::arg("g_scale"), py::arg("u_scale"),
-        py::arg("a_scale"), py::arg("force_bnj") = 0);
+        py::arg("a_scale"), py::arg("force_bnj") = 0,
+        py::arg("xcd_swizzle") = false);
   m.def("gemm_gate_up_silu", &gemm_gate_up_silu, py::arg("A"), py::arg("Wg"),
         py::arg("Wu"), py::arg("xcd_swizzle") = true);
   m.def("gemm_bf16_nt_p2", &gemm_bf16_nt_p2, py::arg("A"), py::arg("Bt"),
