@@ -36,7 +36,7 @@ def main():
     gq, g_s = g.quantize_fp8_rowwise(wg)
     uq, u_s = g.quantize_fp8_rowwise(wu)
     flops = 2 * 2 * M * K * N
-    for bnj in (2, 4):
+    for bnj in (1, 2, 4):
         us = t_us(lambda: ext.gemm_gate_up_silu_fp8(aq, gq, uq, g_s, u_s, a_s, bnj))
         out = ext.gemm_gate_up_silu_fp8(aq, gq, uq, g_s, u_s, a_s, bnj).float()
         adq = aq.view(torch.float8_e4m3fn).float() * a_s

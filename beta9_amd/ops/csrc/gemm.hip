@@ -1606,6 +1606,25 @@ torch::Tensor gemm_gate_up_silu_fp8(torch::Tensor A, torch::Tensor Wg,
   auto stream = at::cuda::getCurrentCUDAStream();
   int64_t tiles128 = (M / BM) * (N / 128);
   bool wide = (N % 128 == 0) && (force_bnj == 4 || (force_bnj == 0 && tiles128 >= 256));
+  if (force_bnj == 1) {
+    TORCH_CHECK(N % 32 == 0);
+    dim3 grid((M / BM) * (N / 32));
+    if (xcd_swizzle)
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<1, false, true>), grid,
+                         dim3(THREADS), 0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    else
+      hipLaunchKernelGGL((gemm_gate_up_silu_fp8_kernel<1>), grid, dim3(THREADS),
+                         0, stream, (const fp8_t*)A.data_ptr(),
+                         (const fp8_t*)Wg.data_ptr(), (const fp8_t*)Wu.data_ptr(),
+                         (bf16_t*)Out.data_ptr(), (const float*)g_scale.data_ptr(),
+                         (const float*)u_scale.data_ptr(), (float)a_scale,
+                         (int)M, (int)N, (int)K, (fp8_t*)nullptr, 1.0f);
+    return Out;
+  }
   if (wide) {
     dim3 grid(tiles128);
     if (xcd_swizzle)
