@@ -100,6 +100,7 @@ class EndpointInstance:
             gpu=cfg.runtime.gpu,
             gpu_request=list(cfg.runtime.gpus),
             gpu_count=cfg.runtime.gpu_count,
+            gpu_fraction=cfg.runtime.gpu_fraction,
             image_id=cfg.runtime.image_id,
             stub_id=self.stub_id,
             workspace_id=self.stub_row["workspace_id"],

@@ -57,6 +57,7 @@ class FunctionExecutor:
             gpu=cfg.runtime.gpu,
             gpu_request=list(cfg.runtime.gpus),
             gpu_count=cfg.runtime.gpu_count,
+            gpu_fraction=cfg.runtime.gpu_fraction,
             image_id=cfg.runtime.image_id,
             stub_id=task.stub_id,
             workspace_id=task.workspace_id,

@@ -50,6 +50,7 @@ class PodService:
             gpu=cfg.runtime.gpu,
             gpu_request=list(cfg.runtime.gpus),
             gpu_count=cfg.runtime.gpu_count,
+            gpu_fraction=cfg.runtime.gpu_fraction,
             image_id=cfg.runtime.image_id,
             stub_id=stub_id,
             workspace_id=auth["workspace_id"],

@@ -75,7 +75,7 @@ class WorkerRepo:
         cur = await self.get(worker.id)
         if cur is None or cur.resource_version != worker.resource_version:
             return False
-        gpu_needed = request.gpu_count if request.requires_gpu() else 0
+        gpu_needed = request.gpu_units()
         if (
             cur.free_cpu < request.cpu
             or cur.free_memory_mb < request.memory_mb
@@ -118,7 +118,7 @@ class WorkerRepo:
         cur = await self.get(worker_id)
         if cur is None:
             return
-        gpu = request.gpu_count if request.requires_gpu() else 0
+        gpu = request.gpu_units()
         cur.free_cpu = min(cur.total_cpu, cur.free_cpu + request.cpu)
         cur.free_memory_mb = min(cur.total_memory_mb, cur.free_memory_mb + request.memory_mb)
         cur.free_gpu_count = min(cur.total_gpu_count, cur.free_gpu_count + gpu)

@@ -144,6 +144,15 @@ class GPUBoot:
                 return
             torch.cuda.init()
             self.phases["hip_preinit_ms"] = (time.perf_counter() - t0) * 1000
+            # Shared-GPU (fractional) container: enforce the HBM budget the
+            # worker assigned — HIP time-shares the GCD natively, memory is
+            # the resource that needs fencing (Thunder-analog; see
+            # GPUManager._assign_fraction).
+            frac_s = os.environ.get("B9_GPU_MEMORY_FRACTION", "")
+            if frac_s:
+                frac = max(0.01, min(1.0, float(frac_s)))
+                torch.cuda.set_per_process_memory_fraction(frac, 0)
+                self.phases["gpu_memory_fraction"] = frac
             t1 = time.perf_counter()
             from ..ops import hipsnap
 
@@ -156,6 +165,13 @@ class GPUBoot:
                 if meta is not None:
                     total = sum(r["nbytes"] for r in meta["regions"])
                     size = int(total * 1.25) + (64 << 20)
+            if frac_s:
+                # the arena is raw hipMalloc (outside the torch allocator
+                # that set_per_process_memory_fraction fences) — cap it to
+                # the budgeted share of HBM explicitly
+                total = torch.cuda.get_device_properties(0).total_memory
+                budget = int(total * float(self.phases["gpu_memory_fraction"]) * 0.5)
+                size = min(size or hipsnap.DEFAULT_ARENA_BYTES, budget)
             # prefer the prewarm shell's pre-created arena (hipMalloc +
             # pinned staging already paid before the container started)
             self.arena = hipsnap.take_prewarmed(min_size=size)

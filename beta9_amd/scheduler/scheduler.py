@@ -282,7 +282,7 @@ class Scheduler:
             if request.requires_gpu():
                 if w.gpu not in request.gpu_request and GPU_TYPE_ANY not in request.gpu_request:
                     continue
-                if w.free_gpu_count < request.gpu_count:
+                if w.free_gpu_count < request.gpu_units():
                     continue
             elif w.gpu and w.total_gpu_count > 0 and request.preemptable is False and w.pool_name != request.pool_selector:
                 # keep CPU-only work off GPU workers unless explicitly pooled
@@ -312,7 +312,7 @@ class Scheduler:
         # pending worker already provisioning with room? just retry.
         pending_ok = any(
             w.status == WorkerStatus.pending
-            and (not request.requires_gpu() or (w.gpu in request.gpu_request and w.free_gpu_count >= request.gpu_count))
+            and (not request.requires_gpu() or (w.gpu in request.gpu_request and w.free_gpu_count >= request.gpu_units()))
             for w in workers
         )
         res = self._reservations.get(request.container_id)

@@ -372,6 +372,20 @@ class LLMTokenPressureAutoscaler:
 # ---------------------------------------------------------------------------
 
 
+def _parse_gpu(gpu: str) -> tuple:
+    """"mi355x" -> ("mi355x", 1.0); "mi355x:0.5" -> ("mi355x", 0.5)."""
+    if ":" not in gpu:
+        return gpu, 1.0
+    name, _, frac_s = gpu.partition(":")
+    try:
+        frac = float(frac_s)
+    except ValueError:
+        raise ValueError(f"bad gpu spec {gpu!r}: fraction must be a number")
+    if not (0.0 < frac <= 1.0):
+        raise ValueError(f"bad gpu spec {gpu!r}: fraction must be in (0, 1]")
+    return name, frac
+
+
 def _handler_spec(fn: Callable) -> str:
     module = fn.__module__
     if module == "__main__":
@@ -434,8 +448,13 @@ class RunnerAbstraction:
         self.func = func
         self.cpu = cpu
         self.memory = memory
-        self.gpu = gpu
-        self.gpu_count = gpu_count if gpu_count else (1 if gpu else 0)
+        # "mi355x:0.5" = half a shared device (Thunder-analog fractional
+        # GPU — pkg/worker/thunder.go; our worker packs shares onto one
+        # GCD and fences the HBM budget per process)
+        self.gpu, self.gpu_fraction = _parse_gpu(gpu)
+        self.gpu_count = gpu_count if gpu_count else (1 if self.gpu else 0)
+        if self.gpu_fraction < 1.0 and self.gpu_count > 1:
+            raise ValueError("fractional gpu requires gpu_count=1")
         self.image = image or Image()
         self.timeout = timeout
         self.retries = retries
@@ -486,6 +505,7 @@ class RunnerAbstraction:
                 gpu=self.gpu,
                 gpus=[self.gpu] if self.gpu else [],
                 gpu_count=self.gpu_count,
+                gpu_fraction=self.gpu_fraction,
                 image_id="",
             ),
             handler=_handler_spec(self.func) if self.func else "",

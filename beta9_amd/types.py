@@ -156,6 +156,14 @@ class RuntimeConfig(BaseModel):
     gpu: str = ""  # requested GPU type, "" = CPU-only
     gpus: List[str] = Field(default_factory=list)  # ordered preference
     gpu_count: int = 0
+    # Fractional GPU sharing (reference: Thunder virtualized pools,
+    # pkg/worker/thunder.go:31-138 + selection worker.go:128-137). On
+    # MI355X there is no external virtualization service: HIP natively
+    # time-shares one GCD across processes (HWS firmware scheduler), so
+    # the worker packs fractional containers onto a shared device and
+    # enforces the HBM budget per process (torch per-process memory
+    # fraction + arena sizing). 1.0 = exclusive whole-GPU.
+    gpu_fraction: float = 1.0
     image_id: str = ""
 
 
@@ -261,6 +269,7 @@ class ContainerRequest(BaseModel):
     gpu: str = ""
     gpu_request: List[str] = Field(default_factory=list)  # ordered preference
     gpu_count: int = 0
+    gpu_fraction: float = 1.0  # <1.0 = share a device (see RuntimeConfig)
     image_id: str = ""
     stub_id: str = ""
     workspace_id: str = ""
@@ -285,6 +294,16 @@ class ContainerRequest(BaseModel):
     def requires_gpu(self) -> bool:
         return bool(self.gpu) or bool(self.gpu_request) or self.gpu_count > 0
 
+    def gpu_units(self) -> float:
+        """Scheduling demand in GPU units: whole devices, or the shared
+        fraction for gpu_fraction<1 requests (only valid with one GPU)."""
+        if not self.requires_gpu():
+            return 0.0
+        count = max(1, self.gpu_count)
+        if self.gpu_fraction < 1.0 and count == 1:
+            return self.gpu_fraction
+        return float(count)
+
 
 class Worker(BaseModel):
     """Worker state snapshot (reference: pkg/types/scheduler.go:58-80)."""
@@ -296,7 +315,8 @@ class Worker(BaseModel):
     total_gpu_count: int = 0
     free_cpu: int = 0
     free_memory_mb: int = 0
-    free_gpu_count: int = 0
+    # float: fractional (shared-device) containers subtract their share
+    free_gpu_count: float = 0
     gpu: str = ""  # GPU type on this worker
     pool_name: str = ""
     machine_id: str = ""

@@ -157,9 +157,13 @@ class Lifecycle:
 
             if request.requires_gpu():
                 with rec.span("worker.gpu_assignment"):
-                    if self.gpu is None or self.gpu.free_count < request.gpu_count:
+                    if self.gpu is None or not self.gpu.can_fit(
+                        max(1, request.gpu_count), request.gpu_fraction
+                    ):
                         raise AllocationError("no GPUs available on worker")
-                    gpu_assignment = self.gpu.assign(cid, max(1, request.gpu_count))
+                    gpu_assignment = self.gpu.assign(
+                        cid, max(1, request.gpu_count), fraction=request.gpu_fraction
+                    )
                     spec.env.update(gpu_assignment.env)
                     spec.devices.extend(gpu_assignment.devices)
                     rdzv_env = self.gpu.provision_rendezvous(gpu_assignment)
@@ -223,6 +227,10 @@ class Lifecycle:
                     and not spec.upper_dir
                     and not spec.block_network
                     and spec.env.get("B9_DISABLE_PREWARM") != "1"
+                    # shared-GPU containers boot cold: exact-GPU shells have
+                    # a full-size arena already reserved, which would bust
+                    # the fractional HBM budget the runner must enforce
+                    and request.gpu_fraction >= 1.0
                     and self.prewarm.sees_path(spec.env.get("B9_WORKSPACE_DIR", ""))
                 )
                 if prewarm_ok:

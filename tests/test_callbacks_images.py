@@ -179,3 +179,23 @@ async def test_sandbox_snapshot_memory_reports_unsupported():
             assert r["supported"] is False and "hipsnap" in r["reason"]
         finally:
             await asyncio.to_thread(sb.terminate)
+
+
+def test_sdk_fractional_gpu_spec():
+    """gpu="mi355x:0.5" -> shared-device request (Thunder-analog syntax;
+    fraction flows stub config -> ContainerRequest -> worker)."""
+    import pytest as _pytest
+
+    import beta9_amd
+
+    fn = beta9_amd.function(gpu="MI355X:0.5")(lambda: 1)
+    cfg = fn._stub_config()
+    assert cfg.runtime.gpu == "MI355X"
+    assert cfg.runtime.gpu_fraction == 0.5
+    assert cfg.runtime.gpu_count == 1
+    with _pytest.raises(ValueError):
+        beta9_amd.function(gpu="MI355X:1.5")(lambda: 1)
+    with _pytest.raises(ValueError):
+        beta9_amd.function(gpu="MI355X:0.5", gpu_count=2)(lambda: 1)
+    # whole-GPU spec unchanged
+    assert beta9_amd.function(gpu="MI355X")(lambda: 1)._stub_config().runtime.gpu_fraction == 1.0

@@ -106,3 +106,53 @@ def test_device_spec(topo):
     assert spec["kind"] == "amd.com/gpu"
     assert len(spec["devices"]) == 8
     assert spec["devices"][0]["nodes"][0] == "/dev/kfd"
+
+
+def test_fractional_packing(topo):
+    """Thunder-analog GPU sharing: fractional containers best-fit pack onto
+    one shared device, keeping whole GPUs free for exclusive requests
+    (reference: pkg/worker/thunder.go:31-138 selection worker.go:128-137)."""
+    m = GPUManager(topology_path=topo)
+    a = m.assign("f1", 1, fraction=0.5)
+    b = m.assign("f2", 1, fraction=0.5)
+    # both halves land on the SAME device; 7 whole GPUs untouched
+    assert a.ordinals == b.ordinals
+    assert m.free_count == 7
+    assert a.env["B9_GPU_MEMORY_FRACTION"] == "0.5"
+    assert len(a.devices) == 2 and "/dev/kfd" in a.devices
+    # the shared device is exhausted: a third half must take a NEW device
+    c = m.assign("f3", 1, fraction=0.5)
+    assert c.ordinals != a.ordinals
+    assert m.free_count == 6
+    # release one half of the first device -> next half packs back onto it
+    m.unassign("f1")
+    d = m.assign("f4", 1, fraction=0.5)
+    assert d.ordinals == a.ordinals
+    # releasing everything restores whole-device capacity
+    for cid in ("f2", "f3", "f4"):
+        m.unassign(cid)
+    assert m.free_count == 8
+    assert m.free_units == 8.0
+
+
+def test_fractional_vs_exclusive(topo):
+    m = GPUManager(topology_path=topo)
+    m.assign("big", 7)  # one whole device left
+    m.assign("f1", 1, fraction=0.25)
+    # the only device is now shared: exclusive requests must fail...
+    assert m.free_count == 0
+    with pytest.raises(AllocationError):
+        m.assign("excl", 1)
+    # ...but fractions that still fit succeed, and oversubscription fails
+    m.assign("f2", 1, fraction=0.5)
+    assert not m.can_fit(1, 0.5)
+    with pytest.raises(AllocationError):
+        m.assign("f3", 1, fraction=0.5)
+    m.assign("f3", 1, fraction=0.25)
+    assert m.free_units == pytest.approx(0.0)
+
+
+def test_fractional_invalid(topo):
+    m = GPUManager(topology_path=topo)
+    with pytest.raises(AllocationError):
+        m.assign("bad", 1, fraction=-0.5)

@@ -194,3 +194,27 @@ def test_oom_classification(tmp_path):
     entry.cgroup_dir = str(cg)
     entry.oom_killed = False
     assert rt._check_oom(entry) is True
+
+
+@pytest.mark.asyncio
+async def test_fractional_gpu_env_injection(tmp_path):
+    """A gpu_fraction=0.5 container gets the shared device + the HBM budget
+    env the runner enforces (Thunder-analog — pkg/worker/thunder.go)."""
+    lc, gm = make_lifecycle(tmp_path, gpus=1)
+    req = ContainerRequest(
+        container_id="frac1",
+        stub_id="s1",
+        workspace_id="ws1",
+        stub_type="function",
+        gpu="MI355X",
+        gpu_count=1,
+        gpu_fraction=0.5,
+        cpu=100,
+        memory_mb=64,
+        entry_point=["env"],
+    )
+    code = await lc.run_container(req)
+    assert code == 0
+    assert gm.free_count == 1 and gm.free_units == 1.0  # released after exit
+    logs = "\n".join(lc.repo.logs)
+    assert "B9_GPU_MEMORY_FRACTION=0.5" in logs

@@ -246,3 +246,33 @@ def test_zygote_miss_path_on_gpu():
                 await w.shutdown()
 
     asyncio.run(amain())
+
+
+def test_fractional_gpu_memory_budget():
+    """Shared-GPU container enforcement on hardware: with
+    B9_GPU_MEMORY_FRACTION set, GPUBoot fences the torch allocator — an
+    in-budget allocation succeeds, an over-budget one raises OOM instead
+    of eating the neighbor's HBM (Thunder-analog, pkg/worker/thunder.go)."""
+    code = (
+        "import os, torch\n"
+        "os.environ['B9_GPU_MEMORY_FRACTION'] = '0.05'\n"
+        "from beta9_amd.runner.common import RunnerEnv, GPUBoot\n"
+        "env = RunnerEnv.from_env()\n"
+        "env.gpu_devices = ['0']\n"
+        "boot = GPUBoot(env); boot.boot()\n"
+        "assert boot.phases.get('gpu_memory_fraction') == 0.05, boot.phases\n"
+        "total = torch.cuda.get_device_properties(0).total_memory\n"
+        "ok = torch.empty(int(total*0.005), dtype=torch.uint8, device='cuda:0')\n"
+        "try:\n"
+        "    big = torch.empty(int(total*0.2), dtype=torch.uint8, device='cuda:0')\n"
+        "    raise SystemExit('over-budget allocation was NOT fenced')\n"
+        "except torch.cuda.OutOfMemoryError:\n"
+        "    print('budget enforced')\n"
+    )
+    r = subprocess.run(
+        [sys.executable, "-c", code],
+        capture_output=True, text=True, timeout=180,
+        env={**os.environ, "PYTHONPATH": os.getcwd()},
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "budget enforced" in r.stdout

@@ -208,3 +208,37 @@ async def test_reservation_prevents_duplicate_provisioning():
         assert "c-res" not in s._reservations  # cleared on commit
     finally:
         await s.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_fractional_gpu_capacity():
+    """Two half-GPU containers fit a 1-GPU worker; a whole-GPU request is
+    then rejected until a share is released (Thunder-analog sharing —
+    pkg/worker/thunder.go; capacity in fractional GPU units)."""
+    s, wr, cr, store = await make_sched()
+    w = make_worker("w1", gpus=1, gpu="MI355X")
+    await wr.register(w)
+    half = lambda cid: ContainerRequest(
+        container_id=cid, cpu=100, memory_mb=128, gpu="MI355X", gpu_count=1,
+        gpu_fraction=0.5,
+    )
+    r1, r2 = half("c1"), half("c2")
+    s._normalize_gpu(r1); s._normalize_gpu(r2)
+    assert r1.gpu_units() == 0.5
+    assert await wr.schedule_container_request(await wr.get("w1"), r1)
+    assert await wr.schedule_container_request(await wr.get("w1"), r2)
+    cur = await wr.get("w1")
+    assert cur.free_gpu_count == pytest.approx(0.0)
+    # whole-GPU request no longer fits this worker
+    whole = ContainerRequest(container_id="c3", cpu=100, memory_mb=128, gpu="MI355X", gpu_count=1)
+    s._normalize_gpu(whole)
+    assert s._filter(whole, [cur]) == []
+    assert not await wr.schedule_container_request(cur, whole)
+    # releasing one share frees half a unit — still not enough for a whole
+    await wr.release_capacity("w1", r1)
+    cur = await wr.get("w1")
+    assert cur.free_gpu_count == pytest.approx(0.5)
+    assert not await wr.schedule_container_request(cur, whole)
+    # but enough for another half
+    r4 = half("c4"); s._normalize_gpu(r4)
+    assert await wr.schedule_container_request(cur, r4)
