@@ -92,3 +92,10 @@ def pidns_probe(**kw):
     with open("/isolation-canary", "w") as f:
         f.write("x")
     return {"pid": _os.getpid(), "canary": _os.path.exists("/isolation-canary")}
+
+
+def gpu_envdump():
+    return {
+        "visible": os.environ.get("HIP_VISIBLE_DEVICES", ""),
+        "fraction": os.environ.get("B9_GPU_MEMORY_FRACTION", ""),
+    }

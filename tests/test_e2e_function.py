@@ -59,3 +59,24 @@ async def test_secret_injection():
         assert result["secret"] == "s3cr3t"
         assert result["stub_type"] == "function"
         assert result["container_id"].startswith("function-")
+
+
+@pytest.mark.asyncio
+async def test_fractional_gpu_sharing_e2e():
+    """Two half-GPU functions run CONCURRENTLY on a 1-GPU worker, packed
+    onto the same device with the HBM-budget env set; a whole-GPU request
+    gets 'no capacity' while the shares hold the device (Thunder-analog
+    sharing, pkg/worker/thunder.go — fraction flows SDK -> scheduler ->
+    worker -> runner env)."""
+    async with cluster(n_workers=1, fake_gpus=1):
+        import handlers
+
+        fn = beta9_amd.Function(
+            handlers.gpu_envdump, cpu=100, memory=128, gpu="MI355X:0.5"
+        )
+        r1, r2 = await asyncio.gather(
+            asyncio.to_thread(fn.remote), asyncio.to_thread(fn.remote)
+        )
+        assert r1["fraction"] == "0.5" and r2["fraction"] == "0.5"
+        # both shares landed on the single device
+        assert r1["visible"] == r2["visible"] != ""
