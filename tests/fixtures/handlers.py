@@ -99,3 +99,29 @@ def gpu_envdump():
         "visible": os.environ.get("HIP_VISIBLE_DEVICES", ""),
         "fraction": os.environ.get("B9_GPU_MEMORY_FRACTION", ""),
     }
+
+
+def frac_serving_steps(steps=100, tokens=256, hidden=2048, inter=4096, layers=4):
+    """GPU serving burst for the fractional-sharing demo: N inference steps
+    of the synthetic serving model, returns per-container throughput."""
+    import time as _time
+
+    import torch
+
+    from beta9_amd.models.mlp import SyntheticServingModel
+
+    m = SyntheticServingModel(int(hidden), int(inter), int(layers), "cuda:0")
+    for _ in range(3):
+        m.infer(batch_tokens=int(tokens))
+    torch.cuda.synchronize()
+    t0 = _time.perf_counter()
+    for _ in range(int(steps)):
+        m.infer(batch_tokens=int(tokens))
+    torch.cuda.synchronize()
+    dt = _time.perf_counter() - t0
+    return {
+        "steps_per_s": int(steps) / dt,
+        "ms_per_step": dt / int(steps) * 1000,
+        "visible": __import__("os").environ.get("HIP_VISIBLE_DEVICES", ""),
+        "fraction": __import__("os").environ.get("B9_GPU_MEMORY_FRACTION", ""),
+    }
