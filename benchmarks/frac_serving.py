@@ -14,9 +14,16 @@ import os
 import sys
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+FIXTURES = os.path.join(REPO, "tests", "fixtures")
 sys.path.insert(0, REPO)
-sys.path.insert(0, os.path.join(REPO, "tests", "fixtures"))
+sys.path.insert(0, FIXTURES)
 sys.path.insert(0, os.path.join(REPO, "tests"))
+# container runners import the handler by module name — they need the
+# fixtures dir on PYTHONPATH too (same as tests/conftest.py does)
+_pp = os.environ.get("PYTHONPATH", "")
+os.environ["PYTHONPATH"] = os.pathsep.join(
+    [FIXTURES, REPO] + ([_pp] if _pp else [])
+)
 
 
 async def main() -> None:
