@@ -265,6 +265,7 @@ class Lifecycle:
             self.usage.container_started(
                 cid, gpu=request.gpu, gpu_count=request.gpu_count,
                 cpu=request.cpu, memory_mb=request.memory_mb,
+                gpu_fraction=request.gpu_fraction,
             )
             if needs_addr:
                 with rec.span("worker.publish_address"):

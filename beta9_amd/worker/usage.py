@@ -14,11 +14,13 @@ class UsageTracker:
         self.repo = repo
         self._open: Dict[str, Dict] = {}
 
-    def container_started(self, container_id: str, gpu: str = "", gpu_count: int = 0, cpu: int = 0, memory_mb: int = 0) -> None:
+    def container_started(self, container_id: str, gpu: str = "", gpu_count: int = 0, cpu: int = 0, memory_mb: int = 0, gpu_fraction: float = 1.0) -> None:
         self._open[container_id] = {
             "started_at": time.time(),
             "gpu": gpu,
             "gpu_count": gpu_count,
+            # shared-device containers bill their fraction of a GPU
+            "gpu_units": (gpu_count or 0) * gpu_fraction if gpu_fraction < 1.0 else gpu_count,
             "cpu": cpu,
             "memory_mb": memory_mb,
         }
@@ -28,16 +30,17 @@ class UsageTracker:
         if rec is None:
             return None
         dur = time.time() - rec["started_at"]
+        units = rec.get("gpu_units", rec["gpu_count"])
         record = {
             "container_id": container_id,
             "duration_s": round(dur, 3),
-            "gpu_seconds": round(dur * rec["gpu_count"], 3),
+            "gpu_seconds": round(dur * units, 3),
             "cpu_millicore_seconds": round(dur * rec["cpu"], 1),
             "exit_code": exit_code,
             **{k: rec[k] for k in ("gpu", "gpu_count", "cpu", "memory_mb")},
         }
         REGISTRY.inc("container_seconds_total", dur, gpu=rec["gpu"] or "cpu")
-        if rec["gpu_count"]:
-            REGISTRY.inc("gpu_seconds_total", dur * rec["gpu_count"], gpu=rec["gpu"])
+        if units:
+            REGISTRY.inc("gpu_seconds_total", dur * units, gpu=rec["gpu"])
         REGISTRY.observe("container_duration_s", dur)
         return record
