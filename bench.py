@@ -259,19 +259,29 @@ async def run_bench(args, rank: int, world: int, local_rank: int) -> dict:
     except (KeyError, TypeError):
         first_load = {}
 
-    async def step() -> int:
-        results = await asyncio.gather(
-            *(rpc_invoke(client, my_stub, args.tokens, str(rank))
-              for _ in range(args.concurrency))
-        )
-        failed = sum(1 for r in results if not r["ok"])
-        if failed:
-            raise RuntimeError(f"{failed}/{len(results)} requests failed in step")
-        return len(results)
+    # ---- load generator subprocess -----------------------------------------
+    # Serving clients are separate processes in production; driving the
+    # request loop from THIS process would share rank 0's event loop with
+    # the gateway+worker and measure loop contention, not the platform.
+    import subprocess
 
-    dbg("warmup begin")
-    for _ in range(args.warmup):
-        await step()
+    lg_env = dict(os.environ)
+    lg_env["PYTHONPATH"] = REPO + os.pathsep + lg_env.get("PYTHONPATH", "")
+    # the loadgen does no collectives/GPU work — keep torchrun's dist env
+    # out of it so nothing tries to rendezvous
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT"):
+        lg_env.pop(k, None)
+    lg = subprocess.Popen(
+        [sys.executable, os.path.join(REPO, "benchmarks", "loadgen.py"),
+         info["addr"], info["token"], my_stub, str(args.concurrency),
+         str(args.steps), str(args.warmup), str(args.tokens), "600.0"],
+        stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True, env=lg_env,
+    )
+
+    dbg("warmup begin (loadgen)")
+    ready = await asyncio.to_thread(lg.stdout.readline)
+    if ready.strip() != "READY":
+        raise RuntimeError(f"loadgen failed before READY: {ready!r} rc={lg.poll()}")
     dbg("warmup done")
 
     # ---- timed region ------------------------------------------------------
@@ -283,13 +293,21 @@ async def run_bench(args, rank: int, world: int, local_rank: int) -> dict:
         torch.cuda.synchronize()
     dbg("timed begin")
     t_start = time.perf_counter()
-    done_tasks = 0
-    for _ in range(args.steps):
-        done_tasks += await step()
+    lg.stdin.write("GO\n")
+    lg.stdin.flush()
+    line = await asyncio.to_thread(lg.stdout.readline)
+    wall = time.perf_counter() - t_start
     dbg("timed end")
     if torch.cuda.is_available():
         torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t_start
+    try:
+        lg_out = json.loads(line)
+    except ValueError:
+        raise RuntimeError(f"loadgen died: {line!r} rc={lg.poll()}")
+    lg.wait(timeout=30)
+    # parent wall includes GO/result pipe IPC; report the larger (honest)
+    elapsed = max(lg_out["elapsed"], wall - 0.002)
+    done_tasks = lg_out["tasks"]
     if use_dist:
         await asyncio.to_thread(dist.barrier)
         dev = "cuda" if (dist.get_backend() == "nccl") else "cpu"

@@ -12,7 +12,8 @@ from __future__ import annotations
 import asyncio
 import logging
 import os
-from typing import Dict, List, Optional
+import time
+from typing import Dict, List, Optional, Tuple
 
 from ..backend.db import BackendDB
 from ..backend.repos import ContainerRepo, WorkerRepo
@@ -60,18 +61,34 @@ class Gateway:
         self._runner_tokens: Dict[str, str] = {}  # workspace_id -> token
         self._started = False
         self.draining = False  # reference: drainMiddleware + 10 s propagation
+        self._auth_cache: Dict[str, Tuple[Dict, float]] = {}
 
     # -- auth ----------------------------------------------------------------
     async def _auth(self, token: str, full_method: str):
+        # per-token TTL cache: authorize_token is a blocking sqlite query
+        # and this runs on EVERY rpc (reference: workspace auth cache,
+        # pkg/repository — token lookups cached in Redis). Revocation
+        # invalidates via _auth_evict.
+        now = time.monotonic()
+        hit = self._auth_cache.get(token)
+        if hit is not None and hit[1] > now:
+            return hit[0]
         ctx = self.db.authorize_token(token)
         if ctx is None:
             return None
-        return {
+        out = {
             "workspace_id": ctx["workspace_id"],
             "workspace_name": ctx["workspace_name"],
             "token": token,
             "token_type": ctx.get("token_type", "workspace"),
         }
+        if len(self._auth_cache) > 4096:
+            self._auth_cache.clear()
+        self._auth_cache[token] = (out, now + 60.0)
+        return out
+
+    def _auth_evict(self, token: str) -> None:
+        self._auth_cache.pop(token, None)
 
     def bootstrap_workspace(self, name: str = "default") -> Dict[str, str]:
         """Create (or fetch) a workspace + cluster-admin token — first-boot
@@ -222,6 +239,7 @@ class Gateway:
             self.scheduler,
             objects_path=os.path.join(self.cfg.storage.root, "objects"),
         )
+        gw_svc.auth_evict = self._auth_evict  # revocation invalidates cache
         os.makedirs(self.cfg.checkpoint.store_path, exist_ok=True)
         runner_svc = RunnerService(
             self.dispatcher, self.store, self.db,

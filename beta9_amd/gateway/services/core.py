@@ -106,6 +106,9 @@ class GatewayService:
         if row["workspace_id"] != auth["workspace_id"] and not _is_admin(auth):
             raise permission_denied("cannot revoke another workspace's token")
         self.db.revoke_token(req["token"])
+        evict = getattr(self, "auth_evict", None)
+        if evict is not None:
+            evict(req["token"])
         return True
 
     @rpc()
