@@ -132,6 +132,7 @@ class BackendDB:
         self._conn = sqlite3.connect(path, check_same_thread=False)
         self._conn.row_factory = sqlite3.Row
         self._lock = threading.Lock()
+        self._stub_cache: Dict[str, Dict[str, Any]] = {}
         with self._lock:
             self._conn.execute("PRAGMA journal_mode=WAL")
             self._conn.execute("PRAGMA synchronous=NORMAL")
@@ -319,8 +320,20 @@ class BackendDB:
         }
 
     def get_stub(self, stub_id: str) -> Optional[Dict[str, Any]]:
+        # stub rows are immutable after creation (a new config hashes to a
+        # new stub) and this runs on every endpoint invoke — cache hits
+        # skip the blocking sqlite query on the gateway event loop
+        hit = self._stub_cache.get(stub_id)
+        if hit is not None:
+            return dict(hit)
         rows = self._query("SELECT * FROM stubs WHERE id=?", (stub_id,))
-        return dict(rows[0]) if rows else None
+        if not rows:
+            return None
+        row = dict(rows[0])
+        if len(self._stub_cache) > 4096:
+            self._stub_cache.clear()
+        self._stub_cache[stub_id] = dict(row)
+        return row
 
     def get_stub_config(self, stub_id: str) -> Optional[StubConfig]:
         row = self.get_stub(stub_id)
@@ -330,6 +343,7 @@ class BackendDB:
 
     def update_stub_config(self, stub_id: str, config: StubConfig) -> None:
         self._exec("UPDATE stubs SET config=? WHERE id=?", (config.model_dump_json(), stub_id))
+        self._stub_cache.pop(stub_id, None)
 
     # -- deployments -----------------------------------------------------------
     def create_deployment(

@@ -14,10 +14,13 @@ by design (synthetic handlers / random-init weights).
 from __future__ import annotations
 
 import math
+import threading
 import time
 from typing import Dict, List, Optional
 
 import torch
+
+_tls = threading.local()  # per-runner-thread HIP stream (see infer)
 
 from ..ops import gemm as ops
 
@@ -45,13 +48,22 @@ class MLPBlock:
         self._fp8 = None  # lazily-built w8a8 gate/up/down weights
         self._act_scale = None  # calibrated static activation scale
         self._act2_scale = None  # calibrated silu-output scale (down input)
+        self._init_lock = threading.Lock()
 
     def _fp8_weights(self):
+        # double-checked: concurrent warmup requests run on per-thread
+        # streams (infer), so quantize once under a lock and synchronize
+        # BEFORE publishing — a reader that sees non-None must never see
+        # half-written weights from another stream
         if self._fp8 is None:
-            gq, gs = ops.quantize_fp8_rowwise(self.w_gate)
-            uq, us = ops.quantize_fp8_rowwise(self.w_up)
-            dq, ds = ops.quantize_fp8_rowwise(self.w_down)
-            self._fp8 = (gq, gs, uq, us, dq, ds)
+            with self._init_lock:
+                if self._fp8 is None:
+                    gq, gs = ops.quantize_fp8_rowwise(self.w_gate)
+                    uq, us = ops.quantize_fp8_rowwise(self.w_up)
+                    dq, ds = ops.quantize_fp8_rowwise(self.w_down)
+                    if self.w_gate.is_cuda:
+                        torch.cuda.synchronize()
+                    self._fp8 = (gq, gs, uq, us, dq, ds)
         return self._fp8
 
     def _calibrate(self, h: torch.Tensor) -> float:
@@ -193,10 +205,29 @@ class SyntheticServingModel:
         return x
 
     def infer(self, batch_tokens: int = 128) -> Dict:
-        x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
-        t0 = time.perf_counter()
-        y = self.forward(x)
-        torch.cuda.synchronize()
+        # Concurrent requests land here on runner threads; a per-thread HIP
+        # stream lets their kernels overlap on the GPU (the default stream
+        # would serialize them). Weights are read-only so cross-stream
+        # sharing is safe; torch's caching allocator is stream-aware.
+        stream = None
+        if self.device.startswith("cuda"):
+            stream = getattr(_tls, "stream", None)
+            if stream is None:
+                stream = torch.cuda.Stream()
+                _tls.stream = stream
+        if stream is not None:
+            with torch.cuda.stream(stream):
+                # input created on the same stream its consumers run on
+                x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
+                t0 = time.perf_counter()
+                y = self.forward(x)
+            stream.synchronize()
+        else:
+            x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
+            t0 = time.perf_counter()
+            y = self.forward(x)
+            if self.device.startswith("cuda"):
+                torch.cuda.synchronize()
         dt = time.perf_counter() - t0
         flops = 2 * batch_tokens * 3 * self.hidden * self.inter * len(self.blocks)
         return {
