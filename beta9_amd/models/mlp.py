@@ -50,6 +50,17 @@ class MLPBlock:
         self._act2_scale = None  # calibrated silu-output scale (down input)
         self._init_lock = threading.Lock()
 
+    def __getstate__(self):
+        # objsnap pickles the live model object (checkpoint flow); locks
+        # aren't picklable and are recreated on restore
+        d = dict(self.__dict__)
+        d.pop("_init_lock", None)
+        return d
+
+    def __setstate__(self, d):
+        self.__dict__.update(d)
+        self._init_lock = threading.Lock()
+
     def _fp8_weights(self):
         # double-checked: concurrent warmup requests run on per-thread
         # streams (infer), so quantize once under a lock and synchronize
@@ -189,6 +200,10 @@ class SyntheticServingModel:
             b.w_up = view(f"l{i}.w_up", (inter, hidden))
             b.w_down = view(f"l{i}.w_down", (hidden, inter))
             b.norm_w = view(f"l{i}.norm_w", (hidden,))
+            b._fp8 = None
+            b._act_scale = None
+            b._act2_scale = None
+            b._init_lock = threading.Lock()
             self.blocks.append(b)
         return self
 
