@@ -9,10 +9,13 @@ bench.py --model transformer.
 from __future__ import annotations
 
 import math
+import threading
 import time
 from typing import Dict, Optional
 
 import torch
+
+_tls = threading.local()  # per-runner-thread HIP stream (see infer)
 
 from ..ops import gemm as ops
 
@@ -102,11 +105,30 @@ class TransformerServingModel:
         return total
 
     def infer(self, batch_tokens: int = 256) -> Dict:
-        x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
-        t0 = time.perf_counter()
-        for b in self.blocks:
-            x = b.forward(x)
-        torch.cuda.synchronize()
+        # per-thread HIP stream: concurrent serving requests overlap on the
+        # GPU instead of serializing on the default stream (weights are
+        # read-only; torch's allocator is stream-aware) — same pattern as
+        # SyntheticServingModel.infer
+        stream = None
+        if self.device.startswith("cuda"):
+            stream = getattr(_tls, "stream", None)
+            if stream is None:
+                stream = torch.cuda.Stream()
+                _tls.stream = stream
+        if stream is not None:
+            with torch.cuda.stream(stream):
+                x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
+                t0 = time.perf_counter()
+                for b in self.blocks:
+                    x = b.forward(x)
+            stream.synchronize()
+        else:
+            x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)
+            t0 = time.perf_counter()
+            for b in self.blocks:
+                x = b.forward(x)
+            if self.device.startswith("cuda"):
+                torch.cuda.synchronize()
         dt = time.perf_counter() - t0
         # flops: attention projections + sdpa + mlp
         L = len(self.blocks)
