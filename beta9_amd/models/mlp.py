@@ -226,10 +226,9 @@ class SyntheticServingModel:
         # sharing is safe; torch's caching allocator is stream-aware.
         stream = None
         if self.device.startswith("cuda"):
-            stream = getattr(_tls, "stream", None)
-            if stream is None:
-                stream = torch.cuda.Stream()
-                _tls.stream = stream
+            from ..ops import streams as _streams
+
+            stream = _streams.get_thread_stream()
         if stream is not None:
             with torch.cuda.stream(stream):
                 # input created on the same stream its consumers run on

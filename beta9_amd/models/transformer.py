@@ -111,10 +111,9 @@ class TransformerServingModel:
         # SyntheticServingModel.infer
         stream = None
         if self.device.startswith("cuda"):
-            stream = getattr(_tls, "stream", None)
-            if stream is None:
-                stream = torch.cuda.Stream()
-                _tls.stream = stream
+            from ..ops import streams as _streams
+
+            stream = _streams.get_thread_stream()
         if stream is not None:
             with torch.cuda.stream(stream):
                 x = torch.randn(batch_tokens, self.hidden, dtype=torch.bfloat16, device=self.device)

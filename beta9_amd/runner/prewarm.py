@@ -50,6 +50,11 @@ def main() -> int:
 
                 if _g.has_native():
                     _g.warm_all()  # every kernel's first-launch cost
+                from beta9_amd.ops import streams as _streams
+
+                # per-request serving streams: pay HSA hw-queue creation
+                # here, not in the first concurrent requests
+                _streams.prewarm(8)
             except Exception:
                 pass
             # pre-create the HBM arena + pinned staging (hipMalloc slab and
