@@ -20,7 +20,6 @@ from typing import Dict, List, Optional
 
 import torch
 
-_tls = threading.local()  # per-runner-thread HIP stream (see infer)
 
 from ..ops import gemm as ops
 

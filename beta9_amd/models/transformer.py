@@ -15,7 +15,6 @@ from typing import Dict, Optional
 
 import torch
 
-_tls = threading.local()  # per-runner-thread HIP stream (see infer)
 
 from ..ops import gemm as ops
 
