@@ -149,3 +149,31 @@ def test_agent_join(tmp_path):
             await gw.shutdown()
 
     asyncio.run(amain())
+
+
+def test_bench_world2_cpu_contract():
+    """The driver's SCALE run shape: torchrun world=2 over gloo on CPU —
+    one gateway+worker on rank 0, per-rank out-of-process load generators,
+    barriers + reduction, ONE JSON line from rank 0 (bench.py contract)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29377", os.path.join(repo, "bench.py"),
+         "--gpus", "2", "--steps", "3", "--warmup", "1", "--cold-iters", "1",
+         "--cold-miss-iters", "0", "--hidden", "64", "--inter", "128",
+         "--layers", "1", "--tokens", "8"],
+        capture_output=True, text=True, timeout=420, env=env, cwd=repo,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["metric"] == "tasks_per_sec" and d["n_gpus"] == 2
+    assert d["value"] > 0 and len(d["cold_start_concurrent_ms"]) == 2
