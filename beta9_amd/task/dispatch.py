@@ -73,8 +73,12 @@ class Dispatcher:
             executor=executor_name,
         )
         self.db.create_task_async(task.task_id, workspace_id, stub_id, task.status.value)
-        await self.store.set(keys.task_state(task.task_id), task.model_dump())
-        await self.store.set(keys.task_args(task.task_id), payload)
+        # kv mirrors are the HOT copies; the sqlite row is the durable
+        # record. TTL them so sustained traffic doesn't grow the store
+        # (and its WAL snapshots) without bound.
+        ttl = max(3600.0, (policy.timeout if policy else 3600.0) * 2)
+        await self.store.set(keys.task_state(task.task_id), task.model_dump(), ttl=ttl)
+        await self.store.set(keys.task_args(task.task_id), payload, ttl=ttl)
         await self._execute(task, payload)
         return task
 
@@ -99,7 +103,10 @@ class Dispatcher:
             task.started_at = time.time()
         if status.is_terminal:
             task.ended_at = time.time()
-        await self.store.set(keys.task_state(task_id), task.model_dump())
+        # terminal states keep a short kv tail for result polling; running
+        # tasks keep the long lease (see send_and_execute)
+        ttl = 3600.0 if status.is_terminal else max(3600.0, task.policy.timeout * 2)
+        await self.store.set(keys.task_state(task_id), task.model_dump(), ttl=ttl)
         self.db.update_task_async(
             task_id,
             status=status.value,
@@ -273,7 +280,10 @@ class Dispatcher:
         task.retries += 1
         task.status = TaskStatus.retry
         await self.store.delete(keys.task_claim(task.task_id))
-        await self.store.set(keys.task_state(task.task_id), task.model_dump())
+        await self.store.set(
+            keys.task_state(task.task_id), task.model_dump(),
+            ttl=max(3600.0, task.policy.timeout * 2),
+        )
         payload = await self.store.get(keys.task_args(task.task_id))
         log.info("retrying task %s (attempt %d)", task.task_id, task.retries)
         await self._execute(task, payload)
