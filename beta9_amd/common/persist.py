@@ -118,6 +118,10 @@ class PersistentStateStore(StateStore):
                         else None
                     )
                     self._kv[k] = _Entry(v, exp)
+                    if exp is not None:
+                        import heapq
+
+                        heapq.heappush(self._exp_heap, (exp, k))
                 self._hashes.update(state.get("hashes", {}))
                 for k, scores in state.get("zsets", {}).items():
                     z = self._zsets.setdefault(k, _ZSet())
@@ -151,6 +155,10 @@ class PersistentStateStore(StateStore):
                 else None
             )
             self._kv[key] = _Entry(value, exp)
+            if exp is not None:
+                import heapq
+
+                heapq.heappush(self._exp_heap, (exp, key))
         elif op == "del":
             self._kv.pop(rec[1], None)
             self._hashes.pop(rec[1], None)

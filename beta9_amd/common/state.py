@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import asyncio
 import fnmatch
+import heapq
 import time
 from dataclasses import dataclass, field
 from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
@@ -59,6 +60,12 @@ class StateStore:
         self._list_cond: Dict[str, asyncio.Condition] = {}
         self._subs: List[Tuple[str, asyncio.Queue]] = []
         self._sweeper: Optional[asyncio.Task] = None
+        # expiry heap (deadline, key): the sweeper pops only DUE entries —
+        # scanning every key each 50 ms tick was O(total keys) and measured
+        # quadratic under sustained serving load (74M expired() calls in a
+        # 1000-step bench). Stale heap entries (key overwritten with a new
+        # ttl) are re-checked against the live entry and dropped.
+        self._exp_heap: List[Tuple[float, str]] = []
 
     # -- lifecycle ---------------------------------------------------------
     async def start(self) -> None:
@@ -78,10 +85,13 @@ class StateStore:
         while True:
             await asyncio.sleep(0.05)
             now = time.monotonic()
-            dead = [k for k, e in self._kv.items() if e.expired(now)]
-            for k in dead:
-                del self._kv[k]
-                self._publish(KeyEvent.EXPIRED, k)
+            heap = self._exp_heap
+            while heap and heap[0][0] <= now:
+                _, k = heapq.heappop(heap)
+                e = self._kv.get(k)
+                if e is not None and e.expired(now):
+                    del self._kv[k]
+                    self._publish(KeyEvent.EXPIRED, k)
 
     # -- events ------------------------------------------------------------
     def _publish(self, event: str, key: str) -> None:
@@ -99,6 +109,8 @@ class StateStore:
     async def set(self, key: str, value: Any, ttl: Optional[float] = None) -> None:
         exp = time.monotonic() + ttl if ttl else None
         self._kv[key] = _Entry(value, exp)
+        if exp is not None:
+            heapq.heappush(self._exp_heap, (exp, key))
         self._publish(KeyEvent.SET, key)
 
     async def set_nx(self, key: str, value: Any, ttl: Optional[float] = None) -> bool:
@@ -141,6 +153,7 @@ class StateStore:
         if e is None or e.expired(time.monotonic()):
             return False
         e.expires_at = time.monotonic() + ttl
+        heapq.heappush(self._exp_heap, (e.expires_at, key))
         return True
 
     async def ttl(self, key: str) -> Optional[float]:
